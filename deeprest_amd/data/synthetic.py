@@ -1,0 +1,290 @@
+"""Synthetic application / trace generator.
+
+Replaces the reference's L1-L3 stack (DeathStarBench social network + Locust
+load generation + Jaeger/Prometheus collection) with a
+configurable generator that emits data in the raw-data contract:
+
+- a random microservice call graph: each API endpoint owns a few trace
+  *shapes* (span trees over components/operations) with empirical weights,
+  mirroring how compose/read endpoints fan out in the reference app
+  (reference: locust/locustfile-normal.py:88-144 drives those shapes);
+- diurnal traffic with two Gaussian peaks per simulated day and noise
+  (reference: locust/locustfile-normal.py:53-74), API mix from a Zipf
+  popularity prior;
+- ground-truth resource model: per component and resource type the
+  utilization is an affine/EMA function of that component's invocation
+  counts plus noise — so an estimator that learns traffic->resource
+  causality can actually fit it;
+- anomaly injection (utilization NOT justified by traffic) for the
+  sanity-check capability (reference README.md:3: ransomware/cryptojacking
+  detection).
+
+Two output paths:
+- ``generate_raw()``   — full span trees in the contract format (tests,
+  small configs);
+- ``generate_featurized()`` — the (T, P) traffic matrix and (T,) resource
+  series directly, skipping tree construction (bench-scale configs:
+  256/4096 endpoints).  Both paths agree: shapes' path-count vectors are
+  precomputed from the same trees the raw path emits.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+
+from .contract import validate_raw_data
+from .featurize import FeatureSpace, Featurizer, FeaturizedData
+
+DEFAULT_RESOURCES = ("cpu", "memory", "write-iops")
+
+
+@dataclass
+class SyntheticAppConfig:
+    n_apis: int = 8                      # API endpoints (root operations)
+    n_components: int = 12               # microservice components
+    resources: Tuple[str, ...] = DEFAULT_RESOURCES
+    shapes_per_api: int = 3              # distinct trace shapes per endpoint
+    max_depth: int = 4
+    max_children: int = 3
+    windows_per_day: int = 240           # discretization windows per simulated day
+    n_days: int = 4
+    base_calls: float = 20.0             # mean calls/window at trough
+    peak_calls: float = 120.0            # mean calls/window at peaks
+    noise: float = 0.05                  # traffic noise fraction
+    resource_noise: float = 0.02         # resource observation noise fraction
+    seed: int = 1234
+
+    @property
+    def n_windows(self) -> int:
+        return self.windows_per_day * self.n_days
+
+
+def _make_shape(
+    rng: np.random.Generator,
+    components: List[str],
+    root_component: str,
+    root_op: str,
+    max_depth: int,
+    max_children: int,
+) -> Dict[str, Any]:
+    """Random span tree rooted at (root_component, root_op)."""
+
+    def grow(depth: int) -> List[Dict[str, Any]]:
+        if depth >= max_depth:
+            return []
+        n_children = int(rng.integers(0, max_children + 1)) if depth > 0 else int(
+            rng.integers(1, max_children + 1)
+        )
+        children = []
+        for _ in range(n_children):
+            comp = components[int(rng.integers(0, len(components)))]
+            op = f"op{int(rng.integers(0, 4))}"
+            children.append(
+                {"component": comp, "operation": op, "children": grow(depth + 1)}
+            )
+        return children
+
+    return {"component": root_component, "operation": root_op, "children": grow(0)}
+
+
+class SyntheticApp:
+    def __init__(self, config: Optional[SyntheticAppConfig] = None, **kwargs) -> None:
+        if config is None:
+            config = SyntheticAppConfig(**kwargs)
+        self.config = config
+        rng = np.random.default_rng(config.seed)
+        self._rng = rng
+
+        self.components = [f"svc-{i:04d}" for i in range(config.n_components)]
+        # frontend component hosts all API root spans (like nginx-thrift)
+        self.frontend = "frontend"
+        self.apis: List[str] = []
+        self.shapes: Dict[str, List[Dict[str, Any]]] = {}
+        self.shape_weights: Dict[str, np.ndarray] = {}
+        for i in range(config.n_apis):
+            op = f"/api-{i:04d}"
+            api_id = f"{self.frontend}_{op}"
+            self.apis.append(api_id)
+            shapes = [
+                _make_shape(rng, self.components, self.frontend, op,
+                            config.max_depth, config.max_children)
+                for _ in range(config.shapes_per_api)
+            ]
+            self.shapes[api_id] = shapes
+            w = rng.dirichlet(np.ones(config.shapes_per_api) * 2.0)
+            self.shape_weights[api_id] = w
+
+        # API popularity (Zipf-ish) and diurnal peaks
+        pop = 1.0 / (1.0 + np.arange(config.n_apis)) ** 0.8
+        self.popularity = pop / pop.sum()
+        self.peak_positions = rng.uniform(0.15, 0.85, size=(config.n_days, 2))
+
+        # shared feature space over all shapes (discovery order: api, then shape)
+        self.feature_space = FeatureSpace()
+        for api in self.apis:
+            for shape in self.shapes[api]:
+                self.feature_space.observe_trace(shape)
+        P = len(self.feature_space)
+
+        # precompute per-shape path-count and component-count vectors
+        self._shape_vec: Dict[str, np.ndarray] = {}
+        self._shape_comp: Dict[str, np.ndarray] = {}
+        all_components = [self.frontend] + self.components
+        self.component_index = {c: i for i, c in enumerate(all_components)}
+        self.all_components = all_components
+        for api in self.apis:
+            vecs = np.zeros((config.shapes_per_api, P), dtype=np.int64)
+            comps = np.zeros((config.shapes_per_api, len(all_components)), dtype=np.int64)
+            for s, shape in enumerate(self.shapes[api]):
+                self.feature_space.count_trace(shape, vecs[s])
+                stack = [shape]
+                while stack:
+                    node = stack.pop()
+                    comps[s, self.component_index[node["component"]]] += 1
+                    stack.extend(node.get("children", []))
+            self._shape_vec[api] = vecs
+            self._shape_comp[api] = comps
+
+        # ground-truth resource model parameters per (component, resource)
+        C = len(all_components)
+        R = len(config.resources)
+        self._res_base = rng.uniform(5.0, 50.0, size=(C, R))
+        self._res_gain = rng.uniform(0.2, 2.0, size=(C, R))
+        self._res_ema = rng.uniform(0.0, 0.9, size=(C, R))  # memory-like persistence
+
+    # ------------------------------------------------------------------ traffic
+    def traffic_plan(self) -> np.ndarray:
+        """(T, n_apis) expected call counts: diurnal two-peak shape x popularity."""
+        cfg = self.config
+        T = cfg.n_windows
+        t = np.arange(cfg.windows_per_day) / cfg.windows_per_day
+        days = []
+        for d in range(cfg.n_days):
+            p1, p2 = self.peak_positions[d]
+            shape = (
+                np.exp(-0.5 * ((t - p1) / 0.08) ** 2)
+                + np.exp(-0.5 * ((t - p2) / 0.08) ** 2)
+            )
+            days.append(shape)
+        shape_all = np.concatenate(days)  # (T,)
+        level = cfg.base_calls + (cfg.peak_calls - cfg.base_calls) * shape_all
+        lam = level[:, None] * self.popularity[None, :]
+        noise = 1.0 + cfg.noise * self._rng.standard_normal(size=lam.shape)
+        counts = self._rng.poisson(np.maximum(lam * noise, 0.0)).astype(np.int64)
+        return counts
+
+    def _sample_shape_counts(self, api_calls: np.ndarray) -> Dict[str, np.ndarray]:
+        """Per api: (T, shapes_per_api) multinomial split of each window's calls."""
+        out = {}
+        for a, api in enumerate(self.apis):
+            w = self.shape_weights[api]
+            calls = api_calls[:, a]
+            picks = np.zeros((len(calls), len(w)), dtype=np.int64)
+            nz = calls > 0
+            if nz.any():
+                picks[nz] = np.stack(
+                    [self._rng.multinomial(int(n), w) for n in calls[nz]]
+                )
+            out[api] = picks
+        return out
+
+    def _resources_from_invocations(self, inv: np.ndarray) -> np.ndarray:
+        """(T, C, R) ground-truth utilization from (T, C) invocation counts."""
+        cfg = self.config
+        T, C = inv.shape
+        R = len(cfg.resources)
+        vals = np.zeros((T, C, R))
+        state = np.zeros((C, R))
+        for t in range(T):
+            drive = inv[t][:, None] * self._res_gain  # (C, R)
+            state = self._res_ema * state + (1.0 - self._res_ema) * drive
+            obs = self._res_base + state
+            vals[t] = obs
+        vals *= 1.0 + cfg.resource_noise * self._rng.standard_normal(size=vals.shape)
+        return np.maximum(vals, 0.0)
+
+    # ------------------------------------------------------------------ outputs
+    def generate_raw(self) -> List[Dict[str, Any]]:
+        """Full contract-format raw_data with span trees."""
+        cfg = self.config
+        api_calls = self.traffic_plan()
+        shape_counts = self._sample_shape_counts(api_calls)
+        T = cfg.n_windows
+        C = len(self.all_components)
+
+        inv = np.zeros((T, C), dtype=np.int64)
+        for api in self.apis:
+            inv += shape_counts[api] @ self._shape_comp[api]
+        res = self._resources_from_invocations(inv)
+
+        raw = []
+        for t in range(T):
+            traces = []
+            for api in self.apis:
+                for s, shape in enumerate(self.shapes[api]):
+                    traces.extend([shape] * int(shape_counts[api][t, s]))
+            metrics = []
+            for ci, comp in enumerate(self.all_components):
+                for ri, resource in enumerate(cfg.resources):
+                    metrics.append(
+                        {"component": comp, "resource": resource,
+                         "value": float(res[t, ci, ri])}
+                    )
+            raw.append({"metrics": metrics, "traces": traces})
+        validate_raw_data(raw)
+        return raw
+
+    def generate_featurized(self) -> FeaturizedData:
+        """Fast path: traffic matrix + resource series without building trees."""
+        cfg = self.config
+        api_calls = self.traffic_plan()
+        shape_counts = self._sample_shape_counts(api_calls)
+        T = cfg.n_windows
+        P = len(self.feature_space)
+        C = len(self.all_components)
+
+        traffic = np.zeros((T, P), dtype=np.int64)
+        inv = np.zeros((T, C), dtype=np.int64)
+        for api in self.apis:
+            traffic += shape_counts[api] @ self._shape_vec[api]
+            inv += shape_counts[api] @ self._shape_comp[api]
+        res = self._resources_from_invocations(inv)
+
+        resources = {}
+        resource_components = {}
+        for ci, comp in enumerate(self.all_components):
+            for ri, resource in enumerate(cfg.resources):
+                ident = f"{comp}_{resource}"
+                resources[ident] = res[:, ci, ri].copy()
+                resource_components[ident] = comp
+
+        invocations = {c: inv[:, i].copy() for c, i in self.component_index.items()}
+        invocations["general"] = api_calls.sum(axis=1).astype(np.int64)
+
+        return FeaturizedData(
+            traffic=traffic,
+            resources=resources,
+            invocations=invocations,
+            feature_space=self.feature_space,
+            resource_components=resource_components,
+        )
+
+    def inject_anomaly(
+        self,
+        data: FeaturizedData,
+        component: str,
+        resource: str,
+        start: int,
+        length: int,
+        magnitude: float = 3.0,
+    ) -> FeaturizedData:
+        """Add traffic-unjustified utilization (cryptojacking-style CPU thief,
+        reference locust/pow.py:29-38) to one component's series in-place."""
+        ident = f"{component}_{resource}"
+        series = data.resources[ident]
+        baseline = float(np.median(series))
+        series[start : start + length] += magnitude * max(baseline, 1.0)
+        return data

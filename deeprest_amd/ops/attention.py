@@ -1,0 +1,73 @@
+"""Multi-head attention over the endpoint x time window.
+
+The traffic encoder attends over the time axis of the encoded endpoint
+window (north star: "multi-head attention over the endpoint x time window on
+MFMA with LDS-staged tiles").  The forward is ONE fused HIP kernel per
+(batch, head): QK^T on MFMA with K staged in LDS (XOR-swizzled), online
+softmax in registers, PV on MFMA — no S x S score tensor is materialized.
+It saves the per-row logsumexp so the backward can recompute P cheaply.
+
+The backward recomputes P = exp(QK^T * scale - lse) and chains plain GEMMs
+(rocBLAS) — dV = P^T dO, dP = dO V^T, dS = P*(dP - rowsum(dP*P)),
+dQ = dS K * scale, dK = dS^T Q * scale.  Sequence lengths here are short
+(T = 60..512), so recompute is cheap and keeps the fused-forward hot path
+simple.
+
+Replaces and upgrades the reference's per-metric feature-mask "attention"
+(reference: resource-estimation/qrnn.py:21-23,34).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from .native import require_native
+
+
+def reference_mha(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                  scale: Optional[float] = None) -> torch.Tensor:
+    """q,k,v: (B, H, T, D). Plain composition (the numerics oracle)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    s = torch.matmul(q, k.transpose(-1, -2)) * scale
+    p = torch.softmax(s.float(), dim=-1).to(q.dtype)
+    return torch.matmul(p, v)
+
+
+class _MHAFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        ext = require_native("mha_forward")
+        o, lse = ext.mha_forward(q, k, v, float(scale))
+        ctx.save_for_backward(q, k, v, lse)
+        ctx.scale = float(scale)
+        return o
+
+    @staticmethod
+    def backward(ctx, grad_o):
+        q, k, v, lse = ctx.saved_tensors
+        scale = ctx.scale
+        grad_o = grad_o.contiguous()
+        # recompute P in fp32 from the saved logsumexp
+        s = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
+        p = torch.exp(s - lse.unsqueeze(-1))                  # (B, H, T, T)
+        dv = torch.matmul(p.transpose(-1, -2), grad_o.float())
+        dp = torch.matmul(grad_o.float(), v.float().transpose(-1, -2))
+        row = (dp * p).sum(dim=-1, keepdim=True)
+        ds = p * (dp - row)
+        dq = torch.matmul(ds, k.float()) * scale
+        dk = torch.matmul(ds.transpose(-1, -2), q.float()) * scale
+        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None
+
+
+def mha_forward(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                scale: Optional[float] = None) -> torch.Tensor:
+    """Fused attention forward; q,k,v: (B, H, T, D) contiguous."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        return _MHAFunction.apply(q.contiguous(), k.contiguous(), v.contiguous(), scale)
+    return reference_mha(q, k, v, scale)

@@ -1,0 +1,116 @@
+"""Fused GRU sequence op.
+
+The per-resource GRU decoders dominate the training step (SURVEY.md section 7:
+"the fused GRU cell first — it dominates runtime").  Design:
+
+- the input-side projection (x @ W_ih^T + b_ih) is a plain large GEMM done
+  once for the whole sequence by rocBLAS/hipBLASLt *outside* this op;
+- this op consumes those precomputed input gates ``x_gates (B, T, 3H)``,
+  broadcasts them over C per-component decoders through a FiLM condition
+  (gamma/beta per component), and runs the *recurrent* part — per time step
+  a (B*C, H) x (H, 3H) MFMA GEMM fused with the sigmoid/tanh gate math —
+  entirely inside ONE kernel launch for the whole sequence.  Rows evolve
+  independently, so there is no cross-workgroup dependency: each workgroup
+  keeps its row tile's hidden state in registers, stages W_hh in LDS once,
+  and loops over T.
+
+Gate layout follows PyTorch nn.GRU (r, z, n) so the CPU oracle can be
+checked against torch.nn.GRUCell:
+
+    r = sigmoid(xg_r + h W_hr + b_hr)
+    z = sigmoid(xg_z + h W_hz + b_hz)
+    n = tanh(xg_n + r * (h W_hn + b_hn))
+    h' = (1 - z) * n + z * h
+
+Replaces the cuDNN GRU call of the reference (reference:
+resource-estimation/qrnn.py:24,41) — and allocates hidden state on-device,
+avoiding the reference's CPU-hidden-state bug (qrnn.py:39-40).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from .native import require_native
+
+
+def reference_gru_sequence(
+    x_gates: torch.Tensor,       # (B, T, 3H) precomputed input gates
+    w_hh: torch.Tensor,          # (3H, H) recurrent weight (PyTorch layout)
+    b_hh: torch.Tensor,          # (3H,)
+    h0: torch.Tensor,            # (B, C, H)
+    gamma: Optional[torch.Tensor] = None,  # (C, 3H) FiLM scale
+    beta: Optional[torch.Tensor] = None,   # (C, 3H) FiLM shift
+    reverse: bool = False,
+) -> torch.Tensor:
+    """Differentiable PyTorch composition; returns h_all (B, T, C, H)."""
+    B, T, G = x_gates.shape
+    _, C, H = h0.shape
+    assert G == 3 * H, f"x_gates last dim {G} != 3*H ({3 * H})"
+    h = h0
+    outs = []
+    steps = range(T - 1, -1, -1) if reverse else range(T)
+    w_hh_t = w_hh.t()  # (H, 3H)
+    for t in steps:
+        g = x_gates[:, t, None, :]                      # (B, 1, 3H)
+        if gamma is not None:
+            g = g * gamma[None] + (beta[None] if beta is not None else 0.0)
+        else:
+            g = g.expand(B, C, G)
+        hh = h @ w_hh_t + b_hh                          # (B, C, 3H)
+        r = torch.sigmoid(g[..., :H] + hh[..., :H])
+        z = torch.sigmoid(g[..., H : 2 * H] + hh[..., H : 2 * H])
+        n = torch.tanh(g[..., 2 * H :] + r * hh[..., 2 * H :])
+        h = (1.0 - z) * n + z * h
+        outs.append(h)
+    if reverse:
+        outs.reverse()
+    return torch.stack(outs, dim=1)                     # (B, T, C, H)
+
+
+class _FusedGRUSequence(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x_gates, w_hh, b_hh, h0, gamma, beta, reverse):
+        ext = require_native("fused_gru_sequence")
+        h_all, gates_saved = ext.gru_seq_forward(
+            x_gates, w_hh, b_hh, h0, gamma, beta, bool(reverse)
+        )
+        ctx.save_for_backward(x_gates, w_hh, b_hh, h0, gamma, beta, h_all, gates_saved)
+        ctx.reverse = bool(reverse)
+        return h_all
+
+    @staticmethod
+    def backward(ctx, grad_h_all):
+        ext = require_native("fused_gru_sequence")
+        x_gates, w_hh, b_hh, h0, gamma, beta, h_all, gates_saved = ctx.saved_tensors
+        dx_gates, dw_hh, db_hh, dh0, dgamma, dbeta = ext.gru_seq_backward(
+            grad_h_all.contiguous(), x_gates, w_hh, b_hh, h0, gamma, beta,
+            h_all, gates_saved, ctx.reverse,
+        )
+        return dx_gates, dw_hh, db_hh, dh0, dgamma, dbeta, None
+
+
+def fused_gru_sequence(
+    x_gates: torch.Tensor,
+    w_hh: torch.Tensor,
+    b_hh: torch.Tensor,
+    h0: torch.Tensor,
+    gamma: Optional[torch.Tensor] = None,
+    beta: Optional[torch.Tensor] = None,
+    reverse: bool = False,
+) -> torch.Tensor:
+    if x_gates.is_cuda:
+        if gamma is None:
+            C = h0.shape[1]
+            G = x_gates.shape[-1]
+            gamma = torch.ones(C, G, device=x_gates.device, dtype=x_gates.dtype)
+            beta = torch.zeros(C, G, device=x_gates.device, dtype=x_gates.dtype)
+        elif beta is None:
+            beta = torch.zeros_like(gamma)
+        return _FusedGRUSequence.apply(
+            x_gates.contiguous(), w_hh.contiguous(), b_hh.contiguous(),
+            h0.contiguous(), gamma.contiguous(), beta.contiguous(), reverse,
+        )
+    return reference_gru_sequence(x_gates, w_hh, b_hh, h0, gamma, beta, reverse)

@@ -1,0 +1,61 @@
+import pickle
+
+import pytest
+
+from deeprest_amd.data.contract import (
+    ContractError,
+    Span,
+    Window,
+    load_raw_data,
+    save_raw_data,
+    validate_raw_data,
+    windows_from_raw,
+)
+
+
+def test_validate_ok(tiny_raw_data):
+    validate_raw_data(tiny_raw_data)
+
+
+def test_validate_rejects_bad_shapes(tiny_raw_data):
+    with pytest.raises(ContractError):
+        validate_raw_data({"not": "a list"})
+    bad = [dict(tiny_raw_data[0])]
+    del bad[0]["traces"]
+    with pytest.raises(ContractError):
+        validate_raw_data(bad)
+    bad = [{"metrics": [{"component": "a"}], "traces": []}]
+    with pytest.raises(ContractError):
+        validate_raw_data(bad)
+    bad = [{"metrics": [], "traces": [{"component": "a"}]}]  # span missing operation
+    with pytest.raises(ContractError):
+        validate_raw_data(bad)
+
+
+def test_roundtrip_pickle(tiny_raw_data, tmp_path):
+    p = str(tmp_path / "raw.pkl")
+    save_raw_data(tiny_raw_data, p)
+    loaded = load_raw_data(p)
+    assert loaded == tiny_raw_data
+    # plain pickled list of dicts — the reference on-disk format
+    with open(p, "rb") as f:
+        assert isinstance(pickle.load(f), list)
+
+
+def test_typed_view_roundtrip(tiny_raw_data):
+    windows = windows_from_raw(tiny_raw_data)
+    assert isinstance(windows[0], Window)
+    assert windows[0].traces[0].component == "frontend"
+    assert [w.to_dict() for w in windows] == tiny_raw_data
+
+
+def test_span_walk_preorder():
+    s = Span.from_dict(
+        {"component": "a", "operation": "1", "children": [
+            {"component": "b", "operation": "2", "children": [
+                {"component": "c", "operation": "3", "children": []}]},
+            {"component": "d", "operation": "4", "children": []},
+        ]}
+    )
+    order = [n.component for n in s.walk()]
+    assert order == ["a", "b", "c", "d"]

@@ -1,0 +1,100 @@
+import numpy as np
+
+from deeprest_amd.data.featurize import FeatureSpace, Featurizer, FeaturizedData
+
+
+def test_feature_space_paths_and_order(tiny_raw_data):
+    fz = Featurizer(use_native=False).fit(tiny_raw_data)
+    fs = fz.feature_space
+    # /register root, its two children, /compose root, its subtree
+    expected_paths = [
+        (("frontend", "/register"),),
+        (("frontend", "/register"), ("user-db", "/find")),
+        (("frontend", "/register"), ("user-db", "/store")),
+        (("frontend", "/compose"),),
+        (("frontend", "/compose"), ("text-svc", "/parse")),
+        (("frontend", "/compose"), ("text-svc", "/parse"), ("user-db", "/find")),
+    ]
+    assert fs.paths == expected_paths
+    assert len(fs) == 6
+
+
+def test_traffic_counts(tiny_raw_data):
+    data = Featurizer(use_native=False).fit_transform(tiny_raw_data)
+    # window 0: 1 register + 1 compose -> every path once
+    assert data.traffic[0].tolist() == [1, 1, 1, 1, 1, 1]
+    # window 1: 2 compose
+    assert data.traffic[1].tolist() == [0, 0, 0, 2, 2, 2]
+    # window 2: 1 register
+    assert data.traffic[2].tolist() == [1, 1, 1, 0, 0, 0]
+
+
+def test_same_subtree_different_roots_distinct_paths():
+    # the SAME (component, op) node under different roots must be distinct features
+    raw = [{
+        "metrics": [{"component": "x", "resource": "cpu", "value": 1.0}],
+        "traces": [
+            {"component": "f", "operation": "/a", "children": [
+                {"component": "x", "operation": "op", "children": []}]},
+            {"component": "f", "operation": "/b", "children": [
+                {"component": "x", "operation": "op", "children": []}]},
+        ],
+    }]
+    data = Featurizer(use_native=False).fit_transform(raw)
+    assert data.num_paths == 4  # /a, /a->x, /b, /b->x
+
+
+def test_resources_and_invocations(tiny_raw_data):
+    data = Featurizer(use_native=False).fit_transform(tiny_raw_data)
+    assert data.metric_names == ["frontend_cpu", "user-db_cpu", "user-db_memory"]
+    np.testing.assert_allclose(data.resources["frontend_cpu"], [5.0, 7.5, 1.0])
+    np.testing.assert_allclose(data.resources["user-db_memory"], [100.0, 110.0, 95.0])
+    # invocations: spans per component; 'general' = #traces
+    assert data.invocations["general"].tolist() == [2, 2, 1]
+    assert data.invocations["frontend"].tolist() == [2, 2, 1]
+    # w0: register(2 user-db children) + compose(1 nested find) = 3
+    assert data.invocations["user-db"].tolist() == [3, 2, 2]
+    assert data.invocations["text-svc"].tolist() == [1, 2, 0]
+    assert data.resource_components["user-db_memory"] == "user-db"
+
+
+def test_underscore_in_names_supported():
+    # the reference breaks on components with '_' (featurize.py:92); we must not
+    raw = [{
+        "metrics": [{"component": "my_svc", "resource": "cpu", "value": 1.0}],
+        "traces": [{"component": "my_svc", "operation": "do_thing", "children": []}],
+    }]
+    data = Featurizer(use_native=False).fit_transform(raw)
+    assert data.num_paths == 1
+    assert "my_svc" in data.invocations
+
+
+def test_input_list_roundtrip(tiny_raw_data, tmp_path):
+    data = Featurizer(use_native=False).fit_transform(tiny_raw_data)
+    p = str(tmp_path / "input.pkl")
+    data.save(p)
+    loaded = FeaturizedData.load(p)
+    np.testing.assert_array_equal(loaded.traffic, data.traffic)
+    assert set(loaded.resources) == set(data.resources)
+    assert set(loaded.invocations) == set(data.invocations)
+
+
+def test_feature_space_state_dict_roundtrip(tiny_raw_data):
+    fz = Featurizer(use_native=False).fit(tiny_raw_data)
+    fs2 = FeatureSpace.from_state_dict(fz.feature_space.state_dict())
+    assert fs2.paths == fz.feature_space.paths
+    vec = np.zeros(len(fs2), dtype=np.int64)
+    fs2.count_trace(tiny_raw_data[0]["traces"][0], vec)
+    assert vec.sum() == 3
+
+
+def test_deep_trace_no_recursion_limit():
+    # 50k-deep chain — the reference's recursive DFS would blow the stack
+    node = {"component": "leaf", "operation": "op", "children": []}
+    for i in range(50_000):
+        node = {"component": f"c{i % 7}", "operation": "op", "children": [node]}
+    raw = [{"metrics": [{"component": "c0", "resource": "cpu", "value": 1.0}],
+            "traces": [node]}]
+    data = Featurizer(use_native=False).fit_transform(raw)
+    assert data.num_paths == 50_001
+    assert data.traffic[0].sum() == 50_001
