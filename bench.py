@@ -1,0 +1,158 @@
+#!/usr/bin/env python3
+"""Flagship training-step benchmark (driver contract).
+
+Measures the BASELINE.json headline metric: training samples(windows)/sec of
+the 256-endpoint / 3-resource estimation model, weak-scaled over 1..8 MI355X
+GPUs (per-GPU batch fixed).  Synthetic traffic of the named shape, random
+init weights (no network access for datasets), full training step timed:
+forward + pinball loss + backward + gradient all-reduce + fused Adam step.
+
+Usage (driver):
+  python bench.py --gpus N --steps K --warmup W
+  torchrun --nnodes=1 --nproc-per-node N bench.py --gpus N --steps K --warmup W
+
+Rank 0 prints exactly one JSON line with the whole-job aggregate.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+from deeprest_amd.engine.dataset import EstimationDataset
+from deeprest_amd.models.net import DeepRestNet, DeepRestNetConfig, build_model_spec
+from deeprest_amd.ops.adam import FusedAdam
+from deeprest_amd.parallel.dist import DistContext, init_distributed
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch", type=int, default=32, help="per-GPU window batch")
+    p.add_argument("--endpoints", type=int, default=256, help="API endpoints")
+    p.add_argument("--components", type=int, default=64)
+    p.add_argument("--seq-len", type=int, default=60)
+    p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--device", type=str, default=None)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    dist_ctx = init_distributed()
+    rank = dist_ctx.rank if dist_ctx else 0
+    world = dist_ctx.world_size if dist_ctx else 1
+
+    on_gpu = torch.cuda.is_available()
+    if args.device:
+        device = torch.device(args.device)
+    elif on_gpu:
+        device = torch.device("cuda", dist_ctx.local_rank if dist_ctx else 0)
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+
+    # ---- synthetic 256-endpoint app (identical on every rank: same seed) ----
+    T = args.seq_len
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=args.endpoints,
+        n_components=args.components,
+        windows_per_day=max(4 * T, 240),
+        n_days=1,
+        shapes_per_api=3,
+        seed=1234,
+    ))
+    data = app.generate_featurized()
+    ds = EstimationDataset(data, step_size=T, split_fraction=0.8)
+    spec = build_model_spec(data)
+
+    torch.manual_seed(1234)  # identical init on all ranks (weak-scaled DP)
+    model = DeepRestNet(spec, DeepRestNetConfig(dropout=0.0)).to(device)
+    opt = FusedAdam(model.parameters(), lr=1e-3)
+
+    X = ds.X_train.to(device)
+    y = ds.y_train.to(device)
+    n = X.shape[0]
+    B = args.batch
+    autocast_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    use_autocast = on_gpu and args.dtype == "bf16"
+
+    def step(i: int):
+        s = (i * B) % max(n - B, 1)
+        xb, yb = X[s : s + B], y[s : s + B]
+        with torch.autocast(device_type="cuda", dtype=autocast_dtype,
+                            enabled=use_autocast):
+            out = model(xb)
+            loss = model.loss(out.float(), yb)
+        opt.zero_grad(set_to_none=True)
+        loss.backward()
+        if dist_ctx is not None:
+            dist_ctx.all_reduce_gradients(model)
+        opt.step()
+        return loss
+
+    # ---- warmup ----
+    for i in range(args.warmup):
+        step(i)
+    if on_gpu:
+        torch.cuda.synchronize()
+    if dist_ctx is not None:
+        dist_ctx.barrier()
+
+    # ---- timed region: exactly K steps ----
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    if on_gpu:
+        torch.cuda.synchronize()
+    if dist_ctx is not None:
+        dist_ctx.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks (slowest rank defines the job)
+    if dist_ctx is not None:
+        elapsed = dist_ctx.all_reduce_scalar(elapsed, op="max")
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    samples_per_sec = (B * world * args.steps) / elapsed
+
+    if rank == 0:
+        n_gpus = world if on_gpu else args.gpus
+        print(json.dumps({
+            "metric": "training samples/sec, 256-endpoint 3-resource estimation model",
+            "value": round(samples_per_sec, 2),
+            "unit": "windows/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype if on_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "deeprest-amd (attention encoder + call-graph + GRU decoders)",
+                "endpoints": args.endpoints,
+                "components": args.components,
+                "num_metrics": spec.num_metrics,
+                "num_paths": spec.num_paths,
+                "global_batch": B * world,
+                "seq_len": T,
+                "parallelism": f"dp{world}",
+            },
+        }))
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

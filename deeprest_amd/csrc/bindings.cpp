@@ -1,0 +1,232 @@
+// Python bindings for the deeprest_amd CDNA4 kernels (torch extension).
+//
+// Tensor checking / allocation / stream plumbing lives here; the kernels in
+// *.hip are torch-free and exposed through a C ABI.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <c10/cuda/CUDAGuard.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+// ---- C ABI launchers from the .hip translation units ----
+extern "C" {
+void dr_layernorm_fwd(const void* x, const float* w, const float* b, void* y,
+                      float* mean, float* rstd, int64_t n_rows, int D, float eps,
+                      int is_bf16, hipStream_t stream);
+void dr_layernorm_bwd(const void* dy, const void* x, const float* w,
+                      const float* mean, const float* rstd, void* dx,
+                      float* dwdb_part, int n_blocks, int64_t n_rows, int D,
+                      int is_bf16, hipStream_t stream);
+void dr_pinball_fwd(const float* out, const float* labels, const float* quantiles,
+                    int Q, int64_t N, float inv_count, float* loss,
+                    hipStream_t stream);
+void dr_pinball_bwd(const float* out, const float* labels, const float* quantiles,
+                    int Q, int64_t N, float gscale, float* dout, hipStream_t stream);
+void dr_fused_adam(const int64_t* meta, int nt, int64_t total, float lr, float beta1,
+                   float beta2, float eps, float weight_decay, int step,
+                   hipStream_t stream);
+void dr_gru_fwd(const void* xg, const void* gamma, const void* beta,
+                const void* w_hh, const float* b_hh, const void* h0, void* h_all,
+                void* saves, int B, int TT, int C, int reverse, int save,
+                int is_bf16, hipStream_t stream);
+void dr_gru_bwd(const void* grad_h, const void* w_hh, const void* h0,
+                const void* h_all, const void* saves, void* dpre_x, float* dh0,
+                int B, int TT, int C, int reverse, int is_bf16, hipStream_t stream);
+void dr_mha_fwd(const void* q, const void* k, const void* v, void* o, float* lse,
+                int64_t BH, int T_len, int D, float scale, int is_bf16,
+                hipStream_t stream);
+}
+
+namespace {
+
+bool is_bf16(const at::Tensor& t) { return t.scalar_type() == at::kBFloat16; }
+
+void check_dtype(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16 || t.scalar_type() == at::kFloat,
+              name, " must be bf16 or f32, got ", t.scalar_type());
+}
+
+hipStream_t cur_stream() {
+  return at::cuda::getCurrentCUDAStream().stream();
+}
+
+// ------------------------------------------------------------- layer norm
+std::vector<at::Tensor> layer_norm_forward(at::Tensor x, at::Tensor w, at::Tensor b,
+                                           double eps) {
+  const at::cuda::CUDAGuard guard(x.device());
+  check_dtype(x, "x");
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && b.is_contiguous());
+  TORCH_CHECK(w.scalar_type() == at::kFloat && b.scalar_type() == at::kFloat,
+              "layer_norm weight/bias must be f32");
+  int D = (int)x.size(-1);
+  int64_t n_rows = x.numel() / D;
+  TORCH_CHECK(w.numel() == D && b.numel() == D);
+  auto y = at::empty_like(x);
+  auto mean = at::empty({n_rows}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({n_rows}, x.options().dtype(at::kFloat));
+  dr_layernorm_fwd(x.data_ptr(), w.data_ptr<float>(), b.data_ptr<float>(),
+                   y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                   n_rows, D, (float)eps, is_bf16(x), cur_stream());
+  return {y, mean, rstd};
+}
+
+std::vector<at::Tensor> layer_norm_backward(at::Tensor dy, at::Tensor x, at::Tensor w,
+                                            at::Tensor mean, at::Tensor rstd) {
+  const at::cuda::CUDAGuard guard(x.device());
+  TORCH_CHECK(dy.is_contiguous() && x.is_contiguous());
+  TORCH_CHECK(dy.scalar_type() == x.scalar_type());
+  int D = (int)x.size(-1);
+  int64_t n_rows = x.numel() / D;
+  int n_blocks = (int)std::min<int64_t>((n_rows + 3) / 4, 512);
+  if (n_blocks == 0) n_blocks = 1;
+  auto dx = at::empty_like(x);
+  auto part = at::zeros({n_blocks, 2, D}, x.options().dtype(at::kFloat));
+  dr_layernorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr<float>(),
+                   mean.data_ptr<float>(), rstd.data_ptr<float>(), dx.data_ptr(),
+                   part.data_ptr<float>(), n_blocks, n_rows, D, is_bf16(x),
+                   cur_stream());
+  auto sums = part.sum(0);  // (2, D)
+  return {dx, sums[0], sums[1]};
+}
+
+// ---------------------------------------------------------------- pinball
+at::Tensor pinball_forward(at::Tensor outputs, at::Tensor labels, at::Tensor q) {
+  const at::cuda::CUDAGuard guard(outputs.device());
+  TORCH_CHECK(outputs.scalar_type() == at::kFloat && labels.scalar_type() == at::kFloat,
+              "pinball expects f32");
+  TORCH_CHECK(outputs.is_contiguous() && labels.is_contiguous());
+  int Q = (int)outputs.size(-1);
+  TORCH_CHECK(Q <= 8, "at most 8 quantiles");
+  int64_t N = outputs.numel() / Q;
+  TORCH_CHECK(labels.numel() == N);
+  auto loss = at::zeros({}, outputs.options());
+  float inv_count = N > 0 ? 1.0f / (float)N : 0.f;
+  dr_pinball_fwd(outputs.data_ptr<float>(), labels.data_ptr<float>(),
+                 q.data_ptr<float>(), Q, N, inv_count, loss.data_ptr<float>(),
+                 cur_stream());
+  return loss;
+}
+
+at::Tensor pinball_backward(at::Tensor grad, at::Tensor outputs, at::Tensor labels,
+                            at::Tensor q) {
+  const at::cuda::CUDAGuard guard(outputs.device());
+  int Q = (int)outputs.size(-1);
+  int64_t N = outputs.numel() / Q;
+  auto dout = at::empty_like(outputs);
+  float g = grad.item<float>();
+  float gscale = N > 0 ? g / (float)N : 0.f;
+  dr_pinball_bwd(outputs.data_ptr<float>(), labels.data_ptr<float>(),
+                 q.data_ptr<float>(), Q, N, gscale, dout.data_ptr<float>(),
+                 cur_stream());
+  return dout;
+}
+
+// ------------------------------------------------------------- fused adam
+void fused_adam(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+                std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+                int64_t step, double lr, double beta1, double beta2, double eps,
+                double weight_decay) {
+  TORCH_CHECK(!params.empty());
+  const at::cuda::CUDAGuard guard(params[0].device());
+  int nt = (int)params.size();
+  std::vector<int64_t> meta(5 * nt);
+  int64_t total = 0;
+  for (int i = 0; i < nt; ++i) {
+    TORCH_CHECK(params[i].scalar_type() == at::kFloat, "fused_adam expects f32 params");
+    TORCH_CHECK(params[i].is_contiguous() && grads[i].is_contiguous());
+    meta[i] = (int64_t)params[i].data_ptr();
+    meta[nt + i] = (int64_t)grads[i].data_ptr();
+    meta[2 * nt + i] = (int64_t)ms[i].data_ptr();
+    meta[3 * nt + i] = (int64_t)vs[i].data_ptr();
+    total += params[i].numel();
+    meta[4 * nt + i] = total;  // cumulative end offsets
+  }
+  auto meta_t = at::from_blob(meta.data(), {(int64_t)meta.size()},
+                              at::TensorOptions().dtype(at::kLong))
+                    .to(params[0].device(), /*non_blocking=*/false);
+  dr_fused_adam(meta_t.data_ptr<int64_t>(), nt, total, (float)lr, (float)beta1,
+                (float)beta2, (float)eps, (float)weight_decay, (int)step,
+                cur_stream());
+}
+
+// -------------------------------------------------------------------- gru
+std::vector<at::Tensor> gru_seq_forward(at::Tensor xg, at::Tensor w_hh,
+                                        at::Tensor b_hh, at::Tensor h0,
+                                        at::Tensor gamma, at::Tensor beta,
+                                        bool reverse, bool save) {
+  const at::cuda::CUDAGuard guard(xg.device());
+  check_dtype(xg, "x_gates");
+  TORCH_CHECK(xg.dim() == 3, "x_gates must be (B, T, 3H)");
+  TORCH_CHECK(h0.dim() == 3, "h0 must be (B, C, H)");
+  int B = (int)xg.size(0), TT = (int)xg.size(1);
+  int C = (int)h0.size(1), H = (int)h0.size(2);
+  TORCH_CHECK(H == 128, "fused GRU kernel requires hidden size 128, got ", H);
+  TORCH_CHECK(C >= 3, "fused GRU kernel requires >= 3 components, got ", C);
+  TORCH_CHECK(xg.size(2) == 3 * H && w_hh.size(0) == 3 * H && w_hh.size(1) == H);
+  TORCH_CHECK(gamma.sizes() == at::IntArrayRef({C, 3 * H}));
+  TORCH_CHECK(b_hh.scalar_type() == at::kFloat, "b_hh must be f32");
+  auto dt = xg.scalar_type();
+  TORCH_CHECK(w_hh.scalar_type() == dt && h0.scalar_type() == dt &&
+                  gamma.scalar_type() == dt && beta.scalar_type() == dt,
+              "gru operand dtypes must match x_gates");
+  TORCH_CHECK(xg.is_contiguous() && w_hh.is_contiguous() && h0.is_contiguous() &&
+              gamma.is_contiguous() && beta.is_contiguous() && b_hh.is_contiguous());
+
+  auto h_all = at::empty({B, TT, C, H}, xg.options());
+  auto saves = save ? at::empty({B, TT, C, 4 * H}, xg.options())
+                    : at::empty({0}, xg.options());
+  dr_gru_fwd(xg.data_ptr(), gamma.data_ptr(), beta.data_ptr(), w_hh.data_ptr(),
+             b_hh.data_ptr<float>(), h0.data_ptr(), h_all.data_ptr(),
+             save ? saves.data_ptr() : nullptr, B, TT, C, reverse ? 1 : 0,
+             save ? 1 : 0, dt == at::kBFloat16, cur_stream());
+  return {h_all, saves};
+}
+
+std::vector<at::Tensor> gru_seq_backward_kernel(at::Tensor grad_h, at::Tensor w_hh,
+                                                at::Tensor h0, at::Tensor h_all,
+                                                at::Tensor saves, bool reverse) {
+  const at::cuda::CUDAGuard guard(grad_h.device());
+  int B = (int)grad_h.size(0), TT = (int)grad_h.size(1);
+  int C = (int)grad_h.size(2), H = (int)grad_h.size(3);
+  TORCH_CHECK(H == 128);
+  auto dt = grad_h.scalar_type();
+  TORCH_CHECK(grad_h.is_contiguous() && h_all.is_contiguous() && saves.is_contiguous());
+  auto dpre_x = at::empty({B, TT, C, 3 * H}, grad_h.options());
+  auto dh0 = at::empty({B, C, H}, grad_h.options().dtype(at::kFloat));
+  dr_gru_bwd(grad_h.data_ptr(), w_hh.data_ptr(), h0.data_ptr(), h_all.data_ptr(),
+             saves.data_ptr(), dpre_x.data_ptr(), dh0.data_ptr<float>(), B, TT, C,
+             reverse ? 1 : 0, dt == at::kBFloat16, cur_stream());
+  return {dpre_x, dh0};
+}
+
+// -------------------------------------------------------------------- mha
+std::vector<at::Tensor> mha_forward(at::Tensor q, at::Tensor k, at::Tensor v,
+                                    double scale) {
+  const at::cuda::CUDAGuard guard(q.device());
+  check_dtype(q, "q");
+  TORCH_CHECK(q.dim() == 4, "q must be (B, H, T, D)");
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  int64_t B = q.size(0), NH = q.size(1);
+  int T_len = (int)q.size(2), D = (int)q.size(3);
+  TORCH_CHECK(D >= 8 && D <= 64 && D % 8 == 0, "head dim must be 8..64, mult of 8");
+  auto o = at::empty_like(q);
+  auto lse = at::empty({B, NH, T_len}, q.options().dtype(at::kFloat));
+  dr_mha_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+             lse.data_ptr<float>(), B * NH, T_len, D, (float)scale,
+             is_bf16(q), cur_stream());
+  return {o, lse};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("layer_norm_forward", &layer_norm_forward);
+  m.def("layer_norm_backward", &layer_norm_backward);
+  m.def("pinball_forward", &pinball_forward);
+  m.def("pinball_backward", &pinball_backward);
+  m.def("fused_adam", &fused_adam);
+  m.def("gru_seq_forward", &gru_seq_forward);
+  m.def("gru_seq_backward_kernel", &gru_seq_backward_kernel);
+  m.def("mha_forward", &mha_forward);
+}

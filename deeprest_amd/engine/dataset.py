@@ -1,0 +1,103 @@
+"""Windowed train/test dataset with checkpointable normalization.
+
+Reproduces the reference's data preparation exactly
+(reference: resource-estimation/estimate.py:22-57):
+
+- stride-1 sliding windows of length ``step_size`` over the traffic matrix
+  and each resource series;
+- ``split = int(len(X) * split_fraction)`` train/test boundary;
+- global min-max normalization of X and per-metric min-max of y, fit on the
+  train split only (qrnn.py:69-75), with the fitted (scale, min) pairs kept
+  as state so inference is reproducible from a checkpoint (the reference
+  keeps them only in process memory).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..data.featurize import FeaturizedData
+from ..data.windows import MinMaxScaler, sliding_window
+
+
+class EstimationDataset:
+    def __init__(
+        self,
+        data: FeaturizedData,
+        step_size: int = 60,
+        split_fraction: float = 0.40,
+    ) -> None:
+        self.data = data
+        self.step_size = step_size
+        self.metric_names = data.metric_names
+
+        traffic = np.asarray(data.traffic, dtype=np.float64)
+        y_flat = np.stack([data.resources[n] for n in self.metric_names], axis=-1)
+
+        X = sliding_window(traffic, step_size)                 # (N, T, P)
+        y = sliding_window(y_flat, step_size)                  # (N, T, M)
+        self.num_windows = len(X)
+        self.split = int(self.num_windows * split_fraction)
+        if self.split <= 0 or self.split >= self.num_windows:
+            raise ValueError(
+                f"split {self.split} out of range for {self.num_windows} windows"
+            )
+
+        # raw (denormalized) labels, needed by the baselines and error eval
+        self.y_raw = y.copy()
+
+        self.x_scaler = MinMaxScaler().fit(X, self.split)
+        X = self.x_scaler.transform(X)
+        self.y_scalers: List[MinMaxScaler] = []
+        for idx in range(y.shape[-1]):
+            sc = MinMaxScaler().fit(y[:, :, idx], self.split)
+            y[:, :, idx] = sc.transform(y[:, :, idx])
+            self.y_scalers.append(sc)
+
+        self.X = torch.from_numpy(np.ascontiguousarray(X, dtype=np.float32))
+        self.y = torch.from_numpy(np.ascontiguousarray(y, dtype=np.float32))
+
+    # ------------------------------------------------------------------ views
+    @property
+    def X_train(self) -> torch.Tensor:
+        return self.X[: self.split]
+
+    @property
+    def y_train(self) -> torch.Tensor:
+        return self.y[: self.split]
+
+    @property
+    def X_test(self) -> torch.Tensor:
+        return self.X[self.split :]
+
+    @property
+    def y_test(self) -> torch.Tensor:
+        return self.y[self.split :]
+
+    def eval_window_indices(self, max_cycles: int = 9) -> List[int]:
+        """Non-overlapping test windows: iv % step == 0, up to max_cycles
+        (reference: estimate.py:85-88)."""
+        out = []
+        n_test = self.num_windows - self.split
+        for iv in range(n_test):
+            if iv % self.step_size == 0:
+                out.append(iv)
+                if len(out) >= max_cycles:
+                    break
+        return out
+
+    def denormalize_metric(self, values: np.ndarray, idx: int) -> np.ndarray:
+        return self.y_scalers[idx].inverse_transform(values)
+
+    # ------------------------------------------------------------- checkpoint
+    def scaler_state(self) -> dict:
+        return {
+            "x_scaler": self.x_scaler.state_dict(),
+            "y_scalers": [s.state_dict() for s in self.y_scalers],
+            "metric_names": self.metric_names,
+            "step_size": self.step_size,
+            "split": self.split,
+        }

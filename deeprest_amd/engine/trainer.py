@@ -1,0 +1,227 @@
+"""Training loop with the three-estimator comparison harness.
+
+Mirrors the reference driver's behavior (reference:
+resource-estimation/estimate.py:21-123): run both baselines up front, train
+the estimator with quantile loss + Adam, and per evaluation report the
+Median/95th/99th/Max absolute-error table for RESRC, COMP and DEEPR on the
+non-overlapping held-out windows, denormalizing the model's median quantile.
+
+Additions over the reference: checkpoint/resume, bf16 autocast on GPU,
+fused-Adam optimizer, optional data parallelism (one process per GPU over
+RCCL; gradients leave through the DistContext hook so the loop itself stays
+backend-agnostic and testable with gloo on CPU).
+"""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+
+from ..data.featurize import FeaturizedData
+from ..models.baselines import ComponentAwareBaseline, ResourceAwareBaseline
+from ..models.net import DeepRestNet, DeepRestNetConfig, build_model_spec
+from ..ops.adam import FusedAdam
+from ..utils.errors import error_percentiles, format_error_table
+from .checkpoint import load_checkpoint, save_checkpoint
+from .config import EngineConfig
+from .dataset import EstimationDataset
+
+
+@dataclass
+class TrainResult:
+    train_losses: List[float] = field(default_factory=list)
+    test_losses: List[float] = field(default_factory=list)
+    error_tables: Dict[str, Dict[str, Dict[str, float]]] = field(default_factory=dict)
+    samples_per_sec: float = 0.0
+
+    def summary(self) -> str:
+        lines = []
+        for name, per_est in self.error_tables.items():
+            lines.append(format_error_table(name, per_est))
+        return "\n".join(lines)
+
+
+class Trainer:
+    def __init__(
+        self,
+        data: FeaturizedData,
+        config: Optional[EngineConfig] = None,
+        device: Optional[torch.device] = None,
+        dist_ctx=None,
+        model: Optional[DeepRestNet] = None,
+    ) -> None:
+        self.cfg = config or EngineConfig()
+        self.device = device or torch.device(
+            "cuda" if torch.cuda.is_available() else "cpu"
+        )
+        self.dist = dist_ctx
+        self.rank = dist_ctx.rank if dist_ctx is not None else 0
+        self.world_size = dist_ctx.world_size if dist_ctx is not None else 1
+
+        self.dataset = EstimationDataset(
+            data, step_size=self.cfg.data.step_size, split_fraction=self.cfg.data.split
+        )
+        if model is None:
+            spec = build_model_spec(data)
+            model = DeepRestNet(spec, self.cfg.model)
+        self.model = model.to(self.device)
+        self.optimizer = FusedAdam(self.model.parameters(), lr=self.cfg.train.lr)
+        self.feature_space_state = (
+            data.feature_space.state_dict() if data.feature_space is not None else None
+        )
+        self.start_epoch = 0
+        self._baseline_preds: Optional[Dict[str, np.ndarray]] = None
+        self.autocast_dtype = (
+            torch.bfloat16 if self.cfg.train.dtype == "bf16" else torch.float32
+        )
+
+    # ------------------------------------------------------------- baselines
+    def run_baselines(self) -> Dict[str, np.ndarray]:
+        """Fit both baselines per metric; predictions for ALL test windows."""
+        if self._baseline_preds is not None:
+            return self._baseline_preds
+        ds = self.dataset
+        t_cfg = self.cfg.train
+        resrc_list, comp_list = [], []
+        for idx, name in enumerate(ds.metric_names):
+            yw = ds.y_raw[:, :, idx]
+            resrc = ResourceAwareBaseline(
+                split=ds.split, window=ds.step_size,
+                epochs=t_cfg.baseline_epochs, seed=t_cfg.seed,
+            ).fit_and_estimate(yw)
+            comp_name = self.model.spec.components[self.model.spec.comp_of[idx]]
+            comp = ComponentAwareBaseline(
+                component=comp_name, invocations=ds.data.invocations,
+                window=ds.step_size, split=ds.split,
+            ).fit_and_estimate(yw)
+            resrc_list.append(resrc[:, :, None])
+            comp_list.append(comp[:, :, None])
+        self._baseline_preds = {
+            "resrc": np.concatenate(resrc_list, axis=-1),   # (N_test, T, M)
+            "comp": np.concatenate(comp_list, axis=-1),
+        }
+        return self._baseline_preds
+
+    # ------------------------------------------------------------------ train
+    def train(self) -> TrainResult:
+        cfg = self.cfg.train
+        ds = self.dataset
+        result = TrainResult()
+
+        baselines = self.run_baselines() if cfg.run_baselines and self.rank == 0 else None
+
+        if cfg.resume and cfg.checkpoint_path:
+            try:
+                self.load(cfg.checkpoint_path)
+            except FileNotFoundError:
+                pass
+
+        X_train = ds.X_train.to(self.device)
+        y_train = ds.y_train.to(self.device)
+        n = X_train.shape[0]
+        gen = torch.Generator().manual_seed(cfg.seed)
+
+        total_samples = 0
+        t_start = time.perf_counter()
+        for epoch in range(self.start_epoch, cfg.epochs):
+            self.model.train()
+            perm = torch.randperm(n, generator=gen)
+            # data-parallel shard: contiguous slice of the permutation per rank
+            if self.world_size > 1:
+                perm = perm[self.rank :: self.world_size]
+            losses = []
+            for s in range(0, len(perm), cfg.batch_size):
+                idx = perm[s : s + cfg.batch_size].to(self.device)
+                xb, yb = X_train[idx], y_train[idx]
+                with torch.autocast(
+                    device_type=self.device.type, dtype=self.autocast_dtype,
+                    enabled=(self.device.type == "cuda"),
+                ):
+                    out = self.model(xb)
+                    loss = self.model.loss(out.float(), yb)
+                self.optimizer.zero_grad(set_to_none=True)
+                loss.backward()
+                if self.dist is not None:
+                    self.dist.all_reduce_gradients(self.model)
+                self.optimizer.step()
+                losses.append(loss.item())
+                total_samples += xb.shape[0] * self.world_size
+            result.train_losses.append(float(np.mean(losses)))
+
+            if self.rank == 0:
+                test_loss, tables = self.evaluate(baselines)
+                result.test_losses.append(test_loss)
+                result.error_tables = tables
+                if cfg.log_every and (epoch + 1) % cfg.log_every == 0:
+                    print(
+                        f"Epoch [{epoch + 1}/{cfg.epochs}], "
+                        f"Train Loss: {result.train_losses[-1]:.6f}, "
+                        f"Test Loss: {test_loss:.6f}"
+                    )
+                    for name, per_est in tables.items():
+                        print(format_error_table(name, per_est))
+                if cfg.checkpoint_path:
+                    self.save(cfg.checkpoint_path, epoch + 1)
+
+        elapsed = time.perf_counter() - t_start
+        result.samples_per_sec = total_samples / max(elapsed, 1e-9)
+        return result
+
+    # ------------------------------------------------------------------- eval
+    @torch.no_grad()
+    def evaluate(self, baselines: Optional[Dict[str, np.ndarray]] = None):
+        """Test loss + per-metric error tables on non-overlapping windows."""
+        cfg = self.cfg.train
+        ds = self.dataset
+        self.model.eval()
+        eval_idx = ds.eval_window_indices(cfg.eval_cycles)
+        if not eval_idx:
+            return float("nan"), {}
+        xb = ds.X_test[eval_idx].to(self.device)
+        yb = ds.y_test[eval_idx].to(self.device)
+        with torch.autocast(
+            device_type=self.device.type, dtype=self.autocast_dtype,
+            enabled=(self.device.type == "cuda"),
+        ):
+            out = self.model(xb)
+        out = out.float()
+        test_loss = float(self.model.loss(out, yb).item())
+
+        outputs = np.maximum(out.cpu().numpy(), 1e-6)      # (K, T, M, Q)
+        labels = yb.cpu().numpy()
+        tables: Dict[str, Dict[str, Dict[str, float]]] = {}
+        median_q = len(self.model.cfg.quantiles) // 2
+        for m, name in enumerate(ds.metric_names):
+            labels_d = ds.denormalize_metric(labels[:, :, m], m).ravel()
+            pred_d = ds.denormalize_metric(outputs[:, :, m, median_q], m).ravel()
+            per_est = {}
+            if baselines is not None:
+                for est_name, key in (("resrc", "resrc"), ("comp", "comp")):
+                    bl = baselines[key][eval_idx][:, :, m].ravel()
+                    per_est[est_name] = error_percentiles(np.abs(bl - labels_d))
+            per_est["deepr"] = error_percentiles(np.abs(pred_d - labels_d))
+            tables[name] = per_est
+        return test_loss, tables
+
+    # ------------------------------------------------------------ checkpoints
+    def save(self, path: str, epoch: int) -> None:
+        save_checkpoint(
+            path,
+            self.model,
+            optimizer=self.optimizer,
+            scaler_state=self.dataset.scaler_state(),
+            feature_space_state=self.feature_space_state,
+            epoch=epoch,
+            extra={"config": self.cfg.to_dict()},
+        )
+
+    def load(self, path: str) -> None:
+        state = load_checkpoint(path, map_location=self.device)
+        self.model.load_state_dict(state["model"]["state_dict"])
+        if state.get("optimizer"):
+            self.optimizer.load_state_dict(state["optimizer"])
+        self.start_epoch = int(state.get("epoch", 0))

@@ -74,22 +74,50 @@ class _FusedGRUSequence(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x_gates, w_hh, b_hh, h0, gamma, beta, reverse):
         ext = require_native("fused_gru_sequence")
-        h_all, gates_saved = ext.gru_seq_forward(
-            x_gates, w_hh, b_hh, h0, gamma, beta, bool(reverse)
+        need_grad = any(
+            t.requires_grad for t in (x_gates, w_hh, b_hh, h0, gamma, beta)
+        ) and torch.is_grad_enabled()
+        h_all, saves = ext.gru_seq_forward(
+            x_gates, w_hh, b_hh, h0, gamma, beta, bool(reverse), bool(need_grad)
         )
-        ctx.save_for_backward(x_gates, w_hh, b_hh, h0, gamma, beta, h_all, gates_saved)
+        ctx.save_for_backward(x_gates, w_hh, h0, gamma, h_all, saves)
         ctx.reverse = bool(reverse)
         return h_all
 
     @staticmethod
     def backward(ctx, grad_h_all):
         ext = require_native("fused_gru_sequence")
-        x_gates, w_hh, b_hh, h0, gamma, beta, h_all, gates_saved = ctx.saved_tensors
-        dx_gates, dw_hh, db_hh, dh0, dgamma, dbeta = ext.gru_seq_backward(
-            grad_h_all.contiguous(), x_gates, w_hh, b_hh, h0, gamma, beta,
-            h_all, gates_saved, ctx.reverse,
+        x_gates, w_hh, h0, gamma, h_all, saves = ctx.saved_tensors
+        reverse = ctx.reverse
+        # sequential chain (custom kernel): gate pre-activation grads + dh0
+        dpre_x, dh0 = ext.gru_seq_backward_kernel(
+            grad_h_all.contiguous(), w_hh, h0, h_all, saves, reverse
         )
-        return dx_gates, dw_hh, db_hh, dh0, dgamma, dbeta, None
+        B, T, C, G = dpre_x.shape
+        H = G // 3
+        # batched reductions: plain GEMMs / f32-accumulated sums (rocBLAS/eager)
+        r_saved = saves[..., :H]
+        d_hhn = dpre_x[..., 2 * H :] * r_saved
+        dpre_w = torch.cat([dpre_x[..., : 2 * H], d_hhn], dim=-1)
+        if reverse:
+            h_prev = torch.cat([h_all[:, 1:], h0.unsqueeze(1)], dim=1)
+        else:
+            h_prev = torch.cat([h0.unsqueeze(1), h_all[:, :-1]], dim=1)
+        M = B * T * C
+        dw_hh = dpre_w.reshape(M, G).t() @ h_prev.reshape(M, H)          # (3H, H)
+        db_hh = dpre_w.sum(dim=(0, 1, 2), dtype=torch.float32)
+        dxg = torch.einsum("btcj,cj->btj", dpre_x, gamma)
+        dgamma = torch.einsum("btcj,btj->cj", dpre_x, x_gates)
+        dbeta = dpre_x.sum(dim=(0, 1)).to(gamma.dtype)
+        return (
+            dxg.to(x_gates.dtype),
+            dw_hh.to(w_hh.dtype),
+            db_hh,
+            dh0.to(h0.dtype),
+            dgamma.to(gamma.dtype),
+            dbeta,
+            None,
+        )
 
 
 def fused_gru_sequence(
@@ -102,15 +130,23 @@ def fused_gru_sequence(
     reverse: bool = False,
 ) -> torch.Tensor:
     if x_gates.is_cuda:
+        dt = x_gates.dtype
         if gamma is None:
             C = h0.shape[1]
             G = x_gates.shape[-1]
-            gamma = torch.ones(C, G, device=x_gates.device, dtype=x_gates.dtype)
-            beta = torch.zeros(C, G, device=x_gates.device, dtype=x_gates.dtype)
+            gamma = torch.ones(C, G, device=x_gates.device, dtype=dt)
+            beta = torch.zeros(C, G, device=x_gates.device, dtype=dt)
         elif beta is None:
             beta = torch.zeros_like(gamma)
+        # dtype harmonization outside the Function so the casts are
+        # autograd-tracked back to the fp32 master parameters
         return _FusedGRUSequence.apply(
-            x_gates.contiguous(), w_hh.contiguous(), b_hh.contiguous(),
-            h0.contiguous(), gamma.contiguous(), beta.contiguous(), reverse,
+            x_gates.contiguous(),
+            w_hh.to(dt).contiguous(),
+            b_hh.float().contiguous(),
+            h0.to(dt).contiguous(),
+            gamma.to(dt).contiguous(),
+            beta.to(dt).contiguous(),
+            reverse,
         )
     return reference_gru_sequence(x_gates, w_hh, b_hh, h0, gamma, beta, reverse)

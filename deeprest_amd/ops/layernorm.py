@@ -38,5 +38,7 @@ class _LayerNorm(torch.autograd.Function):
 def layer_norm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
                eps: float = 1e-5) -> torch.Tensor:
     if x.is_cuda:
-        return _LayerNorm.apply(x.contiguous(), weight, bias, eps)
+        return _LayerNorm.apply(
+            x.contiguous(), weight.float().contiguous(), bias.float().contiguous(), eps
+        )
     return reference_layer_norm(x, weight, bias, eps)

@@ -21,6 +21,8 @@ def load_native():
         return _EXT
     _TRIED = True
     try:
+        import torch  # noqa: F401  (the .so links against torch's libs)
+
         _EXT = importlib.import_module("deeprest_amd._C")
     except ImportError:
         _EXT = None

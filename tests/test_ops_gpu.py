@@ -1,0 +1,246 @@
+"""GPU numerics tests: each HIP kernel vs the plain PyTorch fp32 reference.
+
+Tolerance tiers: bf16 internal compute (GRU/MHA GEMM operands, guide section 4)
+is compared against the fp32 oracle at bf16-appropriate tolerances; fp32-only
+kernels (pinball, adam, layernorm stats) at fp32 tolerances.
+"""
+
+import numpy as np
+import pytest
+import torch
+
+gpu = pytest.mark.gpu
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dev():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from deeprest_amd.ops import native_available
+
+    if not native_available():
+        pytest.fail("HIP extension not built — GPU tests must not fall back")
+    return torch.device("cuda:0")
+
+
+# ------------------------------------------------------------------ layernorm
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("shape", [(8, 16, 256), (3, 7, 192), (1, 1, 64), (5, 1024)])
+def test_layernorm_fwd_bwd(dev, dtype, shape):
+    from deeprest_amd.ops import layer_norm, reference_layer_norm
+
+    torch.manual_seed(0)
+    x = torch.randn(*shape, device=dev, dtype=dtype) * 3 + 1
+    w = torch.randn(shape[-1], device=dev) * 0.5 + 1.0
+    b = torch.randn(shape[-1], device=dev) * 0.1
+    x_ref = x.detach().float().requires_grad_(True)
+    w_ref = w.detach().clone().requires_grad_(True)
+    b_ref = b.detach().clone().requires_grad_(True)
+    x_t = x.detach().clone().requires_grad_(True)
+    w_t = w.detach().clone().requires_grad_(True)
+    b_t = b.detach().clone().requires_grad_(True)
+
+    y = layer_norm(x_t, w_t, b_t)
+    y_ref = reference_layer_norm(x_ref, w_ref, b_ref)
+    tol = dict(rtol=2e-2, atol=2e-2) if dtype == torch.bfloat16 else dict(rtol=2e-5, atol=2e-5)
+    torch.testing.assert_close(y.float(), y_ref, **tol)
+
+    g = torch.randn_like(y_ref)
+    y.backward(g.to(dtype))
+    y_ref.backward(g)
+    torch.testing.assert_close(x_t.grad.float(), x_ref.grad, rtol=5e-2 if dtype == torch.bfloat16 else 1e-4,
+                               atol=5e-2 if dtype == torch.bfloat16 else 1e-4)
+    torch.testing.assert_close(w_t.grad, w_ref.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(b_t.grad, b_ref.grad, rtol=5e-2, atol=5e-2)
+
+
+# -------------------------------------------------------------------- pinball
+def test_pinball_fwd_bwd(dev):
+    from deeprest_amd.ops import pinball_loss, reference_pinball_loss
+
+    torch.manual_seed(1)
+    B, T, M, Q = 4, 12, 9, 3
+    quantiles = (0.05, 0.50, 0.95)
+    out = torch.randn(B, T, M, Q, device=dev)
+    lab = torch.randn(B, T, M, device=dev)
+    out_t = out.detach().clone().requires_grad_(True)
+    out_r = out.detach().clone().requires_grad_(True)
+
+    l_t = pinball_loss(out_t, lab, quantiles)
+    l_r = reference_pinball_loss(out_r, lab, quantiles)
+    torch.testing.assert_close(l_t, l_r, rtol=1e-5, atol=1e-6)
+    l_t.backward()
+    l_r.backward()
+    torch.testing.assert_close(out_t.grad, out_r.grad, rtol=1e-5, atol=1e-7)
+
+
+# ----------------------------------------------------------------- fused adam
+def test_fused_adam_matches_torch(dev):
+    from deeprest_amd.ops.adam import FusedAdam
+
+    torch.manual_seed(2)
+    shapes = [(64, 32), (128,), (7, 5, 3), (1000,)]
+    params_a = [torch.randn(*s, device=dev, requires_grad=True) for s in shapes]
+    params_b = [p.detach().clone().requires_grad_(True) for p in params_a]
+    opt_a = FusedAdam(params_a, lr=1e-2, weight_decay=0.01)
+    opt_b = torch.optim.Adam(params_b, lr=1e-2, weight_decay=0.01)
+    for step in range(5):
+        for pa, pb in zip(params_a, params_b):
+            g = torch.randn_like(pa)
+            pa.grad = g.clone()
+            pb.grad = g.clone()
+        opt_a.step()
+        opt_b.step()
+    for pa, pb in zip(params_a, params_b):
+        torch.testing.assert_close(pa, pb, rtol=1e-4, atol=1e-5)
+
+
+# ------------------------------------------------------------------------ gru
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("reverse", [False, True])
+def test_gru_forward_matches_reference(dev, dtype, reverse):
+    from deeprest_amd.ops import fused_gru_sequence, reference_gru_sequence
+
+    torch.manual_seed(3)
+    B, T, C, H = 3, 7, 5, 128
+    xg = torch.randn(B, T, 3 * H, device=dev, dtype=dtype) * 0.5
+    w_hh = (torch.randn(3 * H, H, device=dev) * (1.0 / np.sqrt(H))).to(dtype)
+    b_hh = torch.randn(3 * H, device=dev) * 0.1
+    h0 = torch.randn(B, C, H, device=dev, dtype=dtype) * 0.3
+    gamma = (1.0 + 0.1 * torch.randn(C, 3 * H, device=dev)).to(dtype)
+    beta = (0.1 * torch.randn(C, 3 * H, device=dev)).to(dtype)
+
+    out = fused_gru_sequence(xg, w_hh, b_hh, h0, gamma, beta, reverse=reverse)
+    ref = reference_gru_sequence(
+        xg.float(), w_hh.float(), b_hh.float(), h0.float(),
+        gamma.float(), beta.float(), reverse=reverse,
+    )
+    assert out.shape == (B, T, C, H)
+    torch.testing.assert_close(out.float(), ref, rtol=5e-2, atol=3e-2)
+
+
+def test_gru_forward_large_rows_tail(dev):
+    # R not a multiple of 64 exercises the pad-row path
+    from deeprest_amd.ops import fused_gru_sequence, reference_gru_sequence
+
+    torch.manual_seed(4)
+    B, T, C, H = 5, 4, 13, 128  # R = 65
+    xg = torch.randn(B, T, 3 * H, device=dev) * 0.5
+    w_hh = torch.randn(3 * H, H, device=dev) / np.sqrt(H)
+    b_hh = torch.zeros(3 * H, device=dev)
+    h0 = torch.zeros(B, C, H, device=dev)
+    out = fused_gru_sequence(xg, w_hh, b_hh, h0)
+    ref = reference_gru_sequence(xg, w_hh, b_hh, h0)
+    torch.testing.assert_close(out.float(), ref, rtol=5e-2, atol=3e-2)
+    assert torch.isfinite(out).all()
+
+
+@pytest.mark.parametrize("reverse", [False, True])
+def test_gru_backward_matches_reference(dev, reverse):
+    from deeprest_amd.ops import fused_gru_sequence, reference_gru_sequence
+
+    torch.manual_seed(5)
+    B, T, C, H = 2, 5, 4, 128
+    mk = lambda *s: torch.randn(*s, device=dev) * 0.4
+
+    xg0, w0, b0, h00 = mk(B, T, 3 * H), mk(3 * H, H) / np.sqrt(H), mk(3 * H) * 0.2, mk(B, C, H)
+    g0 = 1.0 + 0.1 * mk(C, 3 * H)
+    be0 = 0.1 * mk(C, 3 * H)
+
+    args_t = [t.detach().clone().requires_grad_(True) for t in (xg0, w0, b0, h00, g0, be0)]
+    args_r = [t.detach().clone().requires_grad_(True) for t in (xg0, w0, b0, h00, g0, be0)]
+
+    out_t = fused_gru_sequence(*args_t, reverse=reverse)
+    out_r = reference_gru_sequence(*args_r, reverse=reverse)
+    grad = torch.randn_like(out_r)
+    out_t.backward(grad)
+    out_r.backward(grad)
+
+    names = ["x_gates", "w_hh", "b_hh", "h0", "gamma", "beta"]
+    for name, at_, ar_ in zip(names, args_t, args_r):
+        assert at_.grad is not None, name
+        torch.testing.assert_close(
+            at_.grad.float(), ar_.grad.float(), rtol=8e-2, atol=5e-2,
+            msg=lambda m, n=name: f"grad mismatch for {n}: {m}",
+        )
+
+
+# ------------------------------------------------------------------------ mha
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("T_len", [60, 64, 37, 128])
+@pytest.mark.parametrize("D", [32, 64, 16])
+def test_mha_forward_matches_reference(dev, dtype, T_len, D):
+    from deeprest_amd.ops import mha_forward, reference_mha
+
+    torch.manual_seed(6)
+    B, NH = 2, 4
+    q = torch.randn(B, NH, T_len, D, device=dev, dtype=dtype)
+    k = torch.randn(B, NH, T_len, D, device=dev, dtype=dtype)
+    v = torch.randn(B, NH, T_len, D, device=dev, dtype=dtype)
+    o = mha_forward(q, k, v)
+    o_ref = reference_mha(q.float(), k.float(), v.float())
+    torch.testing.assert_close(o.float(), o_ref, rtol=3e-2, atol=2e-2)
+
+
+def test_mha_asymmetric_catches_transpose(dev):
+    # asymmetric K/V catch row<->col swaps (guide: always test with asymmetric B)
+    from deeprest_amd.ops import mha_forward, reference_mha
+
+    B, NH, T_len, D = 1, 1, 32, 32
+    q = torch.zeros(B, NH, T_len, D, device=dev)
+    q[0, 0, :, 0] = torch.arange(T_len, device=dev) * 0.1
+    k = torch.zeros_like(q)
+    k[0, 0, :, 0] = torch.arange(T_len, device=dev) * 0.05
+    v = torch.arange(T_len * D, device=dev, dtype=torch.float32).reshape(1, 1, T_len, D) * 0.01
+    o = mha_forward(q, k, v)
+    o_ref = reference_mha(q, k, v)
+    torch.testing.assert_close(o.float(), o_ref, rtol=2e-2, atol=2e-2)
+
+
+def test_mha_backward_matches_reference(dev):
+    from deeprest_amd.ops import mha_forward, reference_mha
+
+    torch.manual_seed(7)
+    B, NH, T_len, D = 2, 3, 60, 32
+    mk = lambda: torch.randn(B, NH, T_len, D, device=dev) * 0.7
+    q0, k0, v0 = mk(), mk(), mk()
+    qt, kt, vt = (t.detach().clone().requires_grad_(True) for t in (q0, k0, v0))
+    qr, kr, vr = (t.detach().clone().requires_grad_(True) for t in (q0, k0, v0))
+    o_t = mha_forward(qt, kt, vt)
+    o_r = reference_mha(qr, kr, vr)
+    g = torch.randn_like(o_r)
+    o_t.backward(g)
+    o_r.backward(g)
+    for a, b in ((qt, qr), (kt, kr), (vt, vr)):
+        torch.testing.assert_close(a.grad, b.grad, rtol=5e-2, atol=3e-2)
+
+
+# -------------------------------------------------------------- model-on-gpu
+def test_model_step_on_gpu(dev):
+    from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+    from deeprest_amd.models.net import DeepRestNet, DeepRestNetConfig, build_model_spec
+    from deeprest_amd.ops.adam import FusedAdam
+
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=8, n_components=8, windows_per_day=120, n_days=1, seed=11))
+    data = app.generate_featurized()
+    spec = build_model_spec(data)
+    model = DeepRestNet(spec, DeepRestNetConfig(
+        d_model=64, n_heads=2, n_layers=1, d_ff=128, hidden=128, comp_dim=16,
+        dropout=0.0)).to(dev)
+    opt = FusedAdam(model.parameters(), lr=1e-3)
+    x = torch.randn(4, 30, spec.num_paths, device=dev)
+    y = torch.rand(4, 30, spec.num_metrics, device=dev)
+    losses = []
+    for _ in range(10):
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+            out = model(x)
+            loss = model.loss(out.float(), y)
+        opt.zero_grad(set_to_none=True)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert all(np.isfinite(losses))
+    assert losses[-1] < losses[0]
