@@ -190,6 +190,9 @@ std::vector<at::Tensor> gru_seq_backward_kernel(at::Tensor grad_h, at::Tensor w_
   int B = (int)grad_h.size(0), TT = (int)grad_h.size(1);
   int C = (int)grad_h.size(2), H = (int)grad_h.size(3);
   TORCH_CHECK(H == 128);
+  TORCH_CHECK(saves.numel() == (int64_t)B * TT * C * 4 * H,
+              "gru backward: saves tensor missing or wrong size (forward must "
+              "run with save=true)");
   auto dt = grad_h.scalar_type();
   TORCH_CHECK(grad_h.is_contiguous() && h_all.is_contiguous() && saves.is_contiguous());
   auto dpre_x = at::empty({B, TT, C, 3 * H}, grad_h.options());

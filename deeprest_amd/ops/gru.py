@@ -74,9 +74,10 @@ class _FusedGRUSequence(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x_gates, w_hh, b_hh, h0, gamma, beta, reverse):
         ext = require_native("fused_gru_sequence")
-        need_grad = any(
-            t.requires_grad for t in (x_gates, w_hh, b_hh, h0, gamma, beta)
-        ) and torch.is_grad_enabled()
+        # NOTE: grad mode is disabled inside Function.forward, so
+        # torch.is_grad_enabled() is useless here — ctx.needs_input_grad is
+        # the correct signal for whether backward will run.
+        need_grad = any(ctx.needs_input_grad)
         h_all, saves = ext.gru_seq_forward(
             x_gates, w_hh, b_hh, h0, gamma, beta, bool(reverse), bool(need_grad)
         )

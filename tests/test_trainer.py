@@ -1,0 +1,110 @@
+import numpy as np
+import torch
+
+from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+from deeprest_amd.engine.config import DataConfig, EngineConfig, TrainConfig
+from deeprest_amd.engine.dataset import EstimationDataset
+from deeprest_amd.engine.trainer import Trainer
+from deeprest_amd.models.net import DeepRestNetConfig
+
+
+def tiny_config(tmp_path=None, epochs=2):
+    cfg = EngineConfig()
+    cfg.data = DataConfig(step_size=20, split=0.4)
+    cfg.train = TrainConfig(epochs=epochs, batch_size=8, baseline_epochs=3,
+                            eval_cycles=3, log_every=0, seed=0)
+    cfg.model = DeepRestNetConfig(d_model=32, n_heads=4, n_layers=1, d_ff=64,
+                                  hidden=16, comp_dim=8, dropout=0.0)
+    if tmp_path is not None:
+        cfg.train.checkpoint_path = str(tmp_path / "ckpt.pt")
+    return cfg
+
+
+def tiny_data():
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=4, n_components=5, windows_per_day=60, n_days=2, seed=21))
+    return app.generate_featurized()
+
+
+def test_dataset_split_and_scalers():
+    data = tiny_data()
+    ds = EstimationDataset(data, step_size=20, split_fraction=0.4)
+    assert ds.X.shape[0] == ds.num_windows
+    assert ds.X.shape[1:] == (20, data.num_paths)
+    assert ds.y.shape[2] == len(data.metric_names)
+    # train normalization maps train split into [0, 1]
+    assert float(ds.X_train.min()) >= 0.0 and float(ds.X_train.max()) <= 1.0
+    idx = ds.eval_window_indices(5)
+    assert idx[0] == 0 and all(i % 20 == 0 for i in idx)
+    # denormalize round-trip
+    m0 = ds.denormalize_metric(ds.y[:, :, 0].numpy(), 0)
+    np.testing.assert_allclose(m0, ds.y_raw[:, :, 0], rtol=1e-5, atol=1e-5)
+
+
+def test_trainer_end_to_end_with_baselines(tmp_path):
+    data = tiny_data()
+    cfg = tiny_config(tmp_path)
+    trainer = Trainer(data, cfg, device=torch.device("cpu"))
+    result = trainer.train()
+    assert len(result.train_losses) == 2
+    assert np.isfinite(result.train_losses).all()
+    # error tables carry all three estimators for every metric
+    assert set(result.error_tables.keys()) == set(data.metric_names)
+    for per_est in result.error_tables.values():
+        assert set(per_est.keys()) == {"resrc", "comp", "deepr"}
+        for stats in per_est.values():
+            assert set(stats.keys()) == {"median", "p95", "p99", "max"}
+            assert np.isfinite(stats["median"])
+    assert "=====" in result.summary()
+    assert result.samples_per_sec > 0
+
+
+def test_trainer_checkpoint_resume(tmp_path):
+    data = tiny_data()
+    cfg = tiny_config(tmp_path, epochs=2)
+    cfg.train.run_baselines = False
+    t1 = Trainer(data, cfg, device=torch.device("cpu"))
+    t1.train()
+
+    cfg2 = tiny_config(tmp_path, epochs=3)
+    cfg2.train.run_baselines = False
+    cfg2.train.resume = True
+    t2 = Trainer(data, cfg2, device=torch.device("cpu"))
+    result = t2.train()
+    assert t2.start_epoch == 2        # resumed from epoch 2
+    assert len(result.train_losses) == 1  # only one more epoch ran
+
+
+def test_checkpoint_contains_scalers_and_feature_space(tmp_path):
+    from deeprest_amd.engine.checkpoint import load_checkpoint
+
+    data = tiny_data()
+    cfg = tiny_config(tmp_path, epochs=1)
+    cfg.train.run_baselines = False
+    t = Trainer(data, cfg, device=torch.device("cpu"))
+    t.train()
+    state = load_checkpoint(cfg.train.checkpoint_path)
+    assert state["epoch"] == 1
+    assert state["scalers"]["metric_names"] == data.metric_names
+    assert len(state["scalers"]["y_scalers"]) == len(data.metric_names)
+    assert state["feature_space"] is not None
+    assert state["model"]["spec"]["num_paths"] == data.num_paths
+
+
+def test_config_yaml_roundtrip(tmp_path):
+    cfg = tiny_config()
+    p = str(tmp_path / "cfg.yaml")
+    cfg.save(p)
+    cfg2 = EngineConfig.load(p)
+    assert cfg2.train.epochs == cfg.train.epochs
+    assert cfg2.model.d_model == cfg.model.d_model
+    assert cfg2.data.step_size == cfg.data.step_size
+
+
+def test_config_cli_overrides():
+    from deeprest_amd.engine.config import apply_cli_overrides
+
+    cfg = tiny_config()
+    out = apply_cli_overrides(cfg, ["train.epochs=9", "model.d_model=128"])
+    assert out.train.epochs == 9
+    assert out.model.d_model == 128
