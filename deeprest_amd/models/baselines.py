@@ -105,6 +105,36 @@ class ResourceAwareBaseline(nn.Module):
         return np.tile(pred, (n_test, 1))
 
 
+class TraceAwareBaseline:
+    """Ridge regression from call-path traffic features to the metric.
+
+    The third comparison estimator the demo UI expects ('bl-trace',
+    reference: web-demo/dataloader.py:112-125): API-aware at call-path
+    granularity but linear — no temporal modeling.  Fit on the train split
+    by closed-form ridge; predictions are per-window.
+    """
+
+    def __init__(self, split: int, ridge: float = 1e-3) -> None:
+        self.split = split
+        self.ridge = ridge
+
+    def fit_and_estimate(self, X_windows: np.ndarray, y_windows: np.ndarray) -> np.ndarray:
+        """X_windows: (N, T, P); y_windows: (N, T). Returns (N - split, T)."""
+        X = np.asarray(X_windows, dtype=np.float64)
+        y = np.asarray(y_windows, dtype=np.float64)
+        N, T, P = X.shape
+        Xf = X[: self.split].reshape(-1, P)
+        yf = y[: self.split].reshape(-1)
+        # bias column + ridge-regularized normal equations
+        Xf = np.concatenate([Xf, np.ones((len(Xf), 1))], axis=1)
+        A = Xf.T @ Xf + self.ridge * np.eye(P + 1)
+        w = np.linalg.solve(A, Xf.T @ yf)
+        Xt = X[self.split :].reshape(-1, P)
+        Xt = np.concatenate([Xt, np.ones((len(Xt), 1))], axis=1)
+        pred = (Xt @ w).reshape(-1, T)
+        return np.maximum(pred, 1e-6)
+
+
 class ComponentAwareBaseline:
     """Min-max linear scaling from component invocation counts to the metric."""
 
