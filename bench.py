@@ -37,7 +37,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch", type=int, default=32, help="per-GPU window batch")
+    p.add_argument("--batch", type=int, default=256, help="per-GPU window batch")
     p.add_argument("--endpoints", type=int, default=256, help="API endpoints")
     p.add_argument("--components", type=int, default=64)
     p.add_argument("--seq-len", type=int, default=60)
@@ -63,10 +63,12 @@ def main():
 
     # ---- synthetic 256-endpoint app (identical on every rank: same seed) ----
     T = args.seq_len
+    # enough raw windows that the train split holds >= one full batch
+    need = int(args.batch / 0.8) + T + 64
     app = SyntheticApp(SyntheticAppConfig(
         n_apis=args.endpoints,
         n_components=args.components,
-        windows_per_day=max(4 * T, 240),
+        windows_per_day=max(4 * T, need),
         n_days=1,
         shapes_per_api=3,
         seed=1234,
@@ -85,6 +87,9 @@ def main():
     B = args.batch
     autocast_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
     use_autocast = on_gpu and args.dtype == "bf16"
+
+    if n < B:
+        raise RuntimeError(f"only {n} train windows for per-GPU batch {B}")
 
     def step(i: int):
         s = (i * B) % max(n - B, 1)
