@@ -1,0 +1,155 @@
+"""REST ingestion + estimation API (north star: "REST ingestion API").
+
+Endpoints (FastAPI; run with `uvicorn deeprest_amd.serve.api:create_app`):
+
+- GET  /health                    liveness + native-extension status
+- POST /ingest                    append raw-data windows (the L4 contract
+                                  as JSON instead of pickle)
+- GET  /ingest/stats              windows/apis/components ingested
+- POST /featurize                 build the call-path feature space over the
+                                  ingested windows
+- POST /estimate                  what-if estimation for a traffic plan
+                                  [{api: count}, ...] using the loaded model
+- POST /anomaly                   sanity-check measured series against the
+                                  model's quantile band
+- GET  /apis                      known API endpoints (for what-if queries)
+
+The app holds an IngestStore plus an optional Predictor loaded from a
+checkpoint; everything is JSON-serializable.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional
+
+import numpy as np
+
+from ..data.contract import ContractError, validate_raw_data
+from ..data.featurize import Featurizer
+from ..data.synthesizer import TraceSynthesizer
+from .anomaly import AnomalyScorer
+from .predictor import Predictor
+
+
+class IngestStore:
+    """In-memory raw-data window store behind the REST surface."""
+
+    def __init__(self) -> None:
+        self.windows: List[Dict[str, Any]] = []
+        self.featurizer: Optional[Featurizer] = None
+        self.synthesizer: Optional[TraceSynthesizer] = None
+
+    def ingest(self, windows: List[Dict[str, Any]]) -> int:
+        validate_raw_data(windows)
+        self.windows.extend(windows)
+        return len(self.windows)
+
+    def featurize(self):
+        self.featurizer = Featurizer(use_native=False).fit(self.windows)
+        data = self.featurizer.transform(self.windows)
+        self.synthesizer = TraceSynthesizer(
+            feature_space=self.featurizer.feature_space
+        ).fit(self.windows)
+        return data
+
+
+def create_app(checkpoint_path: Optional[str] = None, predictor: Optional[Predictor] = None):
+    from fastapi import FastAPI, HTTPException
+
+    app = FastAPI(title="deeprest-amd", version="0.1.0")
+    store = IngestStore()
+    state = {"predictor": predictor}
+    if checkpoint_path and predictor is None:
+        state["predictor"] = Predictor.from_checkpoint(checkpoint_path)
+
+    @app.get("/health")
+    def health():
+        from ..ops import native_available
+
+        return {
+            "status": "ok",
+            "native_extension": native_available(),
+            "model_loaded": state["predictor"] is not None,
+        }
+
+    @app.post("/ingest")
+    def ingest(payload: List[Dict[str, Any]]):
+        try:
+            total = store.ingest(payload)
+        except ContractError as e:
+            raise HTTPException(status_code=422, detail=str(e))
+        return {"windows_total": total}
+
+    @app.get("/ingest/stats")
+    def ingest_stats():
+        n = len(store.windows)
+        apis = set()
+        comps = set()
+        for w in store.windows:
+            for tr in w.get("traces", []):
+                apis.add(f"{tr['component']}_{tr['operation']}")
+            for m in w.get("metrics", []):
+                comps.add(m["component"])
+        return {"windows": n, "apis": sorted(apis), "components": sorted(comps)}
+
+    @app.post("/featurize")
+    def featurize():
+        if not store.windows:
+            raise HTTPException(status_code=400, detail="no ingested windows")
+        data = store.featurize()
+        return {
+            "num_paths": data.num_paths,
+            "num_windows": data.num_windows,
+            "metrics": data.metric_names,
+        }
+
+    @app.get("/apis")
+    def apis():
+        if store.synthesizer is None:
+            raise HTTPException(status_code=400, detail="featurize first")
+        return {"apis": store.synthesizer.apis}
+
+    @app.post("/estimate")
+    def estimate(payload: Dict[str, Any]):
+        """payload: {'traffic_plan': [{api: count}, ...], 'seed': int?}"""
+        pred: Optional[Predictor] = state["predictor"]
+        if pred is None:
+            raise HTTPException(status_code=400, detail="no model loaded")
+        if store.synthesizer is None:
+            raise HTTPException(status_code=400, detail="featurize first")
+        plan = payload.get("traffic_plan")
+        if not plan:
+            raise HTTPException(status_code=422, detail="traffic_plan required")
+        rng = np.random.default_rng(payload.get("seed"))
+        try:
+            out = pred.predict_what_if(
+                store.synthesizer, plan,
+                step_size=int(payload.get("step_size", 60)), rng=rng,
+            )
+        except KeyError as e:
+            raise HTTPException(status_code=422, detail=str(e))
+        return {
+            "quantiles": [0.05, 0.50, 0.95],
+            "predictions": {k: v.tolist() for k, v in out.items()},
+        }
+
+    @app.post("/anomaly")
+    def anomaly(payload: Dict[str, Any]):
+        """payload: {'measured': {metric: [..]}, 'predicted': {metric: [[q05,q50,q95],..]}}"""
+        scorer = AnomalyScorer(
+            threshold=float(payload.get("threshold", 0.25)),
+            min_run=int(payload.get("min_run", 3)),
+        )
+        measured = {k: np.asarray(v) for k, v in payload["measured"].items()}
+        preds = {k: np.asarray(v) for k, v in payload["predicted"].items()}
+        reports = scorer.score_all(measured, preds)
+        return {
+            k: {
+                "anomalous": r.is_anomalous,
+                "windows": r.windows,
+                "max_score": float(r.scores.max()) if len(r.scores) else 0.0,
+            }
+            for k, r in reports.items()
+        }
+
+    return app

@@ -1,0 +1,145 @@
+import numpy as np
+import pytest
+import torch
+
+from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+from deeprest_amd.data.synthesizer import TraceSynthesizer
+from deeprest_amd.engine.config import DataConfig, EngineConfig, TrainConfig
+from deeprest_amd.engine.trainer import Trainer
+from deeprest_amd.models.net import DeepRestNetConfig
+from deeprest_amd.serve.anomaly import AnomalyScorer
+from deeprest_amd.serve.predictor import Predictor
+from deeprest_amd.serve.results import ResultsStore, build_results_entry
+
+
+@pytest.fixture(scope="module")
+def trained(tmp_path_factory):
+    tmp = tmp_path_factory.mktemp("serve")
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=4, n_components=5, windows_per_day=60, n_days=2, seed=33))
+    raw = app.generate_raw()
+    data = app.generate_featurized()
+    cfg = EngineConfig()
+    cfg.data = DataConfig(step_size=20, split=0.4)
+    cfg.train = TrainConfig(epochs=1, batch_size=8, run_baselines=False,
+                            log_every=0, checkpoint_path=str(tmp / "ckpt.pt"))
+    cfg.model = DeepRestNetConfig(d_model=32, n_heads=4, n_layers=1, d_ff=64,
+                                  hidden=16, comp_dim=8, dropout=0.0)
+    trainer = Trainer(data, cfg, device=torch.device("cpu"))
+    trainer.train()
+    return app, raw, data, cfg, str(tmp / "ckpt.pt")
+
+
+def test_predictor_from_checkpoint(trained):
+    app, raw, data, cfg, ckpt = trained
+    pred = Predictor.from_checkpoint(ckpt, device=torch.device("cpu"))
+    windows = np.asarray(data.traffic[:25], dtype=np.float64)[None].repeat(2, axis=0)
+    # use proper (N, T, P) windows
+    from deeprest_amd.data.windows import sliding_window
+
+    w = sliding_window(np.asarray(data.traffic, dtype=np.float64), 20)[:3]
+    out = pred.predict(w)
+    assert set(out.keys()) == set(data.metric_names)
+    for v in out.values():
+        assert v.shape == (3, 20, 3)
+        assert (v >= 0).all()
+
+
+def test_predictor_what_if(trained):
+    app, raw, data, cfg, ckpt = trained
+    pred = Predictor.from_checkpoint(ckpt, device=torch.device("cpu"))
+    syn = TraceSynthesizer(feature_space=app.feature_space).fit(raw)
+    plan = [{app.apis[0]: 5, app.apis[1]: 2}] * 30
+    out = pred.predict_what_if(syn, plan, step_size=20,
+                               rng=np.random.default_rng(0))
+    assert set(out.keys()) == set(data.metric_names)
+
+
+def test_results_store_schema(tmp_path):
+    measurement = np.abs(np.random.default_rng(0).normal(50, 10, size=300))
+    preds = {
+        est: np.abs(np.random.default_rng(1).normal(50, 10, size=(2, 60)))
+        for est in ("bl-resrc", "bl-api", "bl-trace", "ours")
+    }
+    entry = build_results_entry(measurement, preds, calls=[np.arange(300)],
+                                train_len=180)
+    # exact key layout the web-demo reader expects (dataloader.py:112-125)
+    for est in ("bl-resrc", "bl-api", "bl-trace", "ours"):
+        assert f"prediction_{est}" in entry
+        assert f"scale_{est}" in entry
+        assert len(entry[f"scale_{est}"]) == 2
+        assert len(entry[f"prediction_{est}"]) == 120
+    assert "scale_groundtruth" in entry
+    assert len(entry["measurement"]) == 300
+
+    store = ResultsStore()
+    store.add("exp1-waves_seen-1x", "frontend", "cpu", entry)
+    p = str(tmp_path / "results.pkl")
+    store.save(p)
+    loaded = ResultsStore.load(p)
+    assert loaded.get("exp1-waves_seen-1x", "frontend", "cpu")["scale_ours"] == entry["scale_ours"]
+
+
+def test_anomaly_scorer_flags_injected_cpu_thief():
+    T = 100
+    rng = np.random.default_rng(2)
+    q50 = 50 + 5 * np.sin(np.arange(T) / 7)
+    q05, q95 = q50 - 10, q50 + 10
+    measured = q50 + rng.normal(0, 2, T)
+    measured[40:55] += 60.0  # cryptojacking burst not justified by traffic
+    rep = AnomalyScorer(threshold=0.25, min_run=3).score(measured, q05, q50, q95,
+                                                         metric="svc_cpu")
+    assert rep.is_anomalous
+    assert any(s <= 40 < e or (s >= 40 and e <= 56) for s, e in rep.windows)
+    # clean region unflagged
+    assert not rep.flags[:35].any()
+    assert not rep.flags[60:].any()
+
+
+def test_anomaly_scorer_clean_series_unflagged():
+    T = 80
+    q50 = np.full(T, 30.0)
+    rep = AnomalyScorer().score(q50 + 0.5, q50 - 5, q50, q50 + 5)
+    assert not rep.is_anomalous
+
+
+def test_rest_api_end_to_end(trained):
+    from starlette.testclient import TestClient
+
+    from deeprest_amd.serve.api import create_app
+
+    app_obj, raw, data, cfg, ckpt = trained
+    pred = Predictor.from_checkpoint(ckpt, device=torch.device("cpu"))
+    api = create_app(predictor=pred)
+    client = TestClient(api)
+
+    r = client.get("/health")
+    assert r.status_code == 200 and r.json()["status"] == "ok"
+
+    r = client.post("/ingest", json=raw[:40])
+    assert r.status_code == 200 and r.json()["windows_total"] == 40
+
+    r = client.post("/ingest", json=[{"bad": "window"}])
+    assert r.status_code == 422
+
+    r = client.post("/featurize")
+    assert r.status_code == 200
+    assert r.json()["num_windows"] == 40
+
+    r = client.get("/apis")
+    assert r.status_code == 200
+    apis = r.json()["apis"]
+    assert len(apis) == 4
+
+    r = client.post("/estimate", json={"traffic_plan": [{apis[0]: 3}] * 25,
+                                       "step_size": 20, "seed": 0})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["quantiles"] == [0.05, 0.50, 0.95]
+    assert set(body["predictions"].keys()) == set(data.metric_names)
+
+    measured = {data.metric_names[0]: [100.0] * 10}
+    predicted = {data.metric_names[0]: [[10.0, 20.0, 30.0]] * 10}
+    r = client.post("/anomaly", json={"measured": measured, "predicted": predicted})
+    assert r.status_code == 200
+    assert r.json()[data.metric_names[0]]["anomalous"] is True
