@@ -33,6 +33,9 @@ void dr_gru_fwd(const void* xg, const void* gamma, const void* beta,
 void dr_gru_bwd(const void* grad_h, const void* w_hh, const void* h0,
                 const void* h_all, const void* saves, void* dpre_x, float* dh0,
                 int B, int TT, int C, int reverse, int is_bf16, hipStream_t stream);
+void dr_gru_bwd_reduce(const void* dpre, const void* gamma, const void* xg,
+                       void* dxg, float* dgamma, float* dbeta, int64_t BT, int C,
+                       int is_bf16, hipStream_t stream);
 void dr_mha_fwd(const void* q, const void* k, const void* v, void* o, float* lse,
                 int64_t BH, int T_len, int D, float scale, int is_bf16,
                 hipStream_t stream);
@@ -195,12 +198,30 @@ std::vector<at::Tensor> gru_seq_backward_kernel(at::Tensor grad_h, at::Tensor w_
               "run with save=true)");
   auto dt = grad_h.scalar_type();
   TORCH_CHECK(grad_h.is_contiguous() && h_all.is_contiguous() && saves.is_contiguous());
-  auto dpre_x = at::empty({B, TT, C, 3 * H}, grad_h.options());
+  auto dpre = at::empty({B, TT, C, 4 * H}, grad_h.options());
   auto dh0 = at::empty({B, C, H}, grad_h.options().dtype(at::kFloat));
   dr_gru_bwd(grad_h.data_ptr(), w_hh.data_ptr(), h0.data_ptr(), h_all.data_ptr(),
-             saves.data_ptr(), dpre_x.data_ptr(), dh0.data_ptr<float>(), B, TT, C,
+             saves.data_ptr(), dpre.data_ptr(), dh0.data_ptr<float>(), B, TT, C,
              reverse ? 1 : 0, dt == at::kBFloat16, cur_stream());
-  return {dpre_x, dh0};
+  return {dpre, dh0};
+}
+
+// single-pass reductions over dpre: dxg, dgamma, dbeta
+std::vector<at::Tensor> gru_bwd_reduce(at::Tensor dpre, at::Tensor gamma,
+                                       at::Tensor xg) {
+  const at::cuda::CUDAGuard guard(dpre.device());
+  TORCH_CHECK(dpre.dim() == 4 && dpre.size(3) == 512);
+  int64_t BT = dpre.size(0) * dpre.size(1);
+  int C = (int)dpre.size(2);
+  TORCH_CHECK(xg.is_contiguous() && gamma.is_contiguous() && dpre.is_contiguous());
+  auto dxg = at::empty({dpre.size(0), dpre.size(1), 384}, dpre.options());
+  auto dgamma = at::zeros({C, 384}, dpre.options().dtype(at::kFloat));
+  auto dbeta = at::zeros({C, 384}, dpre.options().dtype(at::kFloat));
+  dr_gru_bwd_reduce(dpre.data_ptr(), gamma.data_ptr(), xg.data_ptr(),
+                    dxg.data_ptr(), dgamma.data_ptr<float>(),
+                    dbeta.data_ptr<float>(), BT, C,
+                    dpre.scalar_type() == at::kBFloat16, cur_stream());
+  return {dxg, dgamma, dbeta};
 }
 
 // -------------------------------------------------------------------- mha
@@ -231,5 +252,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam", &fused_adam);
   m.def("gru_seq_forward", &gru_seq_forward);
   m.def("gru_seq_backward_kernel", &gru_seq_backward_kernel);
+  m.def("gru_bwd_reduce", &gru_bwd_reduce);
   m.def("mha_forward", &mha_forward);
 }

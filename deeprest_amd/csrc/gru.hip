@@ -262,7 +262,7 @@ __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
     const T* __restrict__ h0,       // (B, C, H)
     const T* __restrict__ h_all,    // (B, TT, C, H)
     const T* __restrict__ saves,    // (B, TT, C, 4H)
-    T* __restrict__ dpre_x,         // (B, TT, C, 3H)
+    T* __restrict__ dpre_x,         // (B, TT, C, 4H): dr_pre|dz_pre|dn_pre|d_hh_n
     float* __restrict__ dh0,        // (B, C, H)
     int B, int TT, int C, int reverse) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -354,10 +354,11 @@ __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
         dh_carry[i][nt] = dht * zp;    // partial; MFMA adds dpre @ W
         zs[i][nt] = zp;
 
-        T* dx = dpre_x + ((bc + t) * C + comp_of[i]) * G3H;
+        T* dx = dpre_x + ((bc + t) * C + comp_of[i]) * (4 * H);
         stf(dx + col, drp);
         stf(dx + H + col, dzp);
         stf(dx + 2 * H + col, dnp);
+        stf(dx + 3 * H + col, dhhn);
         // LDS dpre_W image (768 B rows): dr_pre | dz_pre | d_hh_n
         *reinterpret_cast<uint16_t*>(DPl + swz768(row_of[i], col)) = f2bf(drp);
         *reinterpret_cast<uint16_t*>(DPl + swz768(row_of[i], H + col)) = f2bf(dzp);
@@ -410,6 +411,74 @@ __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
   }
 }
 
+// ------------------------------------------------- backward reductions
+// One pass over dpre per kernel (replaces einsum broadcasts that dominated
+// the profiled step: 44% elementwise time before this existed).
+
+// dxg[bt, j] = sum_c dpre[bt, c, j] * gamma[c, j]   (j in [0, 3H))
+template <typename T>
+__global__ void gru_dxg_kernel(const T* __restrict__ dpre,   // (BT, C, 4H)
+                               const T* __restrict__ gamma,  // (C, 3H)
+                               T* __restrict__ dxg,          // (BT, 3H)
+                               int64_t BT, int C) {
+  const int64_t n_chunks = BT * (G3H / 8);
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < n_chunks;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t bt = idx / (G3H / 8);
+    const int j0 = (int)(idx % (G3H / 8)) * 8;
+    float acc[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] = 0.f;
+    const T* base = dpre + bt * C * (4 * H) + j0;
+    for (int c = 0; c < C; ++c) {
+      const T* dp = base + (int64_t)c * (4 * H);
+      const T* gm = gamma + (int64_t)c * G3H + j0;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[e] += ldf(dp + e) * ldf(gm + e);
+    }
+    T* out = dxg + bt * G3H + j0;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) stf(out + e, acc[e]);
+  }
+}
+
+// dgamma[c, j] = sum_bt dpre[bt, c, j] * xg[bt, j];  dbeta[c, j] = sum_bt dpre
+// grid.y slices the BT axis; f32 atomics finalize (few M adds total).
+template <typename T>
+__global__ void gru_dgamma_kernel(const T* __restrict__ dpre,  // (BT, C, 4H)
+                                  const T* __restrict__ xg,    // (BT, 3H)
+                                  float* __restrict__ dgamma,  // (C, 3H) f32, zeroed
+                                  float* __restrict__ dbeta,   // (C, 3H) f32, zeroed
+                                  int64_t BT, int C) {
+  const int n_threads_needed = C * (G3H / 8);
+  const int tid_g = blockIdx.x * blockDim.x + threadIdx.x;
+  if (tid_g >= n_threads_needed) return;
+  const int c = tid_g / (G3H / 8);
+  const int j0 = (tid_g % (G3H / 8)) * 8;
+  const int64_t bt_lo = BT * blockIdx.y / gridDim.y;
+  const int64_t bt_hi = BT * (blockIdx.y + 1) / gridDim.y;
+  float accg[8], accb[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) { accg[e] = 0.f; accb[e] = 0.f; }
+  for (int64_t bt = bt_lo; bt < bt_hi; ++bt) {
+    const T* dp = dpre + (bt * C + c) * (4 * H) + j0;
+    const T* x = xg + bt * G3H + j0;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float d = ldf(dp + e);
+      accg[e] += d * ldf(x + e);
+      accb[e] += d;
+    }
+  }
+  float* g_out = dgamma + (int64_t)c * G3H + j0;
+  float* b_out = dbeta + (int64_t)c * G3H + j0;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    atomicAdd(g_out + e, accg[e]);
+    atomicAdd(b_out + e, accb[e]);
+  }
+}
+
 template <typename T>
 static void gru_fwd_launch_t(const void* xg, const void* gamma, const void* beta,
                              const void* w_hh, const float* b_hh, const void* h0,
@@ -459,9 +528,39 @@ static void gru_bwd_launch_t(const void* grad_h, const void* w_hh, const void* h
                      dh0, B, TT, C, reverse);
 }
 
+template <typename T>
+static void gru_reduce_launch_t(const void* dpre, const void* gamma, const void* xg,
+                                void* dxg, float* dgamma, float* dbeta, int64_t BT,
+                                int C, hipStream_t stream) {
+  {
+    int64_t n = BT * (G3H / 8);
+    int grid = (int)std::min<int64_t>((n + 255) / 256, 4096);
+    hipLaunchKernelGGL((gru_dxg_kernel<T>), dim3(grid), dim3(256), 0, stream,
+                       (const T*)dpre, (const T*)gamma, (T*)dxg, BT, C);
+  }
+  {
+    int n_threads = C * (G3H / 8);
+    int gx = (n_threads + 255) / 256;
+    int gy = 32;  // BT slices
+    hipLaunchKernelGGL((gru_dgamma_kernel<T>), dim3(gx, gy), dim3(256), 0, stream,
+                       (const T*)dpre, (const T*)xg, dgamma, dbeta, BT, C);
+  }
+}
+
 }  // namespace dr
 
 extern "C" {
+
+void dr_gru_bwd_reduce(const void* dpre, const void* gamma, const void* xg,
+                       void* dxg, float* dgamma, float* dbeta, int64_t BT, int C,
+                       int is_bf16, hipStream_t stream) {
+  if (is_bf16)
+    dr::gru_reduce_launch_t<uint16_t>(dpre, gamma, xg, dxg, dgamma, dbeta, BT, C,
+                                      stream);
+  else
+    dr::gru_reduce_launch_t<float>(dpre, gamma, xg, dxg, dgamma, dbeta, BT, C,
+                                   stream);
+}
 
 void dr_gru_fwd(const void* xg, const void* gamma, const void* beta,
                 const void* w_hh, const float* b_hh, const void* h0, void* h_all,

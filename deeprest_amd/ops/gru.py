@@ -90,33 +90,48 @@ class _FusedGRUSequence(torch.autograd.Function):
         ext = require_native("fused_gru_sequence")
         x_gates, w_hh, h0, gamma, h_all, saves = ctx.saved_tensors
         reverse = ctx.reverse
-        # sequential chain (custom kernel): gate pre-activation grads + dh0
-        dpre_x, dh0 = ext.gru_seq_backward_kernel(
+        # sequential chain (custom kernel): dpre = [dr_pre|dz_pre|dn_pre|d_hh_n]
+        dpre, dh0 = ext.gru_seq_backward_kernel(
             grad_h_all.contiguous(), w_hh, h0, h_all, saves, reverse
         )
-        B, T, C, G = dpre_x.shape
-        H = G // 3
-        # batched reductions: plain GEMMs / f32-accumulated sums (rocBLAS/eager)
-        r_saved = saves[..., :H]
-        d_hhn = dpre_x[..., 2 * H :] * r_saved
-        dpre_w = torch.cat([dpre_x[..., : 2 * H], d_hhn], dim=-1)
-        if reverse:
-            h_prev = torch.cat([h_all[:, 1:], h0.unsqueeze(1)], dim=1)
-        else:
-            h_prev = torch.cat([h0.unsqueeze(1), h_all[:, :-1]], dim=1)
-        M = B * T * C
-        dw_hh = dpre_w.reshape(M, G).t() @ h_prev.reshape(M, H)          # (3H, H)
-        db_hh = dpre_w.sum(dim=(0, 1, 2), dtype=torch.float32)
-        dxg = torch.einsum("btcj,cj->btj", dpre_x, gamma)
-        dgamma = torch.einsum("btcj,btj->cj", dpre_x, x_gates)
-        dbeta = dpre_x.sum(dim=(0, 1)).to(gamma.dtype)
+        B, T, C, G4 = dpre.shape
+        H = G4 // 4
+
+        # dW_hh = sum over (b,t,c) of [dr|dz|d_hhn]^T h_prev — batched strided
+        # GEMMs (rocBLAS picks per-B batching => proper chip occupancy; no cats:
+        # the t=0/t=T-1 boundary term against h0 is a separate small bmm)
+        def dw_for(cols):
+            if reverse:
+                dp_main = dpre[:, :-1, :, cols]             # h_prev = h_all[t+1]
+                h_main = h_all[:, 1:]
+                dp_bound = dpre[:, -1, :, cols]             # h_prev = h0
+            else:
+                dp_main = dpre[:, 1:, :, cols]              # h_prev = h_all[t-1]
+                h_main = h_all[:, :-1]
+                dp_bound = dpre[:, 0, :, cols]
+            K = dp_main.shape[-1]
+            a = dp_main.reshape(B, (T - 1) * C, K).transpose(1, 2)
+            part = torch.bmm(a, h_main.reshape(B, (T - 1) * C, H))
+            ab = dp_bound.reshape(B, C, K).transpose(1, 2)
+            part = part + torch.bmm(ab, h0)
+            return part.sum(0, dtype=torch.float32)         # (K, H)
+
+        dw_rz = dw_for(slice(0, 2 * H))
+        dw_n = dw_for(slice(3 * H, 4 * H))
+        dw_hh = torch.cat([dw_rz, dw_n], dim=0)             # (3H, H) f32
+
+        s = dpre.sum(dim=(0, 1, 2), dtype=torch.float32)    # (4H,)
+        db_hh = torch.cat([s[: 2 * H], s[3 * H :]])
+
+        # single-pass fused reductions (custom kernel): dxg, dgamma, dbeta
+        dxg, dgamma, dbeta = ext.gru_bwd_reduce(dpre, gamma, x_gates)
         return (
-            dxg.to(x_gates.dtype),
+            dxg,
             dw_hh.to(w_hh.dtype),
             db_hh,
             dh0.to(h0.dtype),
             dgamma.to(gamma.dtype),
-            dbeta,
+            dbeta.to(gamma.dtype),
             None,
         )
 
