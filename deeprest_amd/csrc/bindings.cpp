@@ -216,7 +216,7 @@ std::vector<at::Tensor> gru_bwd_reduce(at::Tensor dpre, at::Tensor gamma,
   TORCH_CHECK(xg.is_contiguous() && gamma.is_contiguous() && dpre.is_contiguous());
   auto dxg = at::empty({dpre.size(0), dpre.size(1), 384}, dpre.options());
   auto dgamma = at::zeros({C, 384}, dpre.options().dtype(at::kFloat));
-  auto dbeta = at::zeros({C, 384}, dpre.options().dtype(at::kFloat));
+  auto dbeta = at::zeros({C, 512}, dpre.options().dtype(at::kFloat));
   dr_gru_bwd_reduce(dpre.data_ptr(), gamma.data_ptr(), xg.data_ptr(),
                     dxg.data_ptr(), dgamma.data_ptr<float>(),
                     dbeta.data_ptr<float>(), BT, C,

@@ -71,7 +71,17 @@ __device__ __forceinline__ float group16_max(float v) {
   return v;
 }
 
-__device__ __forceinline__ float sigmoidf_(float x) { return 1.0f / (1.0f + __expf(-x)); }
+__device__ __forceinline__ float sigmoidf_(float x) {
+  // fast path: v_exp + fast divide; exp(-x) -> inf for x << 0 gives 1/inf = 0
+  return __fdividef(1.0f, 1.0f + __expf(-x));
+}
+
+__device__ __forceinline__ float tanhf_(float x) {
+  // clamp so exp never overflows into inf/inf = NaN; tanh(+-15) == +-1 in f32
+  float xc = fminf(fmaxf(x, -15.f), 15.f);
+  float e = __expf(2.f * xc);
+  return __fdividef(e - 1.f, e + 1.f);
+}
 
 // ---- MFMA fragment types (gfx950) ----
 using f32x4 = __attribute__((__vector_size__(4 * sizeof(float)))) float;

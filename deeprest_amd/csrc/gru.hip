@@ -11,23 +11,30 @@
 // (g = xg * gamma + beta).  State stays fp32; GEMM operands are bf16
 // (SURVEY.md "keep state in fp32, activations in low precision").
 //
-// Fixed geometry: H = 128, 3H = 384, 64 rows/block, 4 waves, 256 threads.
-// LDS: W 96 KB + h-tile (bf16, swizzled) 16 KB + xg slots 24 KB = 136 KB
-// -> one block per CU (the MFMA loop runs from LDS/registers; ILP across 24
-// independent accumulator tiles covers the 1-wave/SIMD occupancy).
+// IO layout note: the training-only side tensors (forward saves r|z|n|hh_n,
+// backward dpre dr|dz|dn|d_hhn) use a within-row permutation "pi":
+//   pi(gate, col) = gate*H + (col & 15)*8 + (col >> 4)
+// so each lane's 8 values per gate (its MFMA C-fragment columns, stride 16)
+// are CONTIGUOUS 16 bytes -> one vector load/store instead of 8 scalars.
+// Only this file and ops/gru.py (the dW row unpermute) know about pi.
 //
-// Backward: the sequential part only — per reversed step compute the gate
-// pre-activation gradients (dpre) and the recurrent chain
-// dh_prev = dh*z + dpre @ W_hh (second MFMA phase, W^T image in LDS), write
-// dpre to global.  The large batched reductions (dW_hh, db_hh, dgamma,
-// dbeta, dx_gates) are plain GEMMs done by rocBLAS on the dpre tensor in the
-// autograd wrapper (deeprest_amd/ops/gru.py).
+// Backward: the sequential chain only — per reversed step compute the gate
+// pre-activation grads (phase A, vectorized pi IO) and the recurrent term
+// dh_prev = dh*z + dpre @ W_hh.  The dpre tile round-trips through L2 (same
+// CU writes then reads it after __syncthreads; per-CU L1 is updated by its
+// own stores), against a pi-row-permuted W image in LDS, so no LDS dpre
+// image and one barrier per step.  Batched reductions (dW_hh, db_hh, dgamma,
+// dbeta, dx_gates) happen outside: rocBLAS GEMMs + the fused reduce kernels
+// below (driven by ops/gru.py).
+//
+// Fixed geometry: H = 128, 3H = 384, 64 rows/block, 4 waves, 256 threads.
 #include "common.h"
 
 namespace dr {
 
 constexpr int H = 128;        // hidden size (fixed)
 constexpr int G3H = 384;      // 3*H
+constexpr int G4H = 512;      // 4*H
 constexpr int ROWS = 64;      // rows per block
 constexpr int WAVES = 4;
 constexpr int THREADS = WAVES * DR_WAVE;
@@ -35,20 +42,16 @@ constexpr int KT = H / 32;    // K-tiles of 32 in the fwd GEMM (4)
 constexpr int NT = G3H / 16;  // N-tiles of 16 (24)
 constexpr int XG_SLOTS = 32;  // distinct batch indices a 64-row tile may span
 
-// LDS byte offsets (single dynamic region, all 16B aligned)
+// forward LDS offsets (single dynamic region, all 16B aligned)
 constexpr int LDS_W = 0;                       // 384 x 128 bf16 swizzled (98304 B)
 constexpr int LDS_H = LDS_W + G3H * H * 2;     // 64 x 128 bf16 swizzled (16384 B)
 constexpr int LDS_XG = LDS_H + ROWS * H * 2;   // 32 x 384 bf16 (24576 B)
 constexpr int LDS_FWD_TOTAL = LDS_XG + XG_SLOTS * G3H * 2;  // 139264 B
 
-// bwd LDS: W^T image (128 x 384 bf16) + dpre tile (64 x 384 bf16)
-constexpr int LDS_WT = 0;                       // 98304 B
-constexpr int LDS_DPRE = LDS_WT + H * G3H * 2;  // 49152 B
-constexpr int LDS_BWD_TOTAL = LDS_DPRE + ROWS * G3H * 2;  // 147456 B
+// backward LDS: the pi-permuted W^T image only (128 rows x 384 bf16)
+constexpr int LDS_BWD_TOTAL = H * G3H * 2;     // 98304 B
 
-// swizzled byte address inside a row-major [rows][128] bf16 tile (256 B rows):
-// 16B block index ^= (row & 15) — spreads a 16-lane ds_read_b128 group over
-// 16 distinct banks (guide T2 / Guideline 4).
+// swizzled byte address inside a row-major [rows][128] bf16 tile (256 B rows)
 __device__ __forceinline__ int swz(int row, int k_elem) {
   int blk = k_elem >> 3;            // 8 bf16 = 16 B per block
   int within = (k_elem & 7) * 2;
@@ -66,6 +69,37 @@ __device__ __forceinline__ bf16x8 lds_read8(const char* base, int byte_off) {
   return *reinterpret_cast<const bf16x8*>(base + byte_off);
 }
 
+// ---- pi-layout vector IO: 8 lane-owned values <-> 16/32 contiguous bytes ----
+template <typename T>
+__device__ __forceinline__ void st8(T* dst, const float* v);
+template <>
+__device__ __forceinline__ void st8<uint16_t>(uint16_t* dst, const float* v) {
+  uint16_t p[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) p[e] = f2bf(v[e]);
+  *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(p);
+}
+template <>
+__device__ __forceinline__ void st8<float>(float* dst, const float* v) {
+  *reinterpret_cast<float4*>(dst) = *reinterpret_cast<const float4*>(v);
+  *reinterpret_cast<float4*>(dst + 4) = *reinterpret_cast<const float4*>(v + 4);
+}
+
+template <typename T>
+__device__ __forceinline__ void ld8(const T* src, float* v);
+template <>
+__device__ __forceinline__ void ld8<uint16_t>(const uint16_t* src, float* v) {
+  uint4 raw = *reinterpret_cast<const uint4*>(src);
+  const uint16_t* p = reinterpret_cast<const uint16_t*>(&raw);
+#pragma unroll
+  for (int e = 0; e < 8; ++e) v[e] = bf2f(p[e]);
+}
+template <>
+__device__ __forceinline__ void ld8<float>(const float* src, float* v) {
+  *reinterpret_cast<float4*>(v) = *reinterpret_cast<const float4*>(src);
+  *reinterpret_cast<float4*>(v + 4) = *reinterpret_cast<const float4*>(src + 4);
+}
+
 // ---------------------------------------------------------------- forward
 template <typename T, bool SAVE>
 __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
@@ -76,7 +110,7 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
     const float* __restrict__ b_hh,  // (3H,)
     const T* __restrict__ h0,      // (B, C, H)
     T* __restrict__ h_all,         // (B, TT, C, H)
-    T* __restrict__ saves,         // (B, TT, C, 4H): r|z|n|hh_n  (SAVE only)
+    T* __restrict__ saves,         // (B, TT, C, 4H) pi layout (SAVE only)
     int B, int TT, int C, int reverse) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* Wl = smem + LDS_W;
@@ -105,7 +139,7 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
   // ---- per-lane static geometry (C-layout of the 16x16 MFMA tile) ----
   const int c_col = lane & 15;           // col within a 16-wide N-tile
   const int rgrp = lane >> 4;            // row group (0..3)
-  int row_of[4];                         // absolute tile row per acc reg
+  int row_of[4];
   int64_t r_abs[4];
   int b_of[4], comp_of[4];
   bool live[4];
@@ -119,9 +153,10 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
     comp_of[i] = (int)(rr % C);
   }
 
-  // ---- preload T-invariant per-lane values: gamma/beta and b_hh ----
-  // gamma/beta for (row i, gate g, h-col tile nt) at col c_col.
-  float gm[4][3][8], bt[4][3][8];
+  // ---- preload T-invariant per-lane values ----
+  // gamma/beta packed as bf16 pairs in one u32 per (row, gate, tile):
+  // 96 registers instead of 192 (the fp32 form spilled at the 512-reg cap)
+  uint32_t gb[4][3][8];
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     const T* grow = gamma + (int64_t)comp_of[i] * G3H;
@@ -131,8 +166,9 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
         int col = g * H + nt * 16 + c_col;
-        gm[i][g][nt] = live[i] ? ldf(grow + col) : 0.f;
-        bt[i][g][nt] = live[i] ? ldf(brow + col) : 0.f;
+        uint16_t gv = live[i] ? f2bf(ldf(grow + col)) : 0;
+        uint16_t bv = live[i] ? f2bf(ldf(brow + col)) : 0;
+        gb[i][g][nt] = ((uint32_t)bv << 16) | gv;
       }
   }
   float bh[3][8];
@@ -197,7 +233,7 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
       acc[nt] = a;
     }
 
-    // ---- fused gate epilogue ----
+    // ---- fused gate epilogue (vectorized pi-layout saves) ----
     const uint16_t* xg_rows[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
@@ -205,30 +241,33 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
           XGl + (b_of[i] - b_lo) * (G3H * 2));
 
 #pragma unroll
-    for (int nt = 0; nt < 8; ++nt) {
-      int col = nt * 16 + c_col;
+    for (int i = 0; i < 4; ++i) {
+      float fr[8], fz[8], fn[8], fh[8];
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
+      for (int nt = 0; nt < 8; ++nt) {
+        int col = nt * 16 + c_col;
         float xr = bf2f(xg_rows[i][col]);
         float xz = bf2f(xg_rows[i][H + col]);
         float xn = bf2f(xg_rows[i][2 * H + col]);
-        float g_r = xr * gm[i][0][nt] + bt[i][0][nt];
-        float g_z = xz * gm[i][1][nt] + bt[i][1][nt];
-        float g_n = xn * gm[i][2][nt] + bt[i][2][nt];
+        float g_r = xr * bf2f((uint16_t)gb[i][0][nt]) + bf2f((uint16_t)(gb[i][0][nt] >> 16));
+        float g_z = xz * bf2f((uint16_t)gb[i][1][nt]) + bf2f((uint16_t)(gb[i][1][nt] >> 16));
+        float g_n = xn * bf2f((uint16_t)gb[i][2][nt]) + bf2f((uint16_t)(gb[i][2][nt] >> 16));
         float rp = sigmoidf_(acc[nt][i] + bh[0][nt] + g_r);
         float zp = sigmoidf_(acc[nt + 8][i] + bh[1][nt] + g_z);
         float hn = acc[nt + 16][i] + bh[2][nt];
-        float nn = tanhf(g_n + rp * hn);
+        float nn = tanhf_(g_n + rp * hn);
         float hnew = (1.f - zp) * nn + zp * h[i][nt];
         h[i][nt] = hnew;
         *reinterpret_cast<uint16_t*>(Hl + swz(row_of[i], col)) = f2bf(hnew);
-        if (SAVE && live[i]) {
-          T* sv = saves + (((int64_t)b_of[i] * TT + t) * C + comp_of[i]) * (4 * H);
-          stf(sv + col, rp);
-          stf(sv + H + col, zp);
-          stf(sv + 2 * H + col, nn);
-          stf(sv + 3 * H + col, hn);
-        }
+        fr[nt] = rp; fz[nt] = zp; fn[nt] = nn; fh[nt] = hn;
+      }
+      if (SAVE && live[i]) {
+        T* sv = saves + (((int64_t)b_of[i] * TT + t) * C + comp_of[i]) * G4H +
+                c_col * 8;
+        st8(sv, fr);
+        st8(sv + H, fz);
+        st8(sv + 2 * H, fn);
+        st8(sv + 3 * H, fh);
       }
     }
     __syncthreads();
@@ -253,21 +292,23 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
 }
 
 // ---------------------------------------------------------------- backward
-// Computes dpre_x (B,TT,C,3H: dr_pre|dz_pre|dn_pre) and dh0; the caller does
-// the batched reductions with rocBLAS.
+// The MFMA K axis runs over pi positions m in [0, 384): regions {dr, dz,
+// d_hhn} of the dpre tensor (skipping the dn region, which only feeds the
+// x-side).  A-fragments read the just-written GLOBAL dpre rows (same-CU L2
+// after __syncthreads); the B image in LDS is W^T with rows permuted to the
+// same m ordering, so both sides agree on any K permutation.
 template <typename T>
 __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
     const T* __restrict__ grad_h,   // (B, TT, C, H)
     const T* __restrict__ w_hh,     // (3H, H)
     const T* __restrict__ h0,       // (B, C, H)
     const T* __restrict__ h_all,    // (B, TT, C, H)
-    const T* __restrict__ saves,    // (B, TT, C, 4H)
-    T* __restrict__ dpre_x,         // (B, TT, C, 4H): dr_pre|dz_pre|dn_pre|d_hh_n
+    const T* __restrict__ saves,    // (B, TT, C, 4H) pi layout
+    T* __restrict__ dpre,           // (B, TT, C, 4H) pi: dr|dz|dn|d_hhn
     float* __restrict__ dh0,        // (B, C, H)
     int B, int TT, int C, int reverse) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* WTl = smem + LDS_WT;
-  char* DPl = smem + LDS_DPRE;
+  char* WTl = smem;  // [k_h 0..127][m 0..383] bf16, 768 B rows, swizzled
 
   const int tid = threadIdx.x;
   const int wv = tid / DR_WAVE;
@@ -275,20 +316,22 @@ __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
   const int64_t R = (int64_t)B * C;
   const int64_t r0 = (int64_t)blockIdx.x * ROWS;
 
-  // ---- stage W^T (128 x 384) into swizzled LDS ----
-  // WT[k][j] = W[j][k]; rows are 384 bf16 = 768 B -> use 16B-block swizzle
-  // with blk ^ (row & 15) on a 48-block row (blk 0..47; XOR over low 4 bits).
+  // ---- stage the pi-permuted W^T image ----
+  // m < 256: natural j = (m>>7)*H + inv_pi(m&127)        (W_hr / W_hz rows)
+  // m >= 256: natural j = 2*H + inv_pi(m&127)            (W_hn rows)
   for (int id = tid; id < G3H * (H / 8); id += THREADS) {
-    int j = id / (H / 8);          // 0..383
-    int kblk = id % (H / 8);       // 0..15 (k block of 8)
+    int j = id / (H / 8);          // natural j (0..383)
+    int kblk = id % (H / 8);
+    int g = j / H;
+    int col = j % H;
+    int m = g * H + ((col & 15) << 3) + (col >> 4);   // pi position
+    float v[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) v[e] = ldf(w_hh + (int64_t)j * H + kblk * 8 + e);
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       int k = kblk * 8 + e;
-      float v = ldf(w_hh + (int64_t)j * H + k);
-      int blk = j >> 3;            // 16B block within WT row k (j/8)
-      int within = (j & 7) * 2;
-      int byte_off = k * 768 + ((blk ^ (k & 15)) << 4) + within;
-      *reinterpret_cast<uint16_t*>(WTl + byte_off) = f2bf(v);
+      *reinterpret_cast<uint16_t*>(WTl + swz768(k, m)) = f2bf(v[e]);
     }
   }
 
@@ -313,11 +356,6 @@ __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
   for (int i = 0; i < 4; ++i)
 #pragma unroll
     for (int nt = 0; nt < 8; ++nt) dh_carry[i][nt] = 0.f;
-
-  // zero the dpre tile once: pad rows (r >= R) never write it afterwards, so
-  // their MFMA inputs stay finite
-  for (int id = tid; id < ROWS * G3H; id += THREADS)
-    reinterpret_cast<uint16_t*>(DPl)[id] = 0;
   __syncthreads();
 
   for (int step = 0; step < TT; ++step) {
@@ -328,57 +366,66 @@ __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
 
     float zs[4][8];  // saved z, needed again after the MFMA phase
 #pragma unroll
-    for (int nt = 0; nt < 8; ++nt) {
-      int col = nt * 16 + c_col;
+    for (int i = 0; i < 4; ++i) {
+      if (!live[i]) {
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        if (!live[i]) { zs[i][nt] = 0.f; continue; }
-        const int64_t bc = (int64_t)b_of[i] * TT;
-        const T* sv = saves + ((bc + t) * C + comp_of[i]) * (4 * H);
-        float rp = ldf(sv + col);
-        float zp = ldf(sv + H + col);
-        float nn = ldf(sv + 2 * H + col);
-        float hn = ldf(sv + 3 * H + col);
+        for (int nt = 0; nt < 8; ++nt) zs[i][nt] = 0.f;
+        continue;
+      }
+      const int64_t bc = (int64_t)b_of[i] * TT;
+      const T* sv = saves + ((bc + t) * C + comp_of[i]) * G4H + c_col * 8;
+      float rp[8], zp[8], nn[8], hn[8];
+      ld8(sv, rp);
+      ld8(sv + H, zp);
+      ld8(sv + 2 * H, nn);
+      ld8(sv + 3 * H, hn);
+      float fdr[8], fdz[8], fdn[8], fdh[8];
+#pragma unroll
+      for (int nt = 0; nt < 8; ++nt) {
+        int col = nt * 16 + c_col;
         float hp = use_h0
             ? ldf(h0 + ((int64_t)b_of[i] * C + comp_of[i]) * H + col)
             : ldf(h_all + ((bc + tprev) * C + comp_of[i]) * H + col);
         float g = ldf(grad_h + ((bc + t) * C + comp_of[i]) * H + col);
         float dht = g + dh_carry[i][nt];
-        float dz = dht * (hp - nn);
-        float dn = dht * (1.f - zp);
-        float dnp = dn * (1.f - nn * nn);
-        float dhhn = dnp * rp;
-        float dr = dnp * hn;
-        float drp = dr * rp * (1.f - rp);
-        float dzp = dz * zp * (1.f - zp);
-        dh_carry[i][nt] = dht * zp;    // partial; MFMA adds dpre @ W
-        zs[i][nt] = zp;
-
-        T* dx = dpre_x + ((bc + t) * C + comp_of[i]) * (4 * H);
-        stf(dx + col, drp);
-        stf(dx + H + col, dzp);
-        stf(dx + 2 * H + col, dnp);
-        stf(dx + 3 * H + col, dhhn);
-        // LDS dpre_W image (768 B rows): dr_pre | dz_pre | d_hh_n
-        *reinterpret_cast<uint16_t*>(DPl + swz768(row_of[i], col)) = f2bf(drp);
-        *reinterpret_cast<uint16_t*>(DPl + swz768(row_of[i], H + col)) = f2bf(dzp);
-        *reinterpret_cast<uint16_t*>(DPl + swz768(row_of[i], 2 * H + col)) = f2bf(dhhn);
+        float dz = dht * (hp - nn[nt]);
+        float dn = dht * (1.f - zp[nt]);
+        float dnp = dn * (1.f - nn[nt] * nn[nt]);
+        float dhhn = dnp * rp[nt];
+        float dr = dnp * hn[nt];
+        fdr[nt] = dr * rp[nt] * (1.f - rp[nt]);
+        fdz[nt] = dz * zp[nt] * (1.f - zp[nt]);
+        fdn[nt] = dnp;
+        fdh[nt] = dhhn;
+        dh_carry[i][nt] = dht * zp[nt];    // partial; MFMA adds dpre @ W
+        zs[i][nt] = zp[nt];
       }
+      T* dx = dpre + ((bc + t) * C + comp_of[i]) * G4H + c_col * 8;
+      st8(dx, fdr);
+      st8(dx + H, fdz);
+      st8(dx + 2 * H, fdn);
+      st8(dx + 3 * H, fdh);
     }
-    __syncthreads();
+    __syncthreads();  // all dpre rows visible (same-CU L2) before A-frag reads
 
-    // ---- MFMA: delta = dpre_W (64x384) @ W (384x128), wave's 16 rows ----
-    // A from DPl rows (768 B rows, swizzle over 48 blocks), B from WT image.
+    // ---- MFMA: delta = dpre_pi (64 x 384 over m) @ W_pi, wave's 16 rows ----
     {
       bf16x8 afrag[12];
-      int arow = wv * 16 + (lane & 15);
-      int k0 = (lane >> 4) * 8;
+      const int arow = wv * 16 + (lane & 15);
+      const int64_t ra = r0 + arow < R ? r0 + arow : R - 1;
+      const int ab = (int)(ra / C), ac = (int)(ra % C);
+      const T* arow_ptr = dpre + (((int64_t)ab * TT + t) * C + ac) * G4H;
+      const int k0 = (lane >> 4) * 8;
 #pragma unroll
       for (int kt = 0; kt < 12; ++kt) {
-        int k = kt * 32 + k0;                         // 0..383 (j index)
-        int blk = k >> 3;
-        afrag[kt] = lds_read8(
-            DPl, arow * 768 + ((blk ^ (arow & 15)) << 4) + ((k & 7) * 2));
+        int m = kt * 32 + k0;
+        int off = (m < 256) ? m : (m + 128);   // skip the dn region (g == 2)
+        float tmp[8];
+        ld8(arow_ptr + off, tmp);
+        uint16_t p[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) p[e] = f2bf(tmp[e]);
+        afrag[kt] = *reinterpret_cast<bf16x8*>(p);
       }
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
@@ -386,17 +433,16 @@ __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
 #pragma unroll
         for (int kt = 0; kt < 12; ++kt) {
           int n = nt * 16 + c_col;                    // output h-col
-          int j0 = kt * 32 + (lane >> 4) * 8;         // K (j) index
-          int blk = j0 >> 3;
-          bf16x8 bfrag = lds_read8(
-              WTl, n * 768 + ((blk ^ (n & 15)) << 4) + ((j0 & 7) * 2));
+          int m0 = kt * 32 + (lane >> 4) * 8;
+          bf16x8 bfrag = lds_read8(WTl, swz768(n, m0));
           a = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[kt], bfrag, a, 0, 0, 0);
         }
 #pragma unroll
         for (int i = 0; i < 4; ++i) dh_carry[i][nt] += a[i];
       }
     }
-    __syncthreads();  // DPl consumed; next step may overwrite
+    // no end-of-step barrier: the next step's stores go to DIFFERENT dpre
+    // addresses (per-t storage), and WTl is read-only after the prologue.
   }
 
   // ---- dh0 = final carry ----
@@ -412,73 +458,87 @@ __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
 }
 
 // ------------------------------------------------- backward reductions
-// One pass over dpre per kernel (replaces einsum broadcasts that dominated
-// the profiled step: 44% elementwise time before this existed).
+// One pass over dpre per kernel (replaces the einsum broadcasts that
+// dominated the profiled step).  Both kernels understand the pi layout and
+// emit NATURAL-order outputs.
 
-// dxg[bt, j] = sum_c dpre[bt, c, j] * gamma[c, j]   (j in [0, 3H))
+// dxg[bt, j] = sum_c dpre[bt, c, pi(j)] * gamma[c, j]   (j in [0, 3H))
+// thread chunk = (bt, gate, c_col): 8 pi-contiguous dpre values per c.
 template <typename T>
-__global__ void gru_dxg_kernel(const T* __restrict__ dpre,   // (BT, C, 4H)
+__global__ void gru_dxg_kernel(const T* __restrict__ dpre,   // (BT, C, 4H) pi
                                const T* __restrict__ gamma,  // (C, 3H)
                                T* __restrict__ dxg,          // (BT, 3H)
                                int64_t BT, int C) {
-  const int64_t n_chunks = BT * (G3H / 8);
+  const int64_t n_chunks = BT * 48;      // 3 gates x 16 c_col groups
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < n_chunks;
        idx += (int64_t)gridDim.x * blockDim.x) {
-    const int64_t bt = idx / (G3H / 8);
-    const int j0 = (int)(idx % (G3H / 8)) * 8;
+    const int64_t bt = idx / 48;
+    const int rem = (int)(idx % 48);
+    const int g = rem / 16;
+    const int cc = rem % 16;
+    const int pi0 = g * H + cc * 8;
     float acc[8];
 #pragma unroll
     for (int e = 0; e < 8; ++e) acc[e] = 0.f;
-    const T* base = dpre + bt * C * (4 * H) + j0;
+    const T* base = dpre + bt * C * G4H + pi0;
     for (int c = 0; c < C; ++c) {
-      const T* dp = base + (int64_t)c * (4 * H);
-      const T* gm = gamma + (int64_t)c * G3H + j0;
+      float d[8];
+      ld8(base + (int64_t)c * G4H, d);
+      const T* gm = gamma + (int64_t)c * G3H + g * H + cc;
 #pragma unroll
-      for (int e = 0; e < 8; ++e) acc[e] += ldf(dp + e) * ldf(gm + e);
+      for (int e = 0; e < 8; ++e) acc[e] += d[e] * ldf(gm + e * 16);
     }
-    T* out = dxg + bt * G3H + j0;
+    T* out = dxg + bt * G3H + g * H + cc;
 #pragma unroll
-    for (int e = 0; e < 8; ++e) stf(out + e, acc[e]);
+    for (int e = 0; e < 8; ++e) stf(out + e * 16, acc[e]);
   }
 }
 
-// dgamma[c, j] = sum_bt dpre[bt, c, j] * xg[bt, j];  dbeta[c, j] = sum_bt dpre
-// grid.y slices the BT axis; f32 atomics finalize (few M adds total).
+// dgamma[c, j<3H] = sum_bt dpre[bt, c, pi(j)] * xg[bt, j]
+// dbeta4[c, j<4H] = sum_bt dpre[bt, c, pi(j)]   (4th slice -> db_hh_n)
+// grid.y slices BT; f32 atomics finalize.  Natural-order outputs.
 template <typename T>
-__global__ void gru_dgamma_kernel(const T* __restrict__ dpre,  // (BT, C, 4H)
+__global__ void gru_dgamma_kernel(const T* __restrict__ dpre,  // (BT, C, 4H) pi
                                   const T* __restrict__ xg,    // (BT, 3H)
-                                  float* __restrict__ dgamma,  // (C, 3H) f32, zeroed
-                                  float* __restrict__ dbeta,   // (C, 3H) f32, zeroed
+                                  float* __restrict__ dgamma,  // (C, 3H) zeroed
+                                  float* __restrict__ dbeta4,  // (C, 4H) zeroed
                                   int64_t BT, int C) {
-  const int n_threads_needed = C * (G3H / 8);
+  const int n_threads_needed = C * 64;   // 4 gates x 16 c_col groups
   const int tid_g = blockIdx.x * blockDim.x + threadIdx.x;
   if (tid_g >= n_threads_needed) return;
-  const int c = tid_g / (G3H / 8);
-  const int j0 = (tid_g % (G3H / 8)) * 8;
+  const int c = tid_g / 64;
+  const int rem = tid_g % 64;
+  const int g = rem / 16;
+  const int cc = rem % 16;
+  const bool has_x = g < 3;
+  const int pi0 = g * H + cc * 8;
   const int64_t bt_lo = BT * blockIdx.y / gridDim.y;
   const int64_t bt_hi = BT * (blockIdx.y + 1) / gridDim.y;
   float accg[8], accb[8];
 #pragma unroll
   for (int e = 0; e < 8; ++e) { accg[e] = 0.f; accb[e] = 0.f; }
   for (int64_t bt = bt_lo; bt < bt_hi; ++bt) {
-    const T* dp = dpre + (bt * C + c) * (4 * H) + j0;
-    const T* x = xg + bt * G3H + j0;
+    float d[8];
+    ld8(dpre + (bt * C + c) * G4H + pi0, d);
+    if (has_x) {
+      const T* x = xg + bt * G3H + g * H + cc;
 #pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      float d = ldf(dp + e);
-      accg[e] += d * ldf(x + e);
-      accb[e] += d;
+      for (int e = 0; e < 8; ++e) accg[e] += d[e] * ldf(x + e * 16);
     }
-  }
-  float* g_out = dgamma + (int64_t)c * G3H + j0;
-  float* b_out = dbeta + (int64_t)c * G3H + j0;
 #pragma unroll
-  for (int e = 0; e < 8; ++e) {
-    atomicAdd(g_out + e, accg[e]);
-    atomicAdd(b_out + e, accb[e]);
+    for (int e = 0; e < 8; ++e) accb[e] += d[e];
+  }
+  float* b_out = dbeta4 + (int64_t)c * G4H + g * H + cc;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) atomicAdd(b_out + e * 16, accb[e]);
+  if (has_x) {
+    float* g_out = dgamma + (int64_t)c * G3H + g * H + cc;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) atomicAdd(g_out + e * 16, accg[e]);
   }
 }
 
+// ------------------------------------------------------------- launchers
 template <typename T>
 static void gru_fwd_launch_t(const void* xg, const void* gamma, const void* beta,
                              const void* w_hh, const float* b_hh, const void* h0,
@@ -510,7 +570,7 @@ static void gru_fwd_launch_t(const void* xg, const void* gamma, const void* beta
 
 template <typename T>
 static void gru_bwd_launch_t(const void* grad_h, const void* w_hh, const void* h0,
-                             const void* h_all, const void* saves, void* dpre_x,
+                             const void* h_all, const void* saves, void* dpre,
                              float* dh0, int B, int TT, int C, int reverse,
                              hipStream_t stream) {
   int64_t R = (int64_t)B * C;
@@ -524,7 +584,7 @@ static void gru_bwd_launch_t(const void* grad_h, const void* w_hh, const void* h
   }
   hipLaunchKernelGGL((gru_bwd_kernel<T>), dim3(grid), dim3(THREADS),
                      LDS_BWD_TOTAL, stream, (const T*)grad_h, (const T*)w_hh,
-                     (const T*)h0, (const T*)h_all, (const T*)saves, (T*)dpre_x,
+                     (const T*)h0, (const T*)h_all, (const T*)saves, (T*)dpre,
                      dh0, B, TT, C, reverse);
 }
 
@@ -533,13 +593,13 @@ static void gru_reduce_launch_t(const void* dpre, const void* gamma, const void*
                                 void* dxg, float* dgamma, float* dbeta, int64_t BT,
                                 int C, hipStream_t stream) {
   {
-    int64_t n = BT * (G3H / 8);
+    int64_t n = BT * 48;
     int grid = (int)std::min<int64_t>((n + 255) / 256, 4096);
     hipLaunchKernelGGL((gru_dxg_kernel<T>), dim3(grid), dim3(256), 0, stream,
                        (const T*)dpre, (const T*)gamma, (T*)dxg, BT, C);
   }
   {
-    int n_threads = C * (G3H / 8);
+    int n_threads = C * 64;
     int gx = (n_threads + 255) / 256;
     int gy = 32;  // BT slices
     hipLaunchKernelGGL((gru_dgamma_kernel<T>), dim3(gx, gy), dim3(256), 0, stream,
@@ -550,17 +610,6 @@ static void gru_reduce_launch_t(const void* dpre, const void* gamma, const void*
 }  // namespace dr
 
 extern "C" {
-
-void dr_gru_bwd_reduce(const void* dpre, const void* gamma, const void* xg,
-                       void* dxg, float* dgamma, float* dbeta, int64_t BT, int C,
-                       int is_bf16, hipStream_t stream) {
-  if (is_bf16)
-    dr::gru_reduce_launch_t<uint16_t>(dpre, gamma, xg, dxg, dgamma, dbeta, BT, C,
-                                      stream);
-  else
-    dr::gru_reduce_launch_t<float>(dpre, gamma, xg, dxg, dgamma, dbeta, BT, C,
-                                   stream);
-}
 
 void dr_gru_fwd(const void* xg, const void* gamma, const void* beta,
                 const void* w_hh, const float* b_hh, const void* h0, void* h_all,
@@ -575,14 +624,25 @@ void dr_gru_fwd(const void* xg, const void* gamma, const void* beta,
 }
 
 void dr_gru_bwd(const void* grad_h, const void* w_hh, const void* h0,
-                const void* h_all, const void* saves, void* dpre_x, float* dh0,
+                const void* h_all, const void* saves, void* dpre, float* dh0,
                 int B, int TT, int C, int reverse, int is_bf16, hipStream_t stream) {
   if (is_bf16)
-    dr::gru_bwd_launch_t<uint16_t>(grad_h, w_hh, h0, h_all, saves, dpre_x, dh0,
+    dr::gru_bwd_launch_t<uint16_t>(grad_h, w_hh, h0, h_all, saves, dpre, dh0,
                                    B, TT, C, reverse, stream);
   else
-    dr::gru_bwd_launch_t<float>(grad_h, w_hh, h0, h_all, saves, dpre_x, dh0,
+    dr::gru_bwd_launch_t<float>(grad_h, w_hh, h0, h_all, saves, dpre, dh0,
                                 B, TT, C, reverse, stream);
+}
+
+void dr_gru_bwd_reduce(const void* dpre, const void* gamma, const void* xg,
+                       void* dxg, float* dgamma, float* dbeta, int64_t BT, int C,
+                       int is_bf16, hipStream_t stream) {
+  if (is_bf16)
+    dr::gru_reduce_launch_t<uint16_t>(dpre, gamma, xg, dxg, dgamma, dbeta, BT, C,
+                                      stream);
+  else
+    dr::gru_reduce_launch_t<float>(dpre, gamma, xg, dxg, dgamma, dbeta, BT, C,
+                                   stream);
 }
 
 }  // extern "C"

@@ -36,6 +36,7 @@ import torch.nn.functional as F
 
 from ..data.featurize import FeaturizedData
 from ..ops import fused_gru_sequence, layer_norm, mha_forward, pinball_loss
+from ..ops.linear_bigk import bigk_linear
 
 
 # --------------------------------------------------------------------- spec
@@ -289,6 +290,14 @@ class DeepRestNet(nn.Module):
         self.metric_bias = nn.Parameter(torch.zeros(spec.num_metrics, Q))
         self.register_buffer("comp_of", torch.tensor(spec.comp_of, dtype=torch.long))
         self.register_buffer("res_of", torch.tensor(spec.res_of, dtype=torch.long))
+        # fast path: when the metric order IS (component-major x resource) the
+        # per-metric gather is a plain reshape (true for the synthetic apps;
+        # avoids an indexing_backward scatter over ~10^6 rows)
+        C, R, M = spec.num_components, len(spec.resources), spec.num_metrics
+        grid_c = [c for c in range(C) for _ in range(R)]
+        grid_r = list(range(R)) * C
+        self._gather_is_reshape = (M == C * R and spec.comp_of == grid_c
+                                   and spec.res_of == grid_r)
 
         # positional encoding over the window (sinusoidal, not learned: windows
         # slide, so absolute position has no meaning beyond phase)
@@ -319,10 +328,19 @@ class DeepRestNet(nn.Module):
         comp = self.graph()                                   # (C, comp_dim)
         h_all = self.decoder(x, comp)                         # (B, T, C, H*)
         h_all = self.dropout(h_all)
-        # per-resource-type heads over all components
-        outs = torch.stack([head(h_all) for head in self.heads], dim=3)
-        # (B, T, C, R, Q) -> gather each metric's (component, resource) pair
-        preds = outs[:, :, self.comp_of, self.res_of, :]      # (B, T, M, Q)
+        # per-resource-type heads over all components, as ONE packed GEMM with
+        # a batched-K backward (ops/linear_bigk.py)
+        w_all = torch.cat([h.weight for h in self.heads], dim=0)   # (R*Q, H*)
+        b_all = torch.cat([h.bias for h in self.heads], dim=0)
+        B, T = h_all.shape[0], h_all.shape[1]
+        C = h_all.shape[2]
+        R = len(self.heads)
+        Q = len(self.cfg.quantiles)
+        outs = bigk_linear(h_all, w_all, b_all).view(B, T, C, R, Q)
+        if self._gather_is_reshape:
+            preds = outs.reshape(B, T, C * R, Q)
+        else:
+            preds = outs[:, :, self.comp_of, self.res_of, :]  # (B, T, M, Q)
         return preds + self.metric_bias
 
     def loss(self, outputs: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:

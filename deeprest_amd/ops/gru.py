@@ -97,6 +97,19 @@ class _FusedGRUSequence(torch.autograd.Function):
         B, T, C, G4 = dpre.shape
         H = G4 // 4
 
+        # dpre rows use the kernel's pi packing: position g*128 + cc*8 + nt
+        # holds natural column g*128 + nt*16 + cc — unpermute dW rows after
+        # the GEMM (tiny index_copy on a (K, H) matrix).
+        def unpi(dw_pi):
+            K = dw_pi.shape[0]
+            idx = torch.arange(K, device=dw_pi.device)
+            g = idx // 128
+            q = idx % 128
+            nat = g * 128 + (q % 8) * 16 + (q // 8)
+            out = torch.empty_like(dw_pi)
+            out.index_copy_(0, nat, dw_pi)
+            return out
+
         # dW_hh = sum over (b,t,c) of [dr|dz|d_hhn]^T h_prev — batched strided
         # GEMMs (rocBLAS picks per-B batching => proper chip occupancy; no cats:
         # the t=0/t=T-1 boundary term against h0 is a separate small bmm)
@@ -114,17 +127,18 @@ class _FusedGRUSequence(torch.autograd.Function):
             part = torch.bmm(a, h_main.reshape(B, (T - 1) * C, H))
             ab = dp_bound.reshape(B, C, K).transpose(1, 2)
             part = part + torch.bmm(ab, h0)
-            return part.sum(0, dtype=torch.float32)         # (K, H)
+            return unpi(part.sum(0, dtype=torch.float32))   # (K, H) natural rows
 
         dw_rz = dw_for(slice(0, 2 * H))
         dw_n = dw_for(slice(3 * H, 4 * H))
         dw_hh = torch.cat([dw_rz, dw_n], dim=0)             # (3H, H) f32
 
-        s = dpre.sum(dim=(0, 1, 2), dtype=torch.float32)    # (4H,)
-        db_hh = torch.cat([s[: 2 * H], s[3 * H :]])
-
-        # single-pass fused reductions (custom kernel): dxg, dgamma, dbeta
-        dxg, dgamma, dbeta = ext.gru_bwd_reduce(dpre, gamma, x_gates)
+        # single-pass fused reductions (custom kernel): dxg, dgamma, dbeta4
+        # (dbeta4's 4th slice is sum of d_hh_n -> the n-part of db_hh)
+        dxg, dgamma, dbeta4 = ext.gru_bwd_reduce(dpre, gamma, x_gates)
+        s4 = dbeta4.sum(dim=0)                              # (4H,) f32, tiny
+        db_hh = torch.cat([s4[: 2 * H], s4[3 * H :]])
+        dbeta = dbeta4[:, : 3 * H]
         return (
             dxg,
             dw_hh.to(w_hh.dtype),

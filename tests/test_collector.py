@@ -1,0 +1,80 @@
+import numpy as np
+
+from deeprest_amd.data.collector import Collector, span_tree_from_jaeger
+from deeprest_amd.data.contract import validate_raw_data
+from deeprest_amd.data.featurize import Featurizer
+
+
+def make_jaeger_trace(trace_id, t0_us, svc_ops):
+    """svc_ops: list of (spanID, service, op, parentID or None)."""
+    spans = []
+    for sid, svc, op, parent in svc_ops:
+        span = {
+            "traceID": trace_id,
+            "spanID": sid,
+            "operationName": op,
+            "startTime": t0_us,
+            "serviceName": svc,
+            "references": [],
+        }
+        if parent is not None:
+            span["references"].append({"refType": "CHILD_OF", "spanID": parent})
+        spans.append(span)
+    return {"traceID": trace_id, "spans": spans}
+
+
+def test_span_tree_reconstruction():
+    tr = make_jaeger_trace("t1", 0, [
+        ("a", "nginx", "/compose", None),
+        ("b", "text-svc", "parse", "a"),
+        ("c", "user-db", "find", "b"),
+        ("d", "media-svc", "upload", "a"),
+    ])
+    tree = span_tree_from_jaeger(tr)
+    assert tree["component"] == "nginx"
+    assert len(tree["children"]) == 2
+    kids = {c["component"] for c in tree["children"]}
+    assert kids == {"text-svc", "media-svc"}
+    text = next(c for c in tree["children"] if c["component"] == "text-svc")
+    assert text["children"][0]["component"] == "user-db"
+
+
+def test_collector_windows_and_contract():
+    col = Collector(window_sec=5.0, t0=0.0)
+    # traces in windows 0 and 2
+    col.add_traces([
+        make_jaeger_trace("t1", int(1e6), [("a", "nginx", "/x", None)]),
+        make_jaeger_trace("t2", int(11e6), [("b", "nginx", "/y", None)]),
+    ])
+    col.add_metric_samples([
+        {"component": "nginx", "resource": "cpu", "value": 5.0, "timestamp": 1.0},
+        {"component": "nginx", "resource": "cpu", "value": 7.0, "timestamp": 6.0},
+        {"component": "db", "resource": "memory", "value": 100.0, "timestamp": 11.0},
+    ])
+    windows = col.windows()
+    assert len(windows) == 3
+    validate_raw_data(windows)
+    # windows feed the featurizer directly
+    data = Featurizer(use_native=False).fit_transform(windows)
+    assert data.num_windows == 3
+    assert data.num_paths == 2
+    # gauge carry-forward: window 1 keeps cpu=7, memory still 0
+    cpu = data.resources["nginx_cpu"]
+    np.testing.assert_allclose(cpu, [5.0, 7.0, 7.0])
+    mem = data.resources["db_memory"]
+    np.testing.assert_allclose(mem, [0.0, 0.0, 100.0])
+    # trace placement
+    assert data.invocations["general"].tolist() == [1, 0, 1]
+
+
+def test_collector_jaeger_process_map():
+    tr = {
+        "traceID": "t",
+        "processes": {"p1": {"serviceName": "svc-via-process"}},
+        "spans": [{
+            "traceID": "t", "spanID": "s1", "operationName": "op",
+            "startTime": 0, "processID": "p1", "references": [],
+        }],
+    }
+    tree = span_tree_from_jaeger(tr)
+    assert tree["component"] == "svc-via-process"

@@ -1,0 +1,41 @@
+import numpy as np
+import torch
+
+from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+from deeprest_amd.engine.config import DataConfig, EngineConfig, TrainConfig
+from deeprest_amd.engine.experiment import run_experiment
+from deeprest_amd.models.net import DeepRestNetConfig
+from deeprest_amd.serve.results import ResultsStore
+
+
+def test_run_experiment_produces_results_schema(tmp_path):
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=3, n_components=4, windows_per_day=80, n_days=2, seed=55))
+    data = app.generate_featurized()
+    cfg = EngineConfig()
+    cfg.data = DataConfig(step_size=20, split=0.4)
+    cfg.train = TrainConfig(epochs=1, batch_size=8, baseline_epochs=2,
+                            eval_cycles=3, log_every=0)
+    cfg.model = DeepRestNetConfig(d_model=32, n_heads=4, n_layers=1, d_ff=64,
+                                  hidden=16, comp_dim=8, dropout=0.0)
+
+    calls = [app.traffic_plan()[:, i] for i in range(3)]
+    store = run_experiment(data, "synthetic-waves_seen-1x", config=cfg,
+                           device=torch.device("cpu"), calls_series=calls)
+    assert store.experiments() == ["synthetic-waves_seen-1x"]
+    # every (component, resource) of the app appears
+    comps = store.results["synthetic-waves_seen-1x"]
+    assert set(comps.keys()) == set(app.all_components)
+    entry = comps[app.all_components[0]]["cpu"]
+    for est in ("bl-resrc", "bl-api", "bl-trace", "ours"):
+        assert f"prediction_{est}" in entry
+        assert f"scale_{est}" in entry
+        assert all(np.isfinite(entry[f"prediction_{est}"]))
+    assert len(entry["calls"]) == 3
+    assert "scale_groundtruth" in entry
+
+    # persists in the reference's on-disk format
+    p = str(tmp_path / "results.pkl")
+    store.save(p)
+    loaded = ResultsStore.load(p)
+    assert loaded.get("synthetic-waves_seen-1x", app.all_components[0], "cpu")
