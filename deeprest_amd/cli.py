@@ -1,0 +1,155 @@
+"""Command-line interface.
+
+Replaces the reference's edit-the-constants workflow
+(reference: resource-estimation/estimate.py:13-18)
+with subcommands over the typed config:
+
+  python -m deeprest_amd.cli featurize --raw raw_data.pkl --out input.pkl
+  python -m deeprest_amd.cli train [--config cfg.yaml] [--set train.epochs=10]
+  python -m deeprest_amd.cli experiment --name exp1 --out results.pkl
+  python -m deeprest_amd.cli synthesize --raw raw_data.pkl --list-apis
+  python -m deeprest_amd.cli serve --checkpoint ckpt.pt --port 2021
+"""
+
+from __future__ import annotations
+
+import argparse
+import sys
+
+import numpy as np
+
+
+def _load_config(args):
+    from .engine.config import EngineConfig, apply_cli_overrides
+
+    cfg = EngineConfig.load(args.config) if args.config else EngineConfig()
+    if args.set:
+        cfg = apply_cli_overrides(cfg, args.set)
+    return cfg
+
+
+def _load_data(cfg):
+    from .data.featurize import FeaturizedData, Featurizer
+    from .data.contract import load_raw_data
+    from .data.synthetic import SyntheticApp, SyntheticAppConfig
+
+    d = cfg.data
+    if d.input_path:
+        return FeaturizedData.load(d.input_path)
+    if d.raw_path:
+        return Featurizer().fit_transform(load_raw_data(d.raw_path))
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=d.synth_apis, n_components=d.synth_components,
+        windows_per_day=d.synth_windows_per_day, n_days=d.synth_days,
+        seed=d.synth_seed,
+    ))
+    print(f"[deeprest-amd] no input given; synthetic app "
+          f"({d.synth_apis} endpoints, {d.synth_components} components)")
+    return app.generate_featurized()
+
+
+def cmd_featurize(args):
+    from .data.contract import load_raw_data
+    from .data.featurize import Featurizer
+
+    raw = load_raw_data(args.raw)
+    data = Featurizer().fit_transform(raw)
+    data.save(args.out)
+    print(f"featurized {data.num_windows} windows -> {data.num_paths} call paths, "
+          f"{len(data.metric_names)} metrics; saved {args.out}")
+    return 0
+
+
+def cmd_train(args):
+    import torch
+
+    from .engine.trainer import Trainer
+
+    cfg = _load_config(args)
+    data = _load_data(cfg)
+    trainer = Trainer(data, cfg)
+    result = trainer.train()
+    print(result.summary())
+    print(f"samples/sec: {result.samples_per_sec:.1f}")
+    if cfg.train.checkpoint_path:
+        print(f"checkpoint: {cfg.train.checkpoint_path}")
+    return 0
+
+
+def cmd_experiment(args):
+    from .engine.experiment import run_experiment
+
+    cfg = _load_config(args)
+    data = _load_data(cfg)
+    store = run_experiment(data, args.name, config=cfg)
+    store.save(args.out)
+    print(f"experiment '{args.name}' written to {args.out}")
+    return 0
+
+
+def cmd_synthesize(args):
+    from .data.contract import load_raw_data
+    from .data.synthesizer import TraceSynthesizer
+
+    raw = load_raw_data(args.raw)
+    syn = TraceSynthesizer().fit(raw)
+    print(f"{len(syn.apis)} API endpoints found:")
+    for api in syn.apis:
+        print(f"    > {api}")
+    if args.api:
+        calls = {args.api: args.count}
+        vec = syn.synthesize(calls, rng=np.random.default_rng(args.seed))
+        print(f"feature vector for {calls}: {vec.tolist()}")
+    return 0
+
+
+def cmd_serve(args):
+    import uvicorn
+
+    from .serve.api import create_app
+
+    app = create_app(checkpoint_path=args.checkpoint)
+    uvicorn.run(app, host=args.host, port=args.port)
+    return 0
+
+
+def main(argv=None):
+    p = argparse.ArgumentParser(prog="deeprest-amd")
+    sub = p.add_subparsers(dest="cmd", required=True)
+
+    f = sub.add_parser("featurize")
+    f.add_argument("--raw", required=True)
+    f.add_argument("--out", default="input.pkl")
+
+    for name in ("train", "experiment"):
+        t = sub.add_parser(name)
+        t.add_argument("--config", default=None)
+        t.add_argument("--set", nargs="*", default=[])
+        if name == "experiment":
+            t.add_argument("--name", required=True)
+            t.add_argument("--out", default="results.pkl")
+
+    s = sub.add_parser("synthesize")
+    s.add_argument("--raw", required=True)
+    s.add_argument("--api", default=None)
+    s.add_argument("--count", type=int, default=10)
+    s.add_argument("--seed", type=int, default=0)
+    s.add_argument("--list-apis", action="store_true")
+
+    v = sub.add_parser("serve")
+    v.add_argument("--checkpoint", default=None)
+    v.add_argument("--host", default="0.0.0.0")
+    v.add_argument("--port", type=int, default=2021)
+
+    args = p.parse_args(argv)
+    return {
+        "featurize": cmd_featurize,
+        "train": cmd_train,
+        "experiment": cmd_experiment,
+        "synthesize": cmd_synthesize,
+        "serve": cmd_serve,
+    }[args.cmd](args)
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -19,22 +19,26 @@ class _BigKLinear(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias):
         # x: (B, N, K); weight: (O, K); bias: (O,) or None
-        out = torch.matmul(x, weight.t())
+        w = weight.to(x.dtype)
+        out = torch.matmul(x, w.t())
         if bias is not None:
-            out = out + bias
-        ctx.save_for_backward(x, weight)
+            out = out + bias.to(out.dtype)
+        ctx.save_for_backward(x, w)
         ctx.has_bias = bias is not None
+        ctx.w_dtype = weight.dtype
+        ctx.b_dtype = bias.dtype if bias is not None else None
         return out
 
     @staticmethod
     def backward(ctx, grad):
-        x, weight = ctx.saved_tensors
-        grad = grad.contiguous()
-        dx = torch.matmul(grad, weight)
+        x, w = ctx.saved_tensors
+        g = grad.contiguous().to(x.dtype)
+        dx = torch.matmul(g, w)
         # dW via leading-dim-batched GEMMs, fp32 accumulate of the partials
-        dw = torch.bmm(grad.transpose(1, 2), x).sum(0, dtype=torch.float32)
+        dw = torch.bmm(g.transpose(1, 2), x).sum(0, dtype=torch.float32)
         db = grad.sum(dim=(0, 1), dtype=torch.float32) if ctx.has_bias else None
-        return dx, dw.to(weight.dtype), db.to(weight.dtype) if db is not None else None
+        return (dx, dw.to(ctx.w_dtype),
+                db.to(ctx.b_dtype) if db is not None else None)
 
 
 def bigk_linear(x: torch.Tensor, weight: torch.Tensor,

@@ -71,3 +71,24 @@ def test_full_state_roundtrip():
     model2.eval()
     with torch.no_grad():
         torch.testing.assert_close(model(x), model2(x))
+
+
+def test_bigk_linear_matches_nn_linear():
+    from deeprest_amd.ops.linear_bigk import bigk_linear
+
+    torch.manual_seed(9)
+    x = torch.randn(3, 5, 7, 16, requires_grad=True)
+    lin = torch.nn.Linear(16, 9)
+    x2 = x.detach().clone().requires_grad_(True)
+    out_a = bigk_linear(x, lin.weight, lin.bias)
+    out_b = lin(x2)
+    torch.testing.assert_close(out_a, out_b, rtol=1e-5, atol=1e-6)
+    g = torch.randn_like(out_a)
+    out_a.backward(g)
+    wa, ba = lin.weight.grad.clone(), lin.bias.grad.clone()
+    lin.weight.grad = None
+    lin.bias.grad = None
+    out_b.backward(g)
+    torch.testing.assert_close(x.grad, x2.grad, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(wa, lin.weight.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(ba, lin.bias.grad, rtol=1e-4, atol=1e-5)
