@@ -346,6 +346,84 @@ class DeepRestNet(nn.Module):
     def loss(self, outputs: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
         return pinball_loss(outputs, labels, self.cfg.quantiles)
 
+    # -------------------------------------------------- long-horizon path
+    @torch.no_grad()
+    def forward_long(self, traffic: torch.Tensor, chunk_size: int = 4096) -> torch.Tensor:
+        """Streaming inference over arbitrarily long horizons (SURVEY.md 5.7,
+        BASELINE config 5: 7-day@1s ~ 604,800 steps).
+
+        The encoder attends within chunks (attention cost stays
+        O(chunk^2) per chunk); the GRU decoders carry hidden state across
+        chunk boundaries — a forward sweep left->right and, when
+        bidirectional, a reverse sweep right->left — so recurrence is exact
+        over the whole horizon while peak memory is O(chunk).
+
+        traffic: (B, T_total, P) -> (B, T_total, M, Q)
+        """
+        B, T_total, P = traffic.shape
+        cfg = self.cfg
+        dec = self.decoder
+        comp = self.graph()
+        C = comp.shape[0]
+        H = dec.hidden
+        dev, dt = traffic.device, traffic.dtype
+
+        bounds = list(range(0, T_total, chunk_size)) + [T_total]
+        chunks = list(zip(bounds[:-1], bounds[1:]))
+
+        def encode(x):
+            x = self.in_proj(x)
+            x = x + self._pos_encoding(x.shape[1], x.device, x.dtype)
+            x = self.in_norm(x)
+            for layer in self.layers:
+                x = layer(x)
+            return x
+
+        # forward sweep
+        h_f = torch.tanh(dec.h0_proj(comp)).unsqueeze(0).expand(B, C, H).contiguous().to(dt)
+        gamma = dec.cond_gamma(comp)
+        beta = dec.cond_beta(comp)
+        fwd_outs = []
+        enc_cache = []
+        for (s, e) in chunks:
+            enc = encode(traffic[:, s:e])
+            enc_cache.append(enc)
+            xg = dec.x_proj(enc)
+            out = fused_gru_sequence(xg, dec.w_hh, dec.b_hh, h_f, gamma, beta,
+                                     reverse=False)
+            h_f = out[:, -1].contiguous()
+            fwd_outs.append(out)
+
+        if cfg.bidirectional:
+            gamma_r = dec.cond_gamma_r(comp)
+            beta_r = dec.cond_beta_r(comp)
+            h_r = torch.tanh(dec.h0_proj_r(comp)).unsqueeze(0).expand(B, C, H).contiguous().to(dt)
+            rev_outs = [None] * len(chunks)
+            for ci in range(len(chunks) - 1, -1, -1):
+                xg_r = dec.x_proj_r(enc_cache[ci])
+                out_r = fused_gru_sequence(xg_r, dec.w_hh_r, dec.b_hh_r, h_r,
+                                           gamma_r, beta_r, reverse=True)
+                h_r = out_r[:, 0].contiguous()
+                rev_outs[ci] = out_r
+            h_chunks = [torch.cat([f, r], dim=-1) for f, r in zip(fwd_outs, rev_outs)]
+        else:
+            h_chunks = fwd_outs
+
+        w_all = torch.cat([h.weight for h in self.heads], dim=0)
+        b_all = torch.cat([h.bias for h in self.heads], dim=0)
+        R = len(self.heads)
+        Q = len(cfg.quantiles)
+        preds = []
+        for h_all in h_chunks:
+            Tc = h_all.shape[1]
+            outs = bigk_linear(h_all, w_all, b_all).view(B, Tc, C, R, Q)
+            if self._gather_is_reshape:
+                pc = outs.reshape(B, Tc, C * R, Q)
+            else:
+                pc = outs[:, :, self.comp_of, self.res_of, :]
+            preds.append(pc + self.metric_bias)
+        return torch.cat(preds, dim=1)
+
     # ---- checkpoint helpers (spec travels with the weights) ----
     def full_state(self) -> dict:
         return {
