@@ -246,8 +246,10 @@ class _GRUDecoderBank(nn.Module):
                     b.weight.mul_(0.1)
                     b.bias.zero_()
 
-    def forward(self, enc: torch.Tensor, comp: torch.Tensor) -> torch.Tensor:
-        """enc: (B, T, D); comp: (C, comp_dim) -> (B, T, C, H*dirs)."""
+    def forward(self, enc: torch.Tensor, comp: torch.Tensor):
+        """enc: (B, T, D); comp: (C, comp_dim) -> tuple of direction outputs
+        (each (B, T, C, H)); the heads consume the halves separately so no
+        concat copy is ever materialized."""
         B = enc.shape[0]
         C = comp.shape[0]
         xg = self.x_proj(enc)                                  # (B, T, 3H)
@@ -257,14 +259,14 @@ class _GRUDecoderBank(nn.Module):
         out = fused_gru_sequence(xg, self.w_hh, self.b_hh, h0.contiguous(),
                                  gamma, beta, reverse=False)
         if not self.bidirectional:
-            return out
+            return (out,)
         xg_r = self.x_proj_r(enc)
         gamma_r = self.cond_gamma_r(comp)
         beta_r = self.cond_beta_r(comp)
         h0_r = torch.tanh(self.h0_proj_r(comp)).unsqueeze(0).expand(B, C, self.hidden)
         out_r = fused_gru_sequence(xg_r, self.w_hh_r, self.b_hh_r, h0_r.contiguous(),
                                    gamma_r, beta_r, reverse=True)
-        return torch.cat([out, out_r], dim=-1)                 # (B, T, C, 2H)
+        return (out, out_r)
 
 
 # -------------------------------------------------------------------- model
@@ -284,8 +286,11 @@ class DeepRestNet(nn.Module):
 
         dirs = 2 if cfg.bidirectional else 1
         Q = len(cfg.quantiles)
+        # bias=False: metric_bias below subsumes it, and the per-head bias
+        # gradient was a pathological 28M->9 reduction on the GPU
         self.heads = nn.ModuleList(
-            [nn.Linear(cfg.hidden * dirs, Q) for _ in range(len(spec.resources))]
+            [nn.Linear(cfg.hidden * dirs, Q, bias=False)
+             for _ in range(len(spec.resources))]
         )
         self.metric_bias = nn.Parameter(torch.zeros(spec.num_metrics, Q))
         self.register_buffer("comp_of", torch.tensor(spec.comp_of, dtype=torch.long))
@@ -326,17 +331,23 @@ class DeepRestNet(nn.Module):
         for layer in self.layers:
             x = layer(x)
         comp = self.graph()                                   # (C, comp_dim)
-        h_all = self.decoder(x, comp)                         # (B, T, C, H*)
-        h_all = self.dropout(h_all)
-        # per-resource-type heads over all components, as ONE packed GEMM with
-        # a batched-K backward (ops/linear_bigk.py)
-        w_all = torch.cat([h.weight for h in self.heads], dim=0)   # (R*Q, H*)
-        b_all = torch.cat([h.bias for h in self.heads], dim=0)
-        B, T = h_all.shape[0], h_all.shape[1]
-        C = h_all.shape[2]
+        h_dirs = self.decoder(x, comp)                        # tuple of (B,T,C,H)
+        return self._apply_heads(h_dirs)
+
+    def _apply_heads(self, h_dirs) -> torch.Tensor:
+        """Per-resource heads as a sum of per-direction GEMMs (no concat),
+        each with a batched-K backward (ops/linear_bigk.py)."""
+        H = self.cfg.hidden
+        w_all = torch.cat([h.weight for h in self.heads], dim=0)   # (R*Q, H*dirs)
+        out = None
+        for d, h_d in enumerate(h_dirs):
+            h_d = self.dropout(h_d)
+            part = bigk_linear(h_d, w_all[:, d * H : (d + 1) * H].contiguous())
+            out = part if out is None else out + part
+        B, T, C = h_dirs[0].shape[0], h_dirs[0].shape[1], h_dirs[0].shape[2]
         R = len(self.heads)
         Q = len(self.cfg.quantiles)
-        outs = bigk_linear(h_all, w_all, b_all).view(B, T, C, R, Q)
+        outs = out.view(B, T, C, R, Q)
         if self._gather_is_reshape:
             preds = outs.reshape(B, T, C * R, Q)
         else:
@@ -405,23 +416,11 @@ class DeepRestNet(nn.Module):
                                            gamma_r, beta_r, reverse=True)
                 h_r = out_r[:, 0].contiguous()
                 rev_outs[ci] = out_r
-            h_chunks = [torch.cat([f, r], dim=-1) for f, r in zip(fwd_outs, rev_outs)]
+            h_chunks = list(zip(fwd_outs, rev_outs))
         else:
-            h_chunks = fwd_outs
+            h_chunks = [(f,) for f in fwd_outs]
 
-        w_all = torch.cat([h.weight for h in self.heads], dim=0)
-        b_all = torch.cat([h.bias for h in self.heads], dim=0)
-        R = len(self.heads)
-        Q = len(cfg.quantiles)
-        preds = []
-        for h_all in h_chunks:
-            Tc = h_all.shape[1]
-            outs = bigk_linear(h_all, w_all, b_all).view(B, Tc, C, R, Q)
-            if self._gather_is_reshape:
-                pc = outs.reshape(B, Tc, C * R, Q)
-            else:
-                pc = outs[:, :, self.comp_of, self.res_of, :]
-            preds.append(pc + self.metric_bias)
+        preds = [self._apply_heads(h_dirs) for h_dirs in h_chunks]
         return torch.cat(preds, dim=1)
 
     # ---- checkpoint helpers (spec travels with the weights) ----
