@@ -244,7 +244,10 @@ std::vector<at::Tensor> mha_forward(at::Tensor q, at::Tensor k, at::Tensor v,
 
 }  // namespace
 
+void register_featurize(py::module_& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  register_featurize(m);
   m.def("layer_norm_forward", &layer_norm_forward);
   m.def("layer_norm_backward", &layer_norm_backward);
   m.def("pinball_forward", &pinball_forward);

@@ -110,20 +110,30 @@ py::object featurize_walk(py::list windows, py::list seed_paths, bool grow) {
       }
     }
     // fill in the textual form of newly discovered paths by walking the trie
+    // (iterative — deep path chains must not overflow the C++ stack)
     {
       std::vector<std::pair<std::string, std::string>> chain;
-      std::function<void(TrieNode*)> walk = [&](TrieNode* node) {
-        for (auto& kv : node->children) {
-          const std::string& key = kv.first;
-          size_t z = key.find('\0');
-          chain.emplace_back(key.substr(0, z), key.substr(z + 1));
-          TrieNode* child = kv.second.get();
-          if (child->index >= n_seed) all_paths[child->index] = chain;
-          walk(child);
-          chain.pop_back();
-        }
+      struct WFrame {
+        TrieNode* node;
+        std::unordered_map<std::string, std::unique_ptr<TrieNode>>::iterator it;
       };
-      walk(&root);
+      std::vector<WFrame> wstack;
+      wstack.push_back({&root, root.children.begin()});
+      while (!wstack.empty()) {
+        WFrame& top = wstack.back();
+        if (top.it == top.node->children.end()) {
+          wstack.pop_back();
+          if (!chain.empty()) chain.pop_back();
+          continue;
+        }
+        const std::string& key = top.it->first;
+        TrieNode* child = top.it->second.get();
+        ++top.it;
+        size_t z = key.find('\0');
+        chain.emplace_back(key.substr(0, z), key.substr(z + 1));
+        if (child->index >= n_seed) all_paths[child->index] = chain;
+        wstack.push_back({child, child->children.begin()});
+      }
     }
   }
 

@@ -19,8 +19,8 @@ CSRC = os.path.join(ROOT, "deeprest_amd", "csrc")
 
 sources = [
     os.path.join(CSRC, f)
-    for f in ["bindings.cpp", "layernorm.hip", "pinball.hip", "adam.hip",
-              "gru.hip", "attention.hip"]
+    for f in ["bindings.cpp", "featurize.cpp", "layernorm.hip", "pinball.hip",
+              "adam.hip", "gru.hip", "attention.hip"]
 ]
 
 setup(

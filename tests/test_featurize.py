@@ -89,12 +89,37 @@ def test_feature_space_state_dict_roundtrip(tiny_raw_data):
 
 
 def test_deep_trace_no_recursion_limit():
-    # 50k-deep chain — the reference's recursive DFS would blow the stack
+    # 5k-deep chain — the reference's recursive DFS would blow the stack
+    # (default recursion limit 1000); ours is iterative
+    depth = 5_000
     node = {"component": "leaf", "operation": "op", "children": []}
-    for i in range(50_000):
+    for i in range(depth):
         node = {"component": f"c{i % 7}", "operation": "op", "children": [node]}
     raw = [{"metrics": [{"component": "c0", "resource": "cpu", "value": 1.0}],
             "traces": [node]}]
     data = Featurizer(use_native=False).fit_transform(raw)
-    assert data.num_paths == 50_001
-    assert data.traffic[0].sum() == 50_001
+    assert data.num_paths == depth + 1
+    assert data.traffic[0].sum() == depth + 1
+
+
+def test_native_featurizer_parity(tiny_raw_data):
+    from deeprest_amd.data.featurize import _native_featurize
+
+    if _native_featurize() is None:
+        import pytest
+        pytest.skip("native extension not built")
+    d_py = Featurizer(use_native=False).fit_transform(tiny_raw_data)
+    d_nat = Featurizer(use_native=True).fit_transform(tiny_raw_data)
+    assert d_py.feature_space.paths == d_nat.feature_space.paths
+    np.testing.assert_array_equal(d_py.traffic, d_nat.traffic)
+    for k in d_py.invocations:
+        np.testing.assert_array_equal(d_py.invocations[k], d_nat.invocations[k])
+    # frozen-space transform must ignore unseen paths, not crash
+    unseen = [{"metrics": tiny_raw_data[0]["metrics"],
+               "traces": [{"component": "new-svc", "operation": "/new",
+                           "children": [{"component": "frontend",
+                                         "operation": "/register",
+                                         "children": []}]}]}]
+    fz = Featurizer(use_native=True, feature_space=d_nat.feature_space)
+    out = fz.transform(unseen)
+    assert out.traffic.sum() == 0  # nothing matches known paths
