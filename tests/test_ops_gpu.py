@@ -244,3 +244,34 @@ def test_model_step_on_gpu(dev):
         losses.append(loss.item())
     assert all(np.isfinite(losses))
     assert losses[-1] < losses[0]
+
+
+# --------------------------------------------------------------- predictor
+def test_predictor_hipgraph_capture(dev):
+    import numpy as np
+
+    from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+    from deeprest_amd.data.windows import MinMaxScaler, sliding_window
+    from deeprest_amd.models.net import DeepRestNet, DeepRestNetConfig, build_model_spec
+    from deeprest_amd.serve.predictor import Predictor
+
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=6, n_components=6, windows_per_day=150, n_days=1, seed=17))
+    data = app.generate_featurized()
+    spec = build_model_spec(data)
+    model = DeepRestNet(spec, DeepRestNetConfig(
+        d_model=64, n_heads=2, n_layers=1, d_ff=128, hidden=128, comp_dim=16,
+        dropout=0.0)).to(dev).eval()
+
+    x_scaler = MinMaxScaler().fit(data.traffic.astype(np.float64), 100)
+    y_scalers = [MinMaxScaler() for _ in data.metric_names]
+    pred_graph = Predictor(model, x_scaler, y_scalers, data.metric_names,
+                           device=dev, graph_batch=16, use_graph=True)
+    pred_eager = Predictor(model, x_scaler, y_scalers, data.metric_names,
+                           device=dev, use_graph=False)
+    w = sliding_window(data.traffic.astype(np.float64), 30)[:40]  # 2.5 batches
+    out_g = pred_graph.predict(w)
+    out_e = pred_eager.predict(w)
+    assert pred_graph._graph is not None, "hipGraph was not captured"
+    for name in data.metric_names:
+        np.testing.assert_allclose(out_g[name], out_e[name], rtol=1e-3, atol=1e-3)
