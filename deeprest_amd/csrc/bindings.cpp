@@ -176,17 +176,23 @@ std::vector<at::Tensor> gru_seq_forward(at::Tensor xg, at::Tensor w_hh,
   TORCH_CHECK(xg.is_contiguous() && w_hh.is_contiguous() && h0.is_contiguous() &&
               gamma.is_contiguous() && beta.is_contiguous() && b_hh.is_contiguous());
 
+  TORCH_CHECK(xg.numel() < (1LL << 31),
+              "x_gates too large for 32-bit staging offsets");
   auto h_all = at::empty({B, TT, C, H}, xg.options());
   auto saves = save ? at::empty({B, TT, C, 4 * H}, xg.options())
                     : at::empty({0}, xg.options());
-  dr_gru_fwd(xg.data_ptr(), gamma.data_ptr(), beta.data_ptr(), w_hh.data_ptr(),
+  // the GEMM streams a bf16 weight image from L2 regardless of T
+  auto w_gemm = w_hh.scalar_type() == at::kBFloat16
+                    ? w_hh
+                    : w_hh.to(at::kBFloat16);
+  dr_gru_fwd(xg.data_ptr(), gamma.data_ptr(), beta.data_ptr(), w_gemm.data_ptr(),
              b_hh.data_ptr<float>(), h0.data_ptr(), h_all.data_ptr(),
              save ? saves.data_ptr() : nullptr, B, TT, C, reverse ? 1 : 0,
              save ? 1 : 0, dt == at::kBFloat16, cur_stream());
   return {h_all, saves};
 }
 
-std::vector<at::Tensor> gru_seq_backward_kernel(at::Tensor grad_h, at::Tensor w_hh,
+std::vector<at::Tensor> gru_seq_backward_kernel(at::Tensor grad_h, at::Tensor w_img,
                                                 at::Tensor h0, at::Tensor h_all,
                                                 at::Tensor saves, bool reverse) {
   const at::cuda::CUDAGuard guard(grad_h.device());
@@ -196,11 +202,15 @@ std::vector<at::Tensor> gru_seq_backward_kernel(at::Tensor grad_h, at::Tensor w_
   TORCH_CHECK(saves.numel() == (int64_t)B * TT * C * 4 * H,
               "gru backward: saves tensor missing or wrong size (forward must "
               "run with save=true)");
+  TORCH_CHECK(w_img.scalar_type() == at::kBFloat16 &&
+                  w_img.sizes() == at::IntArrayRef({H, 3 * H}) &&
+                  w_img.is_contiguous(),
+              "w_img must be the (H, 3H) bf16 pi-permuted W image");
   auto dt = grad_h.scalar_type();
   TORCH_CHECK(grad_h.is_contiguous() && h_all.is_contiguous() && saves.is_contiguous());
   auto dpre = at::empty({B, TT, C, 4 * H}, grad_h.options());
   auto dh0 = at::empty({B, C, H}, grad_h.options().dtype(at::kFloat));
-  dr_gru_bwd(grad_h.data_ptr(), w_hh.data_ptr(), h0.data_ptr(), h_all.data_ptr(),
+  dr_gru_bwd(grad_h.data_ptr(), w_img.data_ptr(), h0.data_ptr(), h_all.data_ptr(),
              saves.data_ptr(), dpre.data_ptr(), dh0.data_ptr<float>(), B, TT, C,
              reverse ? 1 : 0, dt == at::kBFloat16, cur_stream());
   return {dpre, dh0};

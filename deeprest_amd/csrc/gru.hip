@@ -42,14 +42,17 @@ constexpr int KT = H / 32;    // K-tiles of 32 in the fwd GEMM (4)
 constexpr int NT = G3H / 16;  // N-tiles of 16 (24)
 constexpr int XG_SLOTS = 32;  // distinct batch indices a 64-row tile may span
 
-// forward LDS offsets (single dynamic region, all 16B aligned)
-constexpr int LDS_W = 0;                       // 384 x 128 bf16 swizzled (98304 B)
-constexpr int LDS_H = LDS_W + G3H * H * 2;     // 64 x 128 bf16 swizzled (16384 B)
+// forward LDS offsets (single dynamic region, all 16B aligned).
+// W is NOT staged: both kernels stream the weight image from L2 (shared by
+// every block) so LDS stays at 40 KB and occupancy is register-bound
+// (the kernels are latency-stalled at 1 wave/SIMD otherwise).
+constexpr int LDS_H = 0;                       // 64 x 128 bf16 swizzled (16384 B)
 constexpr int LDS_XG = LDS_H + ROWS * H * 2;   // 32 x 384 bf16 (24576 B)
-constexpr int LDS_FWD_TOTAL = LDS_XG + XG_SLOTS * G3H * 2;  // 139264 B
+constexpr int LDS_FWD_TOTAL = LDS_XG + XG_SLOTS * G3H * 2;  // 40960 B
 
-// backward LDS: the pi-permuted W^T image only (128 rows x 384 bf16)
-constexpr int LDS_BWD_TOTAL = H * G3H * 2;     // 98304 B
+// backward uses NO LDS: the pi-permuted W image stays in L2 (96 KB, shared
+// by every block) so occupancy is VGPR-bound (2-3 waves/SIMD) instead of
+// LDS-bound (1 wave/SIMD) — the kernel is latency-bound, occupancy hides it.
 
 // swizzled byte address inside a row-major [rows][128] bf16 tile (256 B rows)
 __device__ __forceinline__ int swz(int row, int k_elem) {
@@ -100,20 +103,48 @@ __device__ __forceinline__ void ld8<float>(const float* src, float* v) {
   *reinterpret_cast<float4*>(v + 4) = *reinterpret_cast<const float4*>(src + 4);
 }
 
+// store an 8-wide bf16 fragment to a T-typed tensor (bf16: one 16 B store)
+template <typename T>
+__device__ __forceinline__ void st_frag(T* dst, bf16x8 v);
+template <>
+__device__ __forceinline__ void st_frag<uint16_t>(uint16_t* dst, bf16x8 v) {
+  *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(&v);
+}
+template <>
+__device__ __forceinline__ void st_frag<float>(float* dst, bf16x8 v) {
+  const uint16_t* p = reinterpret_cast<const uint16_t*>(&v);
+#pragma unroll
+  for (int e = 0; e < 8; ++e) dst[e] = bf2f(p[e]);
+}
+
+// load 8 values directly as an MFMA bf16 fragment (no f32 round trip)
+template <typename T>
+__device__ __forceinline__ bf16x8 ld_frag(const T* src);
+template <>
+__device__ __forceinline__ bf16x8 ld_frag<uint16_t>(const uint16_t* src) {
+  return *reinterpret_cast<const bf16x8*>(src);
+}
+template <>
+__device__ __forceinline__ bf16x8 ld_frag<float>(const float* src) {
+  uint16_t p[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) p[e] = f2bf(src[e]);
+  return *reinterpret_cast<bf16x8*>(p);
+}
+
 // ---------------------------------------------------------------- forward
 template <typename T, bool SAVE>
-__global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
+__global__ __launch_bounds__(THREADS, 2) void gru_fwd_kernel(
     const T* __restrict__ xg,      // (B, TT, 3H)
     const T* __restrict__ gamma,   // (C, 3H)
     const T* __restrict__ beta,    // (C, 3H)
-    const T* __restrict__ w_hh,    // (3H, H)
+    const uint16_t* __restrict__ w_gemm,  // (3H, H) bf16 GEMM weight image
     const float* __restrict__ b_hh,  // (3H,)
     const T* __restrict__ h0,      // (B, C, H)
     T* __restrict__ h_all,         // (B, TT, C, H)
     T* __restrict__ saves,         // (B, TT, C, 4H) pi layout (SAVE only)
     int B, int TT, int C, int reverse) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* Wl = smem + LDS_W;
   char* Hl = smem + LDS_H;
   char* XGl = smem + LDS_XG;
 
@@ -123,18 +154,6 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
   const int64_t R = (int64_t)B * C;
   const int64_t r0 = (int64_t)blockIdx.x * ROWS;
   const int b_lo = (int)(r0 / C);
-
-  // ---- prologue: stage W into swizzled LDS ----
-  for (int id = tid; id < G3H * (H / 8); id += THREADS) {
-    int j = id / (H / 8);
-    int blk = id % (H / 8);
-    float v[8];
-#pragma unroll
-    for (int e = 0; e < 8; ++e) v[e] = ldf(w_hh + (int64_t)j * H + blk * 8 + e);
-    uint16_t* dst = reinterpret_cast<uint16_t*>(Wl + j * 256 + ((blk ^ (j & 15)) << 4));
-#pragma unroll
-    for (int e = 0; e < 8; ++e) dst[e] = f2bf(v[e]);
-  }
 
   // ---- per-lane static geometry (C-layout of the 16x16 MFMA tile) ----
   const int c_col = lane & 15;           // col within a 16-wide N-tile
@@ -153,29 +172,12 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
     comp_of[i] = (int)(rr % C);
   }
 
-  // ---- preload T-invariant per-lane values ----
-  // gamma/beta packed as bf16 pairs in one u32 per (row, gate, tile):
-  // 96 registers instead of 192 (the fp32 form spilled at the 512-reg cap)
-  uint32_t gb[4][3][8];
+  // gamma/beta/b_hh row offsets: values re-read from L1/L2 each step
+  // (register residency for them capped occupancy at 1 wave/SIMD — latency
+  // hiding from 2 waves/SIMD is worth more than the reloads)
+  int gb_off[4];
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    const T* grow = gamma + (int64_t)comp_of[i] * G3H;
-    const T* brow = beta + (int64_t)comp_of[i] * G3H;
-#pragma unroll
-    for (int g = 0; g < 3; ++g)
-#pragma unroll
-      for (int nt = 0; nt < 8; ++nt) {
-        int col = g * H + nt * 16 + c_col;
-        uint16_t gv = live[i] ? f2bf(ldf(grow + col)) : 0;
-        uint16_t bv = live[i] ? f2bf(ldf(brow + col)) : 0;
-        gb[i][g][nt] = ((uint32_t)bv << 16) | gv;
-      }
-  }
-  float bh[3][8];
-#pragma unroll
-  for (int g = 0; g < 3; ++g)
-#pragma unroll
-    for (int nt = 0; nt < 8; ++nt) bh[g][nt] = b_hh[g * H + nt * 16 + c_col];
+  for (int i = 0; i < 4; ++i) gb_off[i] = comp_of[i] * G3H;
 
   // ---- fp32 hidden state in registers + bf16 tile in LDS ----
   float h[4][8];
@@ -188,30 +190,85 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
       h[i][nt] = v;
       *reinterpret_cast<uint16_t*>(Hl + swz(row_of[i], col)) = f2bf(v);
     }
+
+  // ---- t-invariant descriptors ----
+  // The expensive per-step address math was the runtime divisions (r / C)
+  // and 64-bit muls: precompute ROW indices once, share ONE per-step time
+  // offset, and derive addresses with shifts/adds.
+  const int t_first = reverse ? (TT - 1) : 0;
+  const int dirC = (reverse ? -1 : 1) * C;           // row-index per-t stride
+
+  // xg staging chunks owned by this thread (slot, blk fixed over t);
+  // offsets are 32-bit (binding checks xg.numel() < 2^31)
+  constexpr int MAXCH = 6;  // covers n_b <= 32 slots at 256 threads
+  int stage_soff[MAXCH];    // xg element offset at t_first
+  int stage_doff[MAXCH];    // LDS byte offset
+  int n_stage = 0;
+  {
+    int n_b = (int)(std::min<int64_t>(r0 + ROWS - 1, R - 1) / C) - b_lo + 1;
+    for (int id = tid; id < n_b * (G3H / 8) && n_stage < MAXCH; id += THREADS) {
+      int slot = id / (G3H / 8);
+      int blk = id % (G3H / 8);
+      stage_soff[n_stage] = (int)((((int64_t)(b_lo + slot) * TT) + t_first) * G3H) + blk * 8;
+      stage_doff[n_stage] = slot * (G3H * 2) + blk * 16;
+      ++n_stage;
+    }
+  }
+  // h_all cooperative store chunks (4 per thread; row index fixed over t)
+  int64_t hout_bc[4];       // (b*TT + t_first)*C + c
+  int hout_blk[4];
+  int hout_lds[4];
+  bool hout_live[4];
+#pragma unroll
+  for (int u = 0; u < 4; ++u) {
+    int id = tid + u * THREADS;             // ROWS*(H/8) = 1024 = 4*THREADS
+    int row = id / (H / 8);
+    int blk = id % (H / 8);
+    int64_t r = r0 + row;
+    hout_live[u] = r < R;
+    int64_t rr = hout_live[u] ? r : (R - 1);
+    int b = (int)(rr / C), c = (int)(rr % C);
+    hout_bc[u] = ((int64_t)b * TT + t_first) * C + c;
+    hout_blk[u] = blk * 8;
+    hout_lds[u] = row * 256 + ((blk ^ (row & 15)) << 4);
+  }
+  // per-lane-row indices for the saves stores (pi layout)
+  int64_t sv_bc[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+    sv_bc[i] = ((int64_t)b_of[i] * TT + t_first) * C + comp_of[i];
+  // per-wave A-fragment LDS offsets
+  int afrag_off[KT];
+  {
+    int arow = wv * 16 + (lane & 15);
+    int k0 = (lane >> 4) * 8;
+#pragma unroll
+    for (int kt = 0; kt < KT; ++kt) afrag_off[kt] = swz(arow, kt * 32 + k0);
+  }
+  // xg LDS row pointers per owned row
+  const uint16_t* xg_rows[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+    xg_rows[i] = reinterpret_cast<const uint16_t*>(XGl + (b_of[i] - b_lo) * (G3H * 2));
+
   __syncthreads();
 
   // ---- time loop ----
+  int stage_toff = 0;                        // += dir*G3H per step (32-bit)
+  int64_t bc_toff = 0;                       // += dirC per step (row-index units)
+  const int stage_tstep = (reverse ? -1 : 1) * G3H;
   for (int step = 0; step < TT; ++step) {
-    const int t = reverse ? (TT - 1 - step) : step;
-
     // A-fragments: this wave's 16 rows of the h tile, all 4 K-tiles
     bf16x8 afrag[KT];
-    {
-      int arow = wv * 16 + (lane & 15);
-      int k0 = (lane >> 4) * 8;
 #pragma unroll
-      for (int kt = 0; kt < KT; ++kt)
-        afrag[kt] = lds_read8(Hl, swz(arow, kt * 32 + k0));
-    }
+    for (int kt = 0; kt < KT; ++kt) afrag[kt] = lds_read8(Hl, afrag_off[kt]);
 
-    // stage xg[b_lo.., t, :] into LDS (vectorized, cooperative)
-    {
-      int n_b = (int)(std::min<int64_t>(r0 + ROWS - 1, R - 1) / C) - b_lo + 1;
-      for (int id = tid; id < n_b * (G3H / 8); id += THREADS) {
-        int slot = id / (G3H / 8);
-        int blk = id % (G3H / 8);
-        const T* src = xg + (((int64_t)(b_lo + slot) * TT) + t) * G3H + blk * 8;
-        uint16_t* dst = reinterpret_cast<uint16_t*>(XGl + slot * (G3H * 2) + blk * 16);
+    // stage xg[., t, :] into LDS (vectorized; 32-bit offset + shared t term)
+#pragma unroll
+    for (int u = 0; u < MAXCH; ++u) {        // compile-time index (rule 20)
+      if (u < n_stage) {
+        const T* src = xg + stage_soff[u] + stage_toff;
+        uint16_t* dst = reinterpret_cast<uint16_t*>(XGl + stage_doff[u]);
 #pragma unroll
         for (int e = 0; e < 8; ++e) dst[e] = f2bf(ldf(src + e));
       }
@@ -219,29 +276,33 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
     __syncthreads();
 
     // ---- MFMA: hh = h_tile @ W^T -> (64, 384), this wave's 16 rows ----
+    // B-fragments stream from the global (L1/L2-resident) W image; the
+    // opaque pointer stops LLVM from hoisting 96 loop-invariant loads
+    // into 384 registers.
+    const uint16_t* w_v = w_gemm;
+    asm volatile("" : "+v"(w_v));
     f32x4 acc[NT];
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt) {
       f32x4 a = {0.f, 0.f, 0.f, 0.f};
+      const uint16_t* wrow = w_v + (nt * 16 + c_col) * H + (lane >> 4) * 8;
 #pragma unroll
       for (int kt = 0; kt < KT; ++kt) {
-        int j = nt * 16 + c_col;                       // W row (gate col)
-        int k0 = kt * 32 + (lane >> 4) * 8;
-        bf16x8 bfrag = lds_read8(Wl, swz(j, k0));
+        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(wrow + kt * 32);
         a = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[kt], bfrag, a, 0, 0, 0);
       }
       acc[nt] = a;
     }
 
     // ---- fused gate epilogue (vectorized pi-layout saves) ----
-    const uint16_t* xg_rows[4];
-#pragma unroll
-    for (int i = 0; i < 4; ++i)
-      xg_rows[i] = reinterpret_cast<const uint16_t*>(
-          XGl + (b_of[i] - b_lo) * (G3H * 2));
-
+    const T* gm_v = gamma;
+    const T* bt_v = beta;
+    const float* bh_v = b_hh;
+    asm volatile("" : "+v"(gm_v), "+v"(bt_v), "+v"(bh_v));
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
+      const T* grow = gm_v + gb_off[i];
+      const T* brow = bt_v + gb_off[i];
       float fr[8], fz[8], fn[8], fh[8];
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
@@ -249,12 +310,12 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
         float xr = bf2f(xg_rows[i][col]);
         float xz = bf2f(xg_rows[i][H + col]);
         float xn = bf2f(xg_rows[i][2 * H + col]);
-        float g_r = xr * bf2f((uint16_t)gb[i][0][nt]) + bf2f((uint16_t)(gb[i][0][nt] >> 16));
-        float g_z = xz * bf2f((uint16_t)gb[i][1][nt]) + bf2f((uint16_t)(gb[i][1][nt] >> 16));
-        float g_n = xn * bf2f((uint16_t)gb[i][2][nt]) + bf2f((uint16_t)(gb[i][2][nt] >> 16));
-        float rp = sigmoidf_(acc[nt][i] + bh[0][nt] + g_r);
-        float zp = sigmoidf_(acc[nt + 8][i] + bh[1][nt] + g_z);
-        float hn = acc[nt + 16][i] + bh[2][nt];
+        float g_r = xr * ldf(grow + col) + ldf(brow + col);
+        float g_z = xz * ldf(grow + H + col) + ldf(brow + H + col);
+        float g_n = xn * ldf(grow + 2 * H + col) + ldf(brow + 2 * H + col);
+        float rp = sigmoidf_(acc[nt][i] + bh_v[col] + g_r);
+        float zp = sigmoidf_(acc[nt + 8][i] + bh_v[H + col] + g_z);
+        float hn = acc[nt + 16][i] + bh_v[2 * H + col];
         float nn = tanhf_(g_n + rp * hn);
         float hnew = (1.f - zp) * nn + zp * h[i][nt];
         h[i][nt] = hnew;
@@ -262,8 +323,7 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
         fr[nt] = rp; fz[nt] = zp; fn[nt] = nn; fh[nt] = hn;
       }
       if (SAVE && live[i]) {
-        T* sv = saves + (((int64_t)b_of[i] * TT + t) * C + comp_of[i]) * G4H +
-                c_col * 8;
+        T* sv = saves + (sv_bc[i] + bc_toff) * G4H + c_col * 8;
         st8(sv, fr);
         st8(sv + H, fz);
         st8(sv + 2 * H, fn);
@@ -273,18 +333,15 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
     __syncthreads();
 
     // ---- cooperative vectorized h_all store (reads the bf16 LDS tile) ----
-    for (int id = tid; id < ROWS * (H / 8); id += THREADS) {
-      int row = id / (H / 8);
-      int blk = id % (H / 8);
-      int64_t r = r0 + row;
-      if (r >= R) continue;
-      int b = (int)(r / C), c = (int)(r % C);
-      bf16x8 v = lds_read8(Hl, row * 256 + ((blk ^ (row & 15)) << 4));
-      T* dst = h_all + (((int64_t)b * TT + t) * C + c) * H + blk * 8;
-      const uint16_t* vu = reinterpret_cast<const uint16_t*>(&v);
 #pragma unroll
-      for (int e = 0; e < 8; ++e) stf(dst + e, bf2f(vu[e]));
+    for (int u = 0; u < 4; ++u) {
+      if (hout_live[u]) {
+        T* dst = h_all + (hout_bc[u] + bc_toff) * H + hout_blk[u];
+        st_frag(dst, lds_read8(Hl, hout_lds[u]));
+      }
     }
+    stage_toff += stage_tstep;
+    bc_toff += dirC;
     // barrier covers both: h-store reads done AND next step's a-frag reads
     // see the same consistent tile until the next epilogue writes it.
     __syncthreads();
@@ -300,40 +357,18 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
 template <typename T>
 __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
     const T* __restrict__ grad_h,   // (B, TT, C, H)
-    const T* __restrict__ w_hh,     // (3H, H)
+    const uint16_t* __restrict__ w_img,  // (H, 384) bf16: w_img[k][m] = W[natJ(m)][k]
     const T* __restrict__ h0,       // (B, C, H)
     const T* __restrict__ h_all,    // (B, TT, C, H)
     const T* __restrict__ saves,    // (B, TT, C, 4H) pi layout
     T* __restrict__ dpre,           // (B, TT, C, 4H) pi: dr|dz|dn|d_hhn
     float* __restrict__ dh0,        // (B, C, H)
     int B, int TT, int C, int reverse) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* WTl = smem;  // [k_h 0..127][m 0..383] bf16, 768 B rows, swizzled
-
   const int tid = threadIdx.x;
   const int wv = tid / DR_WAVE;
   const int lane = tid % DR_WAVE;
   const int64_t R = (int64_t)B * C;
   const int64_t r0 = (int64_t)blockIdx.x * ROWS;
-
-  // ---- stage the pi-permuted W^T image ----
-  // m < 256: natural j = (m>>7)*H + inv_pi(m&127)        (W_hr / W_hz rows)
-  // m >= 256: natural j = 2*H + inv_pi(m&127)            (W_hn rows)
-  for (int id = tid; id < G3H * (H / 8); id += THREADS) {
-    int j = id / (H / 8);          // natural j (0..383)
-    int kblk = id % (H / 8);
-    int g = j / H;
-    int col = j % H;
-    int m = g * H + ((col & 15) << 3) + (col >> 4);   // pi position
-    float v[8];
-#pragma unroll
-    for (int e = 0; e < 8; ++e) v[e] = ldf(w_hh + (int64_t)j * H + kblk * 8 + e);
-#pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      int k = kblk * 8 + e;
-      *reinterpret_cast<uint16_t*>(WTl + swz768(k, m)) = f2bf(v[e]);
-    }
-  }
 
   const int c_col = lane & 15;
   const int rgrp = lane >> 4;
@@ -356,37 +391,53 @@ __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
   for (int i = 0; i < 4; ++i)
 #pragma unroll
     for (int nt = 0; nt < 8; ++nt) dh_carry[i][nt] = 0.f;
-  __syncthreads();
 
+  // ---- compact addressing: per-row indices + ONE shared per-step offset ----
+  // (pointer arrays spilled; full recompute burned VALU on divisions — this
+  // derives every address from bc[i] + bc_toff with shifts/adds)
+  const int t_first = reverse ? 0 : (TT - 1);
+  const int dirC = (reverse ? 1 : -1) * C;
+  int64_t bc0[4];           // (b*TT + t_first)*C + c per owned row
+  int64_t h0_off[4];        // h0 row offset (t-invariant)
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    bc0[i] = ((int64_t)b_of[i] * TT + t_first) * C + comp_of[i];
+    h0_off[i] = ((int64_t)b_of[i] * C + comp_of[i]) * H;
+  }
+  const int arow = wv * 16 + (lane & 15);
+  const int64_t ra = (r0 + arow < R) ? (r0 + arow) : (R - 1);
+  const int ab = (int)(ra / C), ac = (int)(ra % C);
+  const int64_t abc0 = ((int64_t)ab * TT + t_first) * C + ac;
+  const int k0 = (lane >> 4) * 8;
+
+  int64_t bc_toff = 0;      // += dirC per step
   for (int step = 0; step < TT; ++step) {
-    // reversed traversal of the forward processing order
-    const int t = reverse ? step : (TT - 1 - step);
-    const int tprev = reverse ? (t + 1) : (t - 1);
-    const bool use_h0 = reverse ? (t == TT - 1) : (t == 0);
+    // the FIRST processed step is the forward pass's LAST, whose h_prev is
+    // h_all of the step before; h0 is the prev only at the LAST processed
+    // step (= forward t boundary)
+    const bool use_h0 = (step == TT - 1);
 
-    float zs[4][8];  // saved z, needed again after the MFMA phase
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      if (!live[i]) {
-#pragma unroll
-        for (int nt = 0; nt < 8; ++nt) zs[i][nt] = 0.f;
-        continue;
-      }
-      const int64_t bc = (int64_t)b_of[i] * TT;
-      const T* sv = saves + ((bc + t) * C + comp_of[i]) * G4H + c_col * 8;
+      if (!live[i]) continue;
+      const int64_t bc = bc0[i] + bc_toff;
+      const T* sv_p = saves + bc * G4H + c_col * 8;
       float rp[8], zp[8], nn[8], hn[8];
-      ld8(sv, rp);
-      ld8(sv + H, zp);
-      ld8(sv + 2 * H, nn);
-      ld8(sv + 3 * H, hn);
+      ld8(sv_p, rp);
+      ld8(sv_p + H, zp);
+      ld8(sv_p + 2 * H, nn);
+      ld8(sv_p + 3 * H, hn);
       float fdr[8], fdz[8], fdn[8], fdh[8];
+      // h_prev sits at bc + dirC rows (tprev = t+dir); h0 at the boundary.
+      // pointer select -> ONE load (the OOB address is never dereferenced)
+      const T* hsrc = use_h0 ? (h0 + h0_off[i])
+                             : (h_all + (bc + dirC) * H);
+      const T* gr_p = grad_h + bc * H;
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
         int col = nt * 16 + c_col;
-        float hp = use_h0
-            ? ldf(h0 + ((int64_t)b_of[i] * C + comp_of[i]) * H + col)
-            : ldf(h_all + ((bc + tprev) * C + comp_of[i]) * H + col);
-        float g = ldf(grad_h + ((bc + t) * C + comp_of[i]) * H + col);
+        float hp = ldf(hsrc + col);
+        float g = ldf(gr_p + col);
         float dht = g + dh_carry[i][nt];
         float dz = dht * (hp - nn[nt]);
         float dn = dht * (1.f - zp[nt]);
@@ -398,51 +449,50 @@ __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
         fdn[nt] = dnp;
         fdh[nt] = dhhn;
         dh_carry[i][nt] = dht * zp[nt];    // partial; MFMA adds dpre @ W
-        zs[i][nt] = zp[nt];
       }
-      T* dx = dpre + ((bc + t) * C + comp_of[i]) * G4H + c_col * 8;
-      st8(dx, fdr);
-      st8(dx + H, fdz);
-      st8(dx + 2 * H, fdn);
-      st8(dx + 3 * H, fdh);
+      T* dx_p = dpre + bc * G4H + c_col * 8;
+      st8(dx_p, fdr);
+      st8(dx_p + H, fdz);
+      st8(dx_p + 2 * H, fdn);
+      st8(dx_p + 3 * H, fdh);
+      // fence the scheduler: without this it interleaves all 4 row
+      // iterations and the combined live ranges spill to scratch
+      __builtin_amdgcn_sched_barrier(0);
     }
     __syncthreads();  // all dpre rows visible (same-CU L2) before A-frag reads
 
     // ---- MFMA: delta = dpre_pi (64 x 384 over m) @ W_pi, wave's 16 rows ----
     {
+      // opaque copy of the W pointer: stops LLVM from hoisting all 96
+      // loop-invariant B-fragment loads out of the t loop (384 registers ->
+      // scratch).  W streams from L1/L2 each step by design.
+      const uint16_t* w_imgv = w_img;
+      asm volatile("" : "+v"(w_imgv));
+      const T* arow_p = dpre + (abc0 + bc_toff) * G4H;
       bf16x8 afrag[12];
-      const int arow = wv * 16 + (lane & 15);
-      const int64_t ra = r0 + arow < R ? r0 + arow : R - 1;
-      const int ab = (int)(ra / C), ac = (int)(ra % C);
-      const T* arow_ptr = dpre + (((int64_t)ab * TT + t) * C + ac) * G4H;
-      const int k0 = (lane >> 4) * 8;
 #pragma unroll
       for (int kt = 0; kt < 12; ++kt) {
         int m = kt * 32 + k0;
         int off = (m < 256) ? m : (m + 128);   // skip the dn region (g == 2)
-        float tmp[8];
-        ld8(arow_ptr + off, tmp);
-        uint16_t p[8];
-#pragma unroll
-        for (int e = 0; e < 8; ++e) p[e] = f2bf(tmp[e]);
-        afrag[kt] = *reinterpret_cast<bf16x8*>(p);
+        afrag[kt] = ld_frag(arow_p + off);
       }
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
         f32x4 a = {0.f, 0.f, 0.f, 0.f};
+        const int n = nt * 16 + c_col;
+        const uint16_t* wrow = w_imgv + (int64_t)n * G3H;
 #pragma unroll
         for (int kt = 0; kt < 12; ++kt) {
-          int n = nt * 16 + c_col;                    // output h-col
-          int m0 = kt * 32 + (lane >> 4) * 8;
-          bf16x8 bfrag = lds_read8(WTl, swz768(n, m0));
+          bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(wrow + kt * 32 + k0);
           a = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[kt], bfrag, a, 0, 0, 0);
         }
 #pragma unroll
         for (int i = 0; i < 4; ++i) dh_carry[i][nt] += a[i];
       }
     }
+    bc_toff += dirC;
     // no end-of-step barrier: the next step's stores go to DIFFERENT dpre
-    // addresses (per-t storage), and WTl is read-only after the prologue.
+    // addresses (per-t storage); the W image is read-only.
   }
 
   // ---- dh0 = final carry ----
@@ -541,51 +591,34 @@ __global__ void gru_dgamma_kernel(const T* __restrict__ dpre,  // (BT, C, 4H) pi
 // ------------------------------------------------------------- launchers
 template <typename T>
 static void gru_fwd_launch_t(const void* xg, const void* gamma, const void* beta,
-                             const void* w_hh, const float* b_hh, const void* h0,
+                             const void* w_gemm, const float* b_hh, const void* h0,
                              void* h_all, void* saves, int B, int TT, int C,
                              int reverse, int save, hipStream_t stream) {
   int64_t R = (int64_t)B * C;
   int grid = (int)((R + ROWS - 1) / ROWS);
-  static bool attr_set = false;
-  if (!attr_set) {
-    DR_HIP_CHECK(hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&gru_fwd_kernel<T, true>),
-        hipFuncAttributeMaxDynamicSharedMemorySize, LDS_FWD_TOTAL));
-    DR_HIP_CHECK(hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&gru_fwd_kernel<T, false>),
-        hipFuncAttributeMaxDynamicSharedMemorySize, LDS_FWD_TOTAL));
-    attr_set = true;
-  }
   if (save)
     hipLaunchKernelGGL((gru_fwd_kernel<T, true>), dim3(grid), dim3(THREADS),
                        LDS_FWD_TOTAL, stream, (const T*)xg, (const T*)gamma,
-                       (const T*)beta, (const T*)w_hh, b_hh, (const T*)h0,
+                       (const T*)beta, (const uint16_t*)w_gemm, b_hh, (const T*)h0,
                        (T*)h_all, (T*)saves, B, TT, C, reverse);
   else
     hipLaunchKernelGGL((gru_fwd_kernel<T, false>), dim3(grid), dim3(THREADS),
                        LDS_FWD_TOTAL, stream, (const T*)xg, (const T*)gamma,
-                       (const T*)beta, (const T*)w_hh, b_hh, (const T*)h0,
+                       (const T*)beta, (const uint16_t*)w_gemm, b_hh, (const T*)h0,
                        (T*)h_all, (T*)saves, B, TT, C, reverse);
 }
 
 template <typename T>
-static void gru_bwd_launch_t(const void* grad_h, const void* w_hh, const void* h0,
+static void gru_bwd_launch_t(const void* grad_h, const void* w_img, const void* h0,
                              const void* h_all, const void* saves, void* dpre,
                              float* dh0, int B, int TT, int C, int reverse,
                              hipStream_t stream) {
   int64_t R = (int64_t)B * C;
   int grid = (int)((R + ROWS - 1) / ROWS);
-  static bool attr_set = false;
-  if (!attr_set) {
-    DR_HIP_CHECK(hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&gru_bwd_kernel<T>),
-        hipFuncAttributeMaxDynamicSharedMemorySize, LDS_BWD_TOTAL));
-    attr_set = true;
-  }
-  hipLaunchKernelGGL((gru_bwd_kernel<T>), dim3(grid), dim3(THREADS),
-                     LDS_BWD_TOTAL, stream, (const T*)grad_h, (const T*)w_hh,
-                     (const T*)h0, (const T*)h_all, (const T*)saves, (T*)dpre,
-                     dh0, B, TT, C, reverse);
+  hipLaunchKernelGGL((gru_bwd_kernel<T>), dim3(grid), dim3(THREADS), 0, stream,
+                     (const T*)grad_h, (const uint16_t*)w_img, (const T*)h0,
+                     (const T*)h_all, (const T*)saves, (T*)dpre, dh0, B, TT, C,
+                     reverse);
 }
 
 template <typename T>
@@ -601,7 +634,7 @@ static void gru_reduce_launch_t(const void* dpre, const void* gamma, const void*
   {
     int n_threads = C * 64;
     int gx = (n_threads + 255) / 256;
-    int gy = 32;  // BT slices
+    int gy = 128;  // BT slices (parallelism for small C; atomics stay cheap)
     hipLaunchKernelGGL((gru_dgamma_kernel<T>), dim3(gx, gy), dim3(256), 0, stream,
                        (const T*)dpre, (const T*)xg, dgamma, dbeta, BT, C);
   }

@@ -90,9 +90,18 @@ class _FusedGRUSequence(torch.autograd.Function):
         ext = require_native("fused_gru_sequence")
         x_gates, w_hh, h0, gamma, h_all, saves = ctx.saved_tensors
         reverse = ctx.reverse
+        # pi-permuted W image (H, 3H) the kernel streams from L2: row k holds
+        # W[natJ(m), k] over the MFMA K axis m (dr | dz | d_hhn regions)
+        H = w_hh.shape[1]
+        m = torch.arange(3 * H, device=w_hh.device)
+        g = m // H
+        q = m % H
+        col = (q % 8) * 16 + (q // 8)
+        natj = torch.where(m < 2 * H, g * H + col, 2 * H + col)
+        w_img = w_hh[natj, :].t().contiguous().to(torch.bfloat16)
         # sequential chain (custom kernel): dpre = [dr_pre|dz_pre|dn_pre|d_hh_n]
         dpre, dh0 = ext.gru_seq_backward_kernel(
-            grad_h_all.contiguous(), w_hh, h0, h_all, saves, reverse
+            grad_h_all.contiguous(), w_img, h0, h_all, saves, reverse
         )
         B, T, C, G4 = dpre.shape
         H = G4 // 4
