@@ -43,12 +43,14 @@ constexpr int NT = G3H / 16;  // N-tiles of 16 (24)
 constexpr int XG_SLOTS = 32;  // distinct batch indices a 64-row tile may span
 
 // forward LDS offsets (single dynamic region, all 16B aligned).
-// W is NOT staged: both kernels stream the weight image from L2 (shared by
-// every block) so LDS stays at 40 KB and occupancy is register-bound
-// (the kernels are latency-stalled at 1 wave/SIMD otherwise).
-constexpr int LDS_H = 0;                       // 64 x 128 bf16 swizzled (16384 B)
+// Forward stages W in LDS (an L2-streamed W was measured SLOWER for the
+// forward: 96 KB/wave/step of L2 B-fragment traffic outweighed the
+// occupancy gain); the backward streams W from L2 instead (its register
+// budget allows 2 waves/SIMD there, and it has no LDS at all).
+constexpr int LDS_W = 0;                       // 384 x 128 bf16 swizzled (98304 B)
+constexpr int LDS_H = LDS_W + G3H * H * 2;     // 64 x 128 bf16 swizzled (16384 B)
 constexpr int LDS_XG = LDS_H + ROWS * H * 2;   // 32 x 384 bf16 (24576 B)
-constexpr int LDS_FWD_TOTAL = LDS_XG + XG_SLOTS * G3H * 2;  // 40960 B
+constexpr int LDS_FWD_TOTAL = LDS_XG + XG_SLOTS * G3H * 2;  // 139264 B
 
 // backward uses NO LDS: the pi-permuted W image stays in L2 (96 KB, shared
 // by every block) so occupancy is VGPR-bound (2-3 waves/SIMD) instead of
@@ -134,7 +136,7 @@ __device__ __forceinline__ bf16x8 ld_frag<float>(const float* src) {
 
 // ---------------------------------------------------------------- forward
 template <typename T, bool SAVE>
-__global__ __launch_bounds__(THREADS, 2) void gru_fwd_kernel(
+__global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
     const T* __restrict__ xg,      // (B, TT, 3H)
     const T* __restrict__ gamma,   // (C, 3H)
     const T* __restrict__ beta,    // (C, 3H)
@@ -145,6 +147,7 @@ __global__ __launch_bounds__(THREADS, 2) void gru_fwd_kernel(
     T* __restrict__ saves,         // (B, TT, C, 4H) pi layout (SAVE only)
     int B, int TT, int C, int reverse) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* Wl = smem + LDS_W;
   char* Hl = smem + LDS_H;
   char* XGl = smem + LDS_XG;
 
@@ -154,6 +157,14 @@ __global__ __launch_bounds__(THREADS, 2) void gru_fwd_kernel(
   const int64_t R = (int64_t)B * C;
   const int64_t r0 = (int64_t)blockIdx.x * ROWS;
   const int b_lo = (int)(r0 / C);
+
+  // ---- prologue: stage the bf16 W image into swizzled LDS ----
+  for (int id = tid; id < G3H * (H / 8); id += THREADS) {
+    int j = id / (H / 8);
+    int blk = id % (H / 8);
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(w_gemm + (int64_t)j * H + blk * 8);
+    *reinterpret_cast<bf16x8*>(Wl + j * 256 + ((blk ^ (j & 15)) << 4)) = v;
+  }
 
   // ---- per-lane static geometry (C-layout of the 16x16 MFMA tile) ----
   const int c_col = lane & 15;           // col within a 16-wide N-tile
@@ -172,12 +183,30 @@ __global__ __launch_bounds__(THREADS, 2) void gru_fwd_kernel(
     comp_of[i] = (int)(rr % C);
   }
 
-  // gamma/beta/b_hh row offsets: values re-read from L1/L2 each step
-  // (register residency for them capped occupancy at 1 wave/SIMD — latency
-  // hiding from 2 waves/SIMD is worth more than the reloads)
-  int gb_off[4];
+  // ---- preload T-invariant per-lane values ----
+  // gamma/beta packed as bf16 pairs in one u32 per (row, gate, tile): the
+  // LDS budget caps this kernel at 1 block/CU regardless, so registers for
+  // these are free and beat per-step L2 reloads.
+  uint32_t gb[4][3][8];
 #pragma unroll
-  for (int i = 0; i < 4; ++i) gb_off[i] = comp_of[i] * G3H;
+  for (int i = 0; i < 4; ++i) {
+    const T* grow = gamma + (int64_t)comp_of[i] * G3H;
+    const T* brow = beta + (int64_t)comp_of[i] * G3H;
+#pragma unroll
+    for (int g = 0; g < 3; ++g)
+#pragma unroll
+      for (int nt = 0; nt < 8; ++nt) {
+        int col = g * H + nt * 16 + c_col;
+        uint16_t gv = live[i] ? f2bf(ldf(grow + col)) : 0;
+        uint16_t bv = live[i] ? f2bf(ldf(brow + col)) : 0;
+        gb[i][g][nt] = ((uint32_t)bv << 16) | gv;
+      }
+  }
+  float bh[3][8];
+#pragma unroll
+  for (int g = 0; g < 3; ++g)
+#pragma unroll
+    for (int nt = 0; nt < 8; ++nt) bh[g][nt] = b_hh[g * H + nt * 16 + c_col];
 
   // ---- fp32 hidden state in registers + bf16 tile in LDS ----
   float h[4][8];
@@ -276,33 +305,23 @@ __global__ __launch_bounds__(THREADS, 2) void gru_fwd_kernel(
     __syncthreads();
 
     // ---- MFMA: hh = h_tile @ W^T -> (64, 384), this wave's 16 rows ----
-    // B-fragments stream from the global (L1/L2-resident) W image; the
-    // opaque pointer stops LLVM from hoisting 96 loop-invariant loads
-    // into 384 registers.
-    const uint16_t* w_v = w_gemm;
-    asm volatile("" : "+v"(w_v));
     f32x4 acc[NT];
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt) {
       f32x4 a = {0.f, 0.f, 0.f, 0.f};
-      const uint16_t* wrow = w_v + (nt * 16 + c_col) * H + (lane >> 4) * 8;
 #pragma unroll
       for (int kt = 0; kt < KT; ++kt) {
-        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(wrow + kt * 32);
+        int j = nt * 16 + c_col;                       // W row (gate col)
+        int k0 = kt * 32 + (lane >> 4) * 8;
+        bf16x8 bfrag = lds_read8(Wl, swz(j, k0));
         a = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[kt], bfrag, a, 0, 0, 0);
       }
       acc[nt] = a;
     }
 
     // ---- fused gate epilogue (vectorized pi-layout saves) ----
-    const T* gm_v = gamma;
-    const T* bt_v = beta;
-    const float* bh_v = b_hh;
-    asm volatile("" : "+v"(gm_v), "+v"(bt_v), "+v"(bh_v));
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      const T* grow = gm_v + gb_off[i];
-      const T* brow = bt_v + gb_off[i];
       float fr[8], fz[8], fn[8], fh[8];
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
@@ -310,12 +329,12 @@ __global__ __launch_bounds__(THREADS, 2) void gru_fwd_kernel(
         float xr = bf2f(xg_rows[i][col]);
         float xz = bf2f(xg_rows[i][H + col]);
         float xn = bf2f(xg_rows[i][2 * H + col]);
-        float g_r = xr * ldf(grow + col) + ldf(brow + col);
-        float g_z = xz * ldf(grow + H + col) + ldf(brow + H + col);
-        float g_n = xn * ldf(grow + 2 * H + col) + ldf(brow + 2 * H + col);
-        float rp = sigmoidf_(acc[nt][i] + bh_v[col] + g_r);
-        float zp = sigmoidf_(acc[nt + 8][i] + bh_v[H + col] + g_z);
-        float hn = acc[nt + 16][i] + bh_v[2 * H + col];
+        float g_r = xr * bf2f((uint16_t)gb[i][0][nt]) + bf2f((uint16_t)(gb[i][0][nt] >> 16));
+        float g_z = xz * bf2f((uint16_t)gb[i][1][nt]) + bf2f((uint16_t)(gb[i][1][nt] >> 16));
+        float g_n = xn * bf2f((uint16_t)gb[i][2][nt]) + bf2f((uint16_t)(gb[i][2][nt] >> 16));
+        float rp = sigmoidf_(acc[nt][i] + bh[0][nt] + g_r);
+        float zp = sigmoidf_(acc[nt + 8][i] + bh[1][nt] + g_z);
+        float hn = acc[nt + 16][i] + bh[2][nt];
         float nn = tanhf_(g_n + rp * hn);
         float hnew = (1.f - zp) * nn + zp * h[i][nt];
         h[i][nt] = hnew;
@@ -596,6 +615,16 @@ static void gru_fwd_launch_t(const void* xg, const void* gamma, const void* beta
                              int reverse, int save, hipStream_t stream) {
   int64_t R = (int64_t)B * C;
   int grid = (int)((R + ROWS - 1) / ROWS);
+  static bool attr_set = false;
+  if (!attr_set) {
+    DR_HIP_CHECK(hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&gru_fwd_kernel<T, true>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, LDS_FWD_TOTAL));
+    DR_HIP_CHECK(hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&gru_fwd_kernel<T, false>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, LDS_FWD_TOTAL));
+    attr_set = true;
+  }
   if (save)
     hipLaunchKernelGGL((gru_fwd_kernel<T, true>), dim3(grid), dim3(THREADS),
                        LDS_FWD_TOTAL, stream, (const T*)xg, (const T*)gamma,
