@@ -39,7 +39,10 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--batch", type=int, default=256, help="per-GPU window batch")
     p.add_argument("--endpoints", type=int, default=256, help="API endpoints")
-    p.add_argument("--components", type=int, default=64)
+    # 63 + the frontend component = 64 -> batch*64 rows tile exactly onto the
+    # 256 CUs for the fused GRU kernels (a 260th workgroup at 1 block/CU
+    # doubles their wall time)
+    p.add_argument("--components", type=int, default=63)
     p.add_argument("--seq-len", type=int, default=60)
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--device", type=str, default=None)
