@@ -62,6 +62,14 @@ def create_app(checkpoint_path: Optional[str] = None, predictor: Optional[Predic
     if checkpoint_path and predictor is None:
         state["predictor"] = Predictor.from_checkpoint(checkpoint_path)
 
+    @app.get("/demo")
+    def demo():
+        from fastapi.responses import HTMLResponse
+
+        from .demo import DEMO_HTML
+
+        return HTMLResponse(DEMO_HTML)
+
     @app.get("/health")
     def health():
         from ..ops import native_available

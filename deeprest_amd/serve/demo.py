@@ -1,0 +1,118 @@
+"""Browser demo (the reference's web-demo/ equivalent, L8).
+
+The reference ships a Dash UI reading precomputed results.pkl
+(reference: web-demo/app.py).  Ours is a single self-contained HTML page on
+the live REST API: ingest stats, what-if traffic mix sliders per API
+endpoint, and per-metric quantile-band charts rendered with <canvas> — no
+JS build chain, no extra dependencies.  Mount with
+``app.include_router`` via serve.api.create_app (route GET /demo).
+"""
+
+DEMO_HTML = """<!DOCTYPE html>
+<html>
+<head>
+<meta charset="utf-8"/>
+<title>deeprest-amd demo</title>
+<style>
+ body { font-family: system-ui, sans-serif; margin: 2rem; background:#fafafa; }
+ h1 { font-size: 1.3rem; } h2 { font-size: 1.05rem; margin-top: 1.5rem; }
+ .row { display: flex; gap: 2rem; flex-wrap: wrap; }
+ .card { background: white; border: 1px solid #ddd; border-radius: 8px;
+         padding: 1rem; margin: .5rem 0; }
+ canvas { border: 1px solid #eee; background: white; }
+ input[type=number] { width: 5rem; }
+ #status { color: #666; font-size: .9rem; }
+ button { padding: .4rem .9rem; border-radius: 6px; border: 1px solid #888;
+          background: #eee; cursor: pointer; }
+</style>
+</head>
+<body>
+<h1>deeprest-amd &mdash; per-endpoint resource estimation (MI355X)</h1>
+<div id="status">loading&hellip;</div>
+
+<div class="card">
+  <h2>1 &middot; What-if traffic mix (calls per window)</h2>
+  <div id="apis"></div>
+  <p>
+    windows: <input id="nwin" type="number" value="120"/>
+    <button onclick="estimate()">Estimate</button>
+  </p>
+</div>
+
+<div class="card">
+  <h2>2 &middot; Predicted utilization (q05 / median / q95 band)</h2>
+  <div id="charts" class="row"></div>
+</div>
+
+<script>
+let apis = [];
+async function init() {
+  const h = await (await fetch('health')).json();
+  let msg = 'model loaded: ' + h.model_loaded + ' | native HIP extension: ' + h.native_extension;
+  try {
+    const a = await (await fetch('apis')).json();
+    apis = a.apis;
+  } catch (e) { msg += ' | featurize first (POST /ingest then /featurize)'; }
+  document.getElementById('status').textContent = msg;
+  const box = document.getElementById('apis');
+  box.innerHTML = apis.map((a, i) =>
+    `<label style="display:inline-block;margin:.2rem .8rem .2rem 0">${a}
+     <input type="number" id="api${i}" value="${i < 2 ? 20 : 0}"/></label>`).join('');
+}
+function drawBand(canvas, q05, q50, q95, title) {
+  const ctx = canvas.getContext('2d');
+  const W = canvas.width, H = canvas.height, pad = 28;
+  ctx.clearRect(0, 0, W, H);
+  const all = q05.concat(q95);
+  const lo = Math.min(...all), hi = Math.max(...all) || 1;
+  const x = i => pad + (W - 2 * pad) * i / (q50.length - 1 || 1);
+  const y = v => H - pad - (H - 2 * pad) * (v - lo) / (hi - lo || 1);
+  ctx.fillStyle = 'rgba(70,130,220,0.25)';
+  ctx.beginPath();
+  q95.forEach((v, i) => i ? ctx.lineTo(x(i), y(v)) : ctx.moveTo(x(i), y(v)));
+  for (let i = q05.length - 1; i >= 0; i--) ctx.lineTo(x(i), y(q05[i]));
+  ctx.closePath(); ctx.fill();
+  ctx.strokeStyle = 'rgb(30,80,180)'; ctx.lineWidth = 1.5;
+  ctx.beginPath();
+  q50.forEach((v, i) => i ? ctx.lineTo(x(i), y(v)) : ctx.moveTo(x(i), y(v)));
+  ctx.stroke();
+  ctx.fillStyle = '#333'; ctx.font = '11px sans-serif';
+  ctx.fillText(title, 6, 13);
+  ctx.fillText(hi.toFixed(1), 2, pad); ctx.fillText(lo.toFixed(1), 2, H - pad);
+}
+async function estimate() {
+  const plan = {};
+  apis.forEach((a, i) => {
+    const v = parseInt(document.getElementById('api' + i).value) || 0;
+    if (v > 0) plan[a] = v;
+  });
+  const n = parseInt(document.getElementById('nwin').value) || 120;
+  document.getElementById('status').textContent = 'estimating…';
+  const r = await fetch('estimate', {
+    method: 'POST', headers: {'Content-Type': 'application/json'},
+    body: JSON.stringify({traffic_plan: Array(n).fill(plan), seed: 0}),
+  });
+  if (!r.ok) {
+    document.getElementById('status').textContent = 'error: ' + (await r.text());
+    return;
+  }
+  const body = await r.json();
+  const charts = document.getElementById('charts');
+  charts.innerHTML = '';
+  const names = Object.keys(body.predictions).slice(0, 24);
+  for (const name of names) {
+    const w = body.predictions[name][0];   // first window: (T, 3)
+    const c = document.createElement('canvas');
+    c.width = 300; c.height = 150;
+    charts.appendChild(c);
+    drawBand(c, w.map(q => q[0]), w.map(q => q[1]), w.map(q => q[2]), name);
+  }
+  document.getElementById('status').textContent =
+    'estimated ' + names.length + ' metrics for mix ' + JSON.stringify(
+      Object.fromEntries(Object.entries(body.predictions).slice(0,0))) +
+    ' — showing first window quantile bands';
+}
+init();
+</script>
+</body>
+</html>"""

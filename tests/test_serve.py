@@ -143,3 +143,14 @@ def test_rest_api_end_to_end(trained):
     r = client.post("/anomaly", json={"measured": measured, "predicted": predicted})
     assert r.status_code == 200
     assert r.json()[data.metric_names[0]]["anomalous"] is True
+
+
+def test_demo_page_served(trained):
+    from starlette.testclient import TestClient
+
+    from deeprest_amd.serve.api import create_app
+
+    client = TestClient(create_app())
+    r = client.get("/demo")
+    assert r.status_code == 200
+    assert "deeprest-amd" in r.text and "<canvas" in r.text
