@@ -28,12 +28,15 @@ __device__ __forceinline__ float bf2f(uint16_t x) {
 }
 
 __device__ __forceinline__ uint16_t f2bf(float f) {
-  union { uint32_t u; float f; } v;
-  v.f = f;
-  // round-to-nearest-even on the truncated mantissa
-  uint32_t lsb = (v.u >> 16) & 1u;
-  v.u += 0x7fffu + lsb;
-  return static_cast<uint16_t>(v.u >> 16);
+  // single v_cvt_pk_bf16_f32 (RNE) — the manual add/shift form costs 3 VALU
+  __hip_bfloat16 b = __float2bfloat16(f);
+  return *reinterpret_cast<uint16_t*>(&b);
+}
+
+// two f32 -> packed 2x bf16 in one v_cvt_pk_bf16_f32
+__device__ __forceinline__ uint32_t f2bf2(float lo, float hi) {
+  __hip_bfloat162 b2 = __float22bfloat162_rn(make_float2(lo, hi));
+  return *reinterpret_cast<uint32_t*>(&b2);
 }
 
 // generic load/store as float for T in {float, bf16-as-ushort}

@@ -79,10 +79,12 @@ template <typename T>
 __device__ __forceinline__ void st8(T* dst, const float* v);
 template <>
 __device__ __forceinline__ void st8<uint16_t>(uint16_t* dst, const float* v) {
-  uint16_t p[8];
-#pragma unroll
-  for (int e = 0; e < 8; ++e) p[e] = f2bf(v[e]);
-  *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(p);
+  uint4 p;
+  p.x = f2bf2(v[0], v[1]);
+  p.y = f2bf2(v[2], v[3]);
+  p.z = f2bf2(v[4], v[5]);
+  p.w = f2bf2(v[6], v[7]);
+  *reinterpret_cast<uint4*>(dst) = p;
 }
 template <>
 __device__ __forceinline__ void st8<float>(float* dst, const float* v) {
@@ -117,6 +119,20 @@ __device__ __forceinline__ void st_frag<float>(float* dst, bf16x8 v) {
   const uint16_t* p = reinterpret_cast<const uint16_t*>(&v);
 #pragma unroll
   for (int e = 0; e < 8; ++e) dst[e] = bf2f(p[e]);
+}
+
+// load 8 T values as raw bf16 bit patterns (bf16: one 16 B load)
+template <typename T>
+__device__ __forceinline__ void ld_raw8(const T* src, uint16_t* out);
+template <>
+__device__ __forceinline__ void ld_raw8<uint16_t>(const uint16_t* src, uint16_t* out) {
+  uint4 raw = *reinterpret_cast<const uint4*>(src);
+  *reinterpret_cast<uint4*>(out) = raw;
+}
+template <>
+__device__ __forceinline__ void ld_raw8<float>(const float* src, uint16_t* out) {
+#pragma unroll
+  for (int e = 0; e < 8; ++e) out[e] = f2bf(src[e]);
 }
 
 // load 8 values directly as an MFMA bf16 fragment (no f32 round trip)
@@ -231,15 +247,19 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
   // offsets are 32-bit (binding checks xg.numel() < 2^31)
   constexpr int MAXCH = 6;  // covers n_b <= 32 slots at 256 threads
   int stage_soff[MAXCH];    // xg element offset at t_first
-  int stage_doff[MAXCH];    // LDS byte offset
+  int stage_pi0[MAXCH];     // pi-layout LDS element offset of the chunk
   int n_stage = 0;
   {
     int n_b = (int)(std::min<int64_t>(r0 + ROWS - 1, R - 1) / C) - b_lo + 1;
     for (int id = tid; id < n_b * (G3H / 8) && n_stage < MAXCH; id += THREADS) {
       int slot = id / (G3H / 8);
       int blk = id % (G3H / 8);
-      stage_soff[n_stage] = (int)((((int64_t)(b_lo + slot) * TT) + t_first) * G3H) + blk * 8;
-      stage_doff[n_stage] = slot * (G3H * 2) + blk * 16;
+      int j0 = blk * 8;
+      int g = j0 >> 7;
+      int col0 = j0 & 127;
+      stage_soff[n_stage] = (int)((((int64_t)(b_lo + slot) * TT) + t_first) * G3H) + j0;
+      stage_pi0[n_stage] =
+          slot * G3H + g * H + ((col0 & 15) << 3) + (col0 >> 4);
       ++n_stage;
     }
   }
@@ -274,11 +294,12 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
 #pragma unroll
     for (int kt = 0; kt < KT; ++kt) afrag_off[kt] = swz(arow, kt * 32 + k0);
   }
-  // xg LDS row pointers per owned row
+  // xg LDS row bases (pi layout, element-indexed)
   const uint16_t* xg_rows[4];
 #pragma unroll
   for (int i = 0; i < 4; ++i)
-    xg_rows[i] = reinterpret_cast<const uint16_t*>(XGl + (b_of[i] - b_lo) * (G3H * 2));
+    xg_rows[i] = reinterpret_cast<const uint16_t*>(XGl) +
+                 (b_of[i] - b_lo) * G3H + c_col * 8;
 
   __syncthreads();
 
@@ -292,14 +313,18 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
 #pragma unroll
     for (int kt = 0; kt < KT; ++kt) afrag[kt] = lds_read8(Hl, afrag_off[kt]);
 
-    // stage xg[., t, :] into LDS (vectorized; 32-bit offset + shared t term)
+    // stage xg[., t, :] into LDS in pi layout: the 8 loaded naturals land at
+    // positions pi_0 + 8e (16 B stride), so the epilogue reads each (row,
+    // gate) slice as ONE ds_read_b128 instead of 8 scalar reads.
 #pragma unroll
     for (int u = 0; u < MAXCH; ++u) {        // compile-time index (rule 20)
       if (u < n_stage) {
         const T* src = xg + stage_soff[u] + stage_toff;
-        uint16_t* dst = reinterpret_cast<uint16_t*>(XGl + stage_doff[u]);
+        uint16_t* dst = reinterpret_cast<uint16_t*>(XGl) + stage_pi0[u];
+        uint16_t vals[8];
+        ld_raw8(src, vals);
 #pragma unroll
-        for (int e = 0; e < 8; ++e) dst[e] = f2bf(ldf(src + e));
+        for (int e = 0; e < 8; ++e) dst[e * 8] = vals[e];
       }
     }
     __syncthreads();
@@ -325,10 +350,12 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
       float fr[8], fz[8], fn[8], fh[8];
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
+        // pi layout: per-gate values sit at CONSECUTIVE LDS addresses over
+        // nt, so these scalar reads coalesce/merge cheaply
+        float xr = bf2f(xg_rows[i][nt]);
+        float xz = bf2f(xg_rows[i][H + nt]);
+        float xn = bf2f(xg_rows[i][2 * H + nt]);
         int col = nt * 16 + c_col;
-        float xr = bf2f(xg_rows[i][col]);
-        float xz = bf2f(xg_rows[i][H + col]);
-        float xn = bf2f(xg_rows[i][2 * H + col]);
         float g_r = xr * bf2f((uint16_t)gb[i][0][nt]) + bf2f((uint16_t)(gb[i][0][nt] >> 16));
         float g_z = xz * bf2f((uint16_t)gb[i][1][nt]) + bf2f((uint16_t)(gb[i][1][nt] >> 16));
         float g_n = xn * bf2f((uint16_t)gb[i][2][nt]) + bf2f((uint16_t)(gb[i][2][nt] >> 16));
@@ -348,6 +375,8 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
         st8(sv + 2 * H, fn);
         st8(sv + 3 * H, fh);
       }
+      // fence: stop the scheduler interleaving all 4 rows' live ranges
+      __builtin_amdgcn_sched_barrier(0);
     }
     __syncthreads();
 
