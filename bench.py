@@ -37,7 +37,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch", type=int, default=256, help="per-GPU window batch")
+    p.add_argument("--batch", type=int, default=512, help="per-GPU window batch")
     p.add_argument("--endpoints", type=int, default=256, help="API endpoints")
     # 63 + the frontend component = 64 -> batch*64 rows tile exactly onto the
     # 256 CUs for the fused GRU kernels (a 260th workgroup at 1 block/CU
