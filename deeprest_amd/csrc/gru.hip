@@ -402,8 +402,13 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
 // x-side).  A-fragments read the just-written GLOBAL dpre rows (same-CU L2
 // after __syncthreads); the B image in LDS is W^T with rows permuted to the
 // same m ordering, so both sides agree on any K permutation.
-template <typename T>
-__global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
+// NSUB = 1: 4 waves / 64 rows, no LDS, W streamed from L2 (2 blocks/CU can
+//           co-reside when the grid exceeds the CU count).
+// NSUB = 2: 8 waves / 128 rows, the pi-permuted W image staged ONCE in LDS
+//           (96 KB) — 2 waves/SIMD from one block with low-latency
+//           B-fragments; used when the grid still fills the chip.
+template <typename T, int NSUB>
+__global__ __launch_bounds__(NSUB * THREADS) void gru_bwd_kernel(
     const T* __restrict__ grad_h,   // (B, TT, C, H)
     const uint16_t* __restrict__ w_img,  // (H, 384) bf16: w_img[k][m] = W[natJ(m)][k]
     const T* __restrict__ h0,       // (B, C, H)
@@ -412,11 +417,23 @@ __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
     T* __restrict__ dpre,           // (B, TT, C, 4H) pi: dr|dz|dn|d_hhn
     float* __restrict__ dh0,        // (B, C, H)
     int B, int TT, int C, int reverse) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
   const int tid = threadIdx.x;
   const int wv = tid / DR_WAVE;
   const int lane = tid % DR_WAVE;
   const int64_t R = (int64_t)B * C;
-  const int64_t r0 = (int64_t)blockIdx.x * ROWS;
+  const int64_t r0 = (int64_t)blockIdx.x * (ROWS * NSUB);
+
+  if constexpr (NSUB == 2) {
+    // stage the W image into LDS, swz768-swizzled rows [k_h][m]
+    for (int id = tid; id < H * (G3H / 8); id += NSUB * THREADS) {
+      int k = id / (G3H / 8);
+      int mblk = id % (G3H / 8);
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(w_img + (int64_t)k * G3H + mblk * 8);
+      // swizzle per 16B block within the 768 B row
+      *reinterpret_cast<bf16x8*>(smem + k * 768 + ((mblk ^ (k & 15)) << 4)) = v;
+    }
+  }
 
   const int c_col = lane & 15;
   const int rgrp = lane >> 4;
@@ -511,11 +528,11 @@ __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
 
     // ---- MFMA: delta = dpre_pi (64 x 384 over m) @ W_pi, wave's 16 rows ----
     {
-      // opaque copy of the W pointer: stops LLVM from hoisting all 96
-      // loop-invariant B-fragment loads out of the t loop (384 registers ->
-      // scratch).  W streams from L1/L2 each step by design.
+      // NSUB==1: opaque copy of the W pointer stops LLVM from hoisting all
+      // 96 loop-invariant B-fragment loads out of the t loop (384 registers
+      // -> scratch); W streams from L1/L2 each step by design.
       const uint16_t* w_imgv = w_img;
-      asm volatile("" : "+v"(w_imgv));
+      if constexpr (NSUB == 1) asm volatile("" : "+v"(w_imgv));
       const T* arow_p = dpre + (abc0 + bc_toff) * G4H;
       bf16x8 afrag[12];
 #pragma unroll
@@ -531,7 +548,11 @@ __global__ __launch_bounds__(THREADS) void gru_bwd_kernel(
         const uint16_t* wrow = w_imgv + (int64_t)n * G3H;
 #pragma unroll
         for (int kt = 0; kt < 12; ++kt) {
-          bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(wrow + kt * 32 + k0);
+          bf16x8 bfrag;
+          if constexpr (NSUB == 2)
+            bfrag = lds_read8(smem, swz768(n, kt * 32 + k0));
+          else
+            bfrag = *reinterpret_cast<const bf16x8*>(wrow + kt * 32 + k0);
           a = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[kt], bfrag, a, 0, 0, 0);
         }
 #pragma unroll
@@ -672,8 +693,25 @@ static void gru_bwd_launch_t(const void* grad_h, const void* w_img, const void* 
                              float* dh0, int B, int TT, int C, int reverse,
                              hipStream_t stream) {
   int64_t R = (int64_t)B * C;
+  int tiles128 = (int)((R + 2 * ROWS - 1) / (2 * ROWS));
+  if (tiles128 >= 192) {
+    // enough 128-row tiles to fill the chip at 1 block/CU: LDS-W variant
+    constexpr int LDS_WB = H * G3H * 2;  // 96 KB
+    static bool attr_set = false;
+    if (!attr_set) {
+      DR_HIP_CHECK(hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&gru_bwd_kernel<T, 2>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, LDS_WB));
+      attr_set = true;
+    }
+    hipLaunchKernelGGL((gru_bwd_kernel<T, 2>), dim3(tiles128), dim3(2 * THREADS),
+                       LDS_WB, stream, (const T*)grad_h, (const uint16_t*)w_img,
+                       (const T*)h0, (const T*)h_all, (const T*)saves, (T*)dpre,
+                       dh0, B, TT, C, reverse);
+    return;
+  }
   int grid = (int)((R + ROWS - 1) / ROWS);
-  hipLaunchKernelGGL((gru_bwd_kernel<T>), dim3(grid), dim3(THREADS), 0, stream,
+  hipLaunchKernelGGL((gru_bwd_kernel<T, 1>), dim3(grid), dim3(THREADS), 0, stream,
                      (const T*)grad_h, (const uint16_t*)w_img, (const T*)h0,
                      (const T*)h_all, (const T*)saves, (T*)dpre, dh0, B, TT, C,
                      reverse);

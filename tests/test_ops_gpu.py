@@ -275,3 +275,32 @@ def test_predictor_hipgraph_capture(dev):
     assert pred_graph._graph is not None, "hipGraph was not captured"
     for name in data.metric_names:
         np.testing.assert_allclose(out_g[name], out_e[name], rtol=1e-3, atol=1e-3)
+
+
+def test_gru_large_rows_8wave_variant(dev):
+    # R = B*C >= 192*128 exercises the 8-wave LDS-W backward variant that
+    # small shapes never reach
+    from deeprest_amd.ops import fused_gru_sequence, reference_gru_sequence
+
+    torch.manual_seed(8)
+    B, T, C, H = 400, 3, 64, 128  # R = 25600 -> 200 x 128-row tiles
+    xg = torch.randn(B, T, 3 * H, device=dev) * 0.4
+    w_hh = torch.randn(3 * H, H, device=dev) / np.sqrt(H)
+    b_hh = torch.randn(3 * H, device=dev) * 0.1
+    h0 = torch.randn(B, C, H, device=dev) * 0.3
+    gamma = 1.0 + 0.1 * torch.randn(C, 3 * H, device=dev)
+    beta = 0.1 * torch.randn(C, 3 * H, device=dev)
+    args_t = [t.detach().clone().requires_grad_(True)
+              for t in (xg, w_hh, b_hh, h0, gamma, beta)]
+    args_r = [t.detach().clone().requires_grad_(True)
+              for t in (xg, w_hh, b_hh, h0, gamma, beta)]
+    out_t = fused_gru_sequence(*args_t)
+    out_r = reference_gru_sequence(*args_r)
+    torch.testing.assert_close(out_t.float(), out_r, rtol=5e-2, atol=3e-2)
+    g = torch.randn_like(out_r)
+    out_t.backward(g)
+    out_r.backward(g)
+    for name, at_, ar_ in zip(["xg", "w", "b", "h0", "gam", "bet"], args_t, args_r):
+        torch.testing.assert_close(at_.grad.float(), ar_.grad.float(),
+                                   rtol=8e-2, atol=5e-2,
+                                   msg=lambda m, n=name: f"{n}: {m}")
