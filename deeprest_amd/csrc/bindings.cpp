@@ -29,7 +29,7 @@ void dr_fused_adam(const int64_t* meta, int nt, int64_t total, float lr, float b
 void dr_gru_fwd(const void* xg, const void* gamma, const void* beta,
                 const void* w_hh, const float* b_hh, const void* h0, void* h_all,
                 void* saves, int B, int TT, int C, int reverse, int save,
-                int is_bf16, hipStream_t stream);
+                int fp8, int is_bf16, hipStream_t stream);
 void dr_gru_bwd(const void* grad_h, const void* w_hh, const void* h0,
                 const void* h_all, const void* saves, void* dpre_x, float* dh0,
                 int B, int TT, int C, int reverse, int is_bf16, hipStream_t stream);
@@ -157,7 +157,8 @@ void fused_adam(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
 std::vector<at::Tensor> gru_seq_forward(at::Tensor xg, at::Tensor w_hh,
                                         at::Tensor b_hh, at::Tensor h0,
                                         at::Tensor gamma, at::Tensor beta,
-                                        bool reverse, bool save) {
+                                        bool reverse, bool save, bool fp8) {
+  TORCH_CHECK(!(fp8 && save), "fp8 GRU path is inference-only (no saves)");
   const at::cuda::CUDAGuard guard(xg.device());
   check_dtype(xg, "x_gates");
   TORCH_CHECK(xg.dim() == 3, "x_gates must be (B, T, 3H)");
@@ -188,7 +189,7 @@ std::vector<at::Tensor> gru_seq_forward(at::Tensor xg, at::Tensor w_hh,
   dr_gru_fwd(xg.data_ptr(), gamma.data_ptr(), beta.data_ptr(), w_gemm.data_ptr(),
              b_hh.data_ptr<float>(), h0.data_ptr(), h_all.data_ptr(),
              save ? saves.data_ptr() : nullptr, B, TT, C, reverse ? 1 : 0,
-             save ? 1 : 0, dt == at::kBFloat16, cur_stream());
+             save ? 1 : 0, fp8 ? 1 : 0, dt == at::kBFloat16, cur_stream());
   return {h_all, saves};
 }
 
@@ -263,7 +264,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pinball_forward", &pinball_forward);
   m.def("pinball_backward", &pinball_backward);
   m.def("fused_adam", &fused_adam);
-  m.def("gru_seq_forward", &gru_seq_forward);
+  m.def("gru_seq_forward", &gru_seq_forward, py::arg("xg"), py::arg("w_hh"),
+        py::arg("b_hh"), py::arg("h0"), py::arg("gamma"), py::arg("beta"),
+        py::arg("reverse"), py::arg("save"), py::arg("fp8") = false);
   m.def("gru_seq_backward_kernel", &gru_seq_backward_kernel);
   m.def("gru_bwd_reduce", &gru_bwd_reduce);
   m.def("mha_forward", &mha_forward);

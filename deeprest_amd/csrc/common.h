@@ -3,6 +3,7 @@
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <hip/hip_fp8.h>
 #include <stdint.h>
 #include <stdio.h>
 #include <algorithm>

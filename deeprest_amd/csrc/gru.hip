@@ -63,6 +63,17 @@ __device__ __forceinline__ int swz(int row, int k_elem) {
   return row * 256 + ((blk ^ (row & 15)) << 4) + within;
 }
 
+// same for a row-major [rows][128] fp8 tile (128 B rows, 8 B blocks)
+__device__ __forceinline__ int swz8(int row, int k_elem) {
+  int blk = k_elem >> 3;
+  return row * 128 + ((blk ^ (row & 15)) << 3) + (k_elem & 7);
+}
+
+__device__ __forceinline__ uint8_t f2fp8(float v) {
+  __hip_fp8_e4m3 x(v);              // OCP e4m3fn on gfx950 (hardware cvt)
+  return x.__x;
+}
+
 // same swizzle for a row-major [rows][384] bf16 tile (768 B rows)
 __device__ __forceinline__ int swz768(int row, int k_elem) {
   int blk = k_elem >> 3;
@@ -151,7 +162,11 @@ __device__ __forceinline__ bf16x8 ld_frag<float>(const float* src) {
 }
 
 // ---------------------------------------------------------------- forward
-template <typename T, bool SAVE>
+// FP8 = true: W and the h mirror live in LDS as OCP e4m3 and the recurrent
+// GEMM runs on v_mfma_f32_16x16x32_fp8_fp8 (state stays fp32 in registers;
+// activations/outputs keep dtype T).  Inference-only (BASELINE config 5:
+// fp8 MFMA for the long-horizon path); training uses bf16.
+template <typename T, bool SAVE, bool FP8 = false>
 __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
     const T* __restrict__ xg,      // (B, TT, 3H)
     const T* __restrict__ gamma,   // (C, 3H)
@@ -163,9 +178,10 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
     T* __restrict__ saves,         // (B, TT, C, 4H) pi layout (SAVE only)
     int B, int TT, int C, int reverse) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* Wl = smem + LDS_W;
-  char* Hl = smem + LDS_H;
-  char* XGl = smem + LDS_XG;
+  constexpr int ELT = FP8 ? 1 : 2;             // GEMM-tile element bytes
+  char* Wl = smem;
+  char* Hl = smem + G3H * H * ELT;
+  char* XGl = Hl + ROWS * H * ELT;
 
   const int tid = threadIdx.x;
   const int wv = tid / DR_WAVE;
@@ -174,12 +190,21 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
   const int64_t r0 = (int64_t)blockIdx.x * ROWS;
   const int b_lo = (int)(r0 / C);
 
-  // ---- prologue: stage the bf16 W image into swizzled LDS ----
+  // ---- prologue: stage the W image into swizzled LDS ----
   for (int id = tid; id < G3H * (H / 8); id += THREADS) {
     int j = id / (H / 8);
     int blk = id % (H / 8);
     bf16x8 v = *reinterpret_cast<const bf16x8*>(w_gemm + (int64_t)j * H + blk * 8);
-    *reinterpret_cast<bf16x8*>(Wl + j * 256 + ((blk ^ (j & 15)) << 4)) = v;
+    if constexpr (FP8) {
+      const uint16_t* bv = reinterpret_cast<const uint16_t*>(&v);
+      uint8_t q[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) q[e] = f2fp8(bf2f(bv[e]));
+      *reinterpret_cast<uint64_t*>(Wl + j * 128 + ((blk ^ (j & 15)) << 3)) =
+          *reinterpret_cast<const uint64_t*>(q);
+    } else {
+      *reinterpret_cast<bf16x8*>(Wl + j * 256 + ((blk ^ (j & 15)) << 4)) = v;
+    }
   }
 
   // ---- per-lane static geometry (C-layout of the 16x16 MFMA tile) ----
@@ -233,7 +258,10 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
       int col = nt * 16 + c_col;
       float v = live[i] ? ldf(h0 + ((int64_t)b_of[i] * C + comp_of[i]) * H + col) : 0.f;
       h[i][nt] = v;
-      *reinterpret_cast<uint16_t*>(Hl + swz(row_of[i], col)) = f2bf(v);
+      if constexpr (FP8)
+        *reinterpret_cast<uint8_t*>(Hl + swz8(row_of[i], col)) = f2fp8(v);
+      else
+        *reinterpret_cast<uint16_t*>(Hl + swz(row_of[i], col)) = f2bf(v);
     }
 
   // ---- t-invariant descriptors ----
@@ -292,7 +320,8 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
     int arow = wv * 16 + (lane & 15);
     int k0 = (lane >> 4) * 8;
 #pragma unroll
-    for (int kt = 0; kt < KT; ++kt) afrag_off[kt] = swz(arow, kt * 32 + k0);
+    for (int kt = 0; kt < KT; ++kt)
+      afrag_off[kt] = FP8 ? swz8(arow, kt * 32 + k0) : swz(arow, kt * 32 + k0);
   }
   // xg LDS row bases (pi layout, element-indexed)
   const uint16_t* xg_rows[4];
@@ -310,8 +339,14 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
   for (int step = 0; step < TT; ++step) {
     // A-fragments: this wave's 16 rows of the h tile, all 4 K-tiles
     bf16x8 afrag[KT];
+    int64_t afrag8[KT];
 #pragma unroll
-    for (int kt = 0; kt < KT; ++kt) afrag[kt] = lds_read8(Hl, afrag_off[kt]);
+    for (int kt = 0; kt < KT; ++kt) {
+      if constexpr (FP8)
+        afrag8[kt] = *reinterpret_cast<const int64_t*>(Hl + afrag_off[kt]);
+      else
+        afrag[kt] = lds_read8(Hl, afrag_off[kt]);
+    }
 
     // stage xg[., t, :] into LDS in pi layout: the 8 loaded naturals land at
     // positions pi_0 + 8e (16 B stride), so the epilogue reads each (row,
@@ -338,8 +373,13 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
       for (int kt = 0; kt < KT; ++kt) {
         int j = nt * 16 + c_col;                       // W row (gate col)
         int k0 = kt * 32 + (lane >> 4) * 8;
-        bf16x8 bfrag = lds_read8(Wl, swz(j, k0));
-        a = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[kt], bfrag, a, 0, 0, 0);
+        if constexpr (FP8) {
+          int64_t bfrag = *reinterpret_cast<const int64_t*>(Wl + swz8(j, k0));
+          a = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(afrag8[kt], bfrag, a, 0, 0, 0);
+        } else {
+          bf16x8 bfrag = lds_read8(Wl, swz(j, k0));
+          a = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[kt], bfrag, a, 0, 0, 0);
+        }
       }
       acc[nt] = a;
     }
@@ -365,7 +405,10 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
         float nn = tanhf_(g_n + rp * hn);
         float hnew = (1.f - zp) * nn + zp * h[i][nt];
         h[i][nt] = hnew;
-        *reinterpret_cast<uint16_t*>(Hl + swz(row_of[i], col)) = f2bf(hnew);
+        if constexpr (FP8)
+          *reinterpret_cast<uint8_t*>(Hl + swz8(row_of[i], col)) = f2fp8(hnew);
+        else
+          *reinterpret_cast<uint16_t*>(Hl + swz(row_of[i], col)) = f2bf(hnew);
         fr[nt] = rp; fz[nt] = zp; fn[nt] = nn; fh[nt] = hn;
       }
       if (SAVE && live[i]) {
@@ -658,11 +701,14 @@ __global__ void gru_dgamma_kernel(const T* __restrict__ dpre,  // (BT, C, 4H) pi
 }
 
 // ------------------------------------------------------------- launchers
+// fp8 tiles halve the GEMM-side LDS: 48K W + 8K h + 24K xg
+constexpr int LDS_FWD_FP8 = G3H * H + ROWS * H + XG_SLOTS * G3H * 2;  // 81920 B
+
 template <typename T>
 static void gru_fwd_launch_t(const void* xg, const void* gamma, const void* beta,
                              const void* w_gemm, const float* b_hh, const void* h0,
                              void* h_all, void* saves, int B, int TT, int C,
-                             int reverse, int save, hipStream_t stream) {
+                             int reverse, int save, int fp8, hipStream_t stream) {
   int64_t R = (int64_t)B * C;
   int grid = (int)((R + ROWS - 1) / ROWS);
   static bool attr_set = false;
@@ -673,9 +719,17 @@ static void gru_fwd_launch_t(const void* xg, const void* gamma, const void* beta
     DR_HIP_CHECK(hipFuncSetAttribute(
         reinterpret_cast<const void*>(&gru_fwd_kernel<T, false>),
         hipFuncAttributeMaxDynamicSharedMemorySize, LDS_FWD_TOTAL));
+    DR_HIP_CHECK(hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&gru_fwd_kernel<T, false, true>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, LDS_FWD_FP8));
     attr_set = true;
   }
-  if (save)
+  if (fp8)
+    hipLaunchKernelGGL((gru_fwd_kernel<T, false, true>), dim3(grid), dim3(THREADS),
+                       LDS_FWD_FP8, stream, (const T*)xg, (const T*)gamma,
+                       (const T*)beta, (const uint16_t*)w_gemm, b_hh, (const T*)h0,
+                       (T*)h_all, (T*)saves, B, TT, C, reverse);
+  else if (save)
     hipLaunchKernelGGL((gru_fwd_kernel<T, true>), dim3(grid), dim3(THREADS),
                        LDS_FWD_TOTAL, stream, (const T*)xg, (const T*)gamma,
                        (const T*)beta, (const uint16_t*)w_gemm, b_hh, (const T*)h0,
@@ -743,13 +797,13 @@ extern "C" {
 void dr_gru_fwd(const void* xg, const void* gamma, const void* beta,
                 const void* w_hh, const float* b_hh, const void* h0, void* h_all,
                 void* saves, int B, int TT, int C, int reverse, int save,
-                int is_bf16, hipStream_t stream) {
+                int fp8, int is_bf16, hipStream_t stream) {
   if (is_bf16)
     dr::gru_fwd_launch_t<uint16_t>(xg, gamma, beta, w_hh, b_hh, h0, h_all, saves,
-                                   B, TT, C, reverse, save, stream);
+                                   B, TT, C, reverse, save, fp8, stream);
   else
     dr::gru_fwd_launch_t<float>(xg, gamma, beta, w_hh, b_hh, h0, h_all, saves,
-                                B, TT, C, reverse, save, stream);
+                                B, TT, C, reverse, save, fp8, stream);
 }
 
 void dr_gru_bwd(const void* grad_h, const void* w_hh, const void* h0,

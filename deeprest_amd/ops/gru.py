@@ -167,6 +167,7 @@ def fused_gru_sequence(
     gamma: Optional[torch.Tensor] = None,
     beta: Optional[torch.Tensor] = None,
     reverse: bool = False,
+    fp8: bool = False,
 ) -> torch.Tensor:
     if x_gates.is_cuda:
         dt = x_gates.dtype
@@ -177,15 +178,25 @@ def fused_gru_sequence(
             beta = torch.zeros(C, G, device=x_gates.device, dtype=dt)
         elif beta is None:
             beta = torch.zeros_like(gamma)
-        # dtype harmonization outside the Function so the casts are
-        # autograd-tracked back to the fp32 master parameters
-        return _FusedGRUSequence.apply(
+        args = (
             x_gates.contiguous(),
             w_hh.to(dt).contiguous(),
             b_hh.float().contiguous(),
             h0.to(dt).contiguous(),
             gamma.to(dt).contiguous(),
             beta.to(dt).contiguous(),
-            reverse,
         )
+        if fp8:
+            # fp8 MFMA path is inference-only (BASELINE config 5): the
+            # recurrent GEMM runs on e4m3 tiles, state stays fp32 in-kernel
+            if torch.is_grad_enabled() and any(
+                t.requires_grad for t in (x_gates, w_hh, b_hh, h0, gamma, beta)
+            ):
+                raise RuntimeError("fp8 GRU path does not support autograd")
+            ext = require_native("fused_gru_sequence")
+            h_all, _ = ext.gru_seq_forward(*args, reverse, False, True)
+            return h_all
+        # dtype harmonization outside the Function so the casts are
+        # autograd-tracked back to the fp32 master parameters
+        return _FusedGRUSequence.apply(*args, reverse)
     return reference_gru_sequence(x_gates, w_hh, b_hh, h0, gamma, beta, reverse)

@@ -304,3 +304,27 @@ def test_gru_large_rows_8wave_variant(dev):
         torch.testing.assert_close(at_.grad.float(), ar_.grad.float(),
                                    rtol=8e-2, atol=5e-2,
                                    msg=lambda m, n=name: f"{n}: {m}")
+
+
+def test_gru_fp8_inference_path(dev):
+    # fp8 MFMA forward (config 5): looser tolerance vs the fp32 oracle —
+    # e4m3 GEMM operands with fp32 state
+    from deeprest_amd.ops import fused_gru_sequence, reference_gru_sequence
+
+    torch.manual_seed(10)
+    B, T, C, H = 3, 12, 6, 128
+    xg = torch.randn(B, T, 3 * H, device=dev) * 0.4
+    w_hh = torch.randn(3 * H, H, device=dev) / np.sqrt(H)
+    b_hh = torch.randn(3 * H, device=dev) * 0.1
+    h0 = torch.randn(B, C, H, device=dev) * 0.3
+    with torch.no_grad():
+        out8 = fused_gru_sequence(xg, w_hh, b_hh, h0, fp8=True)
+        ref = reference_gru_sequence(xg, w_hh, b_hh, h0)
+    assert torch.isfinite(out8).all()
+    err = (out8.float() - ref).abs().max().item()
+    assert err < 0.25, f"fp8 GRU drifted too far from fp32 oracle: {err}"
+    # and it must be meaningfully closer than noise: correlation check
+    flat8 = out8.float().flatten()
+    flatr = ref.flatten()
+    corr = torch.corrcoef(torch.stack([flat8, flatr]))[0, 1].item()
+    assert corr > 0.99, corr
