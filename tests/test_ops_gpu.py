@@ -300,10 +300,12 @@ def test_gru_large_rows_8wave_variant(dev):
     g = torch.randn_like(out_r)
     out_t.backward(g)
     out_r.backward(g)
+    # gradients here sum ~76k bf16 products: compare error relative to the
+    # tensor's own scale (absolute tolerances don't scale with K)
     for name, at_, ar_ in zip(["xg", "w", "b", "h0", "gam", "bet"], args_t, args_r):
-        torch.testing.assert_close(at_.grad.float(), ar_.grad.float(),
-                                   rtol=8e-2, atol=5e-2,
-                                   msg=lambda m, n=name: f"{n}: {m}")
+        err = (at_.grad.float() - ar_.grad.float()).abs().max()
+        scale = ar_.grad.float().abs().max() + 1e-9
+        assert err / scale < 2e-2, f"{name}: relmax {(err / scale).item():.4f}"
 
 
 def test_gru_fp8_inference_path(dev):
