@@ -131,6 +131,7 @@ class DeepRestNetConfig:
     dropout: float = 0.1
     bidirectional: bool = True
     prop_rounds: int = 2                  # call-graph propagation rounds
+    fp8_inference: bool = False           # fp8 MFMA GRU decode at eval time
 
     def to_dict(self) -> dict:
         return {
@@ -139,6 +140,7 @@ class DeepRestNetConfig:
             "hidden": self.hidden, "comp_dim": self.comp_dim,
             "quantiles": tuple(self.quantiles), "dropout": self.dropout,
             "bidirectional": self.bidirectional, "prop_rounds": self.prop_rounds,
+            "fp8_inference": self.fp8_inference,
         }
 
 
@@ -246,7 +248,7 @@ class _GRUDecoderBank(nn.Module):
                     b.weight.mul_(0.1)
                     b.bias.zero_()
 
-    def forward(self, enc: torch.Tensor, comp: torch.Tensor):
+    def forward(self, enc: torch.Tensor, comp: torch.Tensor, fp8: bool = False):
         """enc: (B, T, D); comp: (C, comp_dim) -> tuple of direction outputs
         (each (B, T, C, H)); the heads consume the halves separately so no
         concat copy is ever materialized."""
@@ -257,7 +259,7 @@ class _GRUDecoderBank(nn.Module):
         beta = self.cond_beta(comp)
         h0 = torch.tanh(self.h0_proj(comp)).unsqueeze(0).expand(B, C, self.hidden)
         out = fused_gru_sequence(xg, self.w_hh, self.b_hh, h0.contiguous(),
-                                 gamma, beta, reverse=False)
+                                 gamma, beta, reverse=False, fp8=fp8)
         if not self.bidirectional:
             return (out,)
         xg_r = self.x_proj_r(enc)
@@ -265,7 +267,7 @@ class _GRUDecoderBank(nn.Module):
         beta_r = self.cond_beta_r(comp)
         h0_r = torch.tanh(self.h0_proj_r(comp)).unsqueeze(0).expand(B, C, self.hidden)
         out_r = fused_gru_sequence(xg_r, self.w_hh_r, self.b_hh_r, h0_r.contiguous(),
-                                   gamma_r, beta_r, reverse=True)
+                                   gamma_r, beta_r, reverse=True, fp8=fp8)
         return (out, out_r)
 
 
@@ -331,7 +333,9 @@ class DeepRestNet(nn.Module):
         for layer in self.layers:
             x = layer(x)
         comp = self.graph()                                   # (C, comp_dim)
-        h_dirs = self.decoder(x, comp)                        # tuple of (B,T,C,H)
+        fp8 = (self.cfg.fp8_inference and not self.training
+               and traffic.is_cuda and not torch.is_grad_enabled())
+        h_dirs = self.decoder(x, comp, fp8=fp8)               # tuple of (B,T,C,H)
         return self._apply_heads(h_dirs)
 
     def _apply_heads(self, h_dirs) -> torch.Tensor:
