@@ -156,22 +156,39 @@ class SyntheticApp:
         self._res_ema = rng.uniform(0.0, 0.9, size=(C, R))  # memory-like persistence
 
     # ------------------------------------------------------------------ traffic
-    def traffic_plan(self) -> np.ndarray:
-        """(T, n_apis) expected call counts: diurnal two-peak shape x popularity."""
+    def traffic_plan(self, scale: float = 1.0, shape: str = "waves",
+                     composition: Optional[Sequence[float]] = None) -> np.ndarray:
+        """(T, n_apis) expected call counts.
+
+        Scenario knobs mirror the reference's evaluation variants
+        (reference: locust/locustfile-{scale,shape,composition}.py):
+        - scale: user multiplier (unseen-scale scenario, e.g. 3.0);
+        - shape: 'waves' (two diurnal Gaussian peaks/day) or 'flat'
+          (steady at peak level — unseen-shape scenario);
+        - composition: per-API mix weights overriding the Zipf popularity
+          (unseen-composition scenario).
+        """
         cfg = self.config
-        T = cfg.n_windows
         t = np.arange(cfg.windows_per_day) / cfg.windows_per_day
         days = []
         for d in range(cfg.n_days):
-            p1, p2 = self.peak_positions[d]
-            shape = (
-                np.exp(-0.5 * ((t - p1) / 0.08) ** 2)
-                + np.exp(-0.5 * ((t - p2) / 0.08) ** 2)
-            )
-            days.append(shape)
+            if shape == "flat":
+                day = np.ones_like(t)
+            else:
+                p1, p2 = self.peak_positions[d]
+                day = (
+                    np.exp(-0.5 * ((t - p1) / 0.08) ** 2)
+                    + np.exp(-0.5 * ((t - p2) / 0.08) ** 2)
+                )
+            days.append(day)
         shape_all = np.concatenate(days)  # (T,)
-        level = cfg.base_calls + (cfg.peak_calls - cfg.base_calls) * shape_all
-        lam = level[:, None] * self.popularity[None, :]
+        level = (cfg.base_calls + (cfg.peak_calls - cfg.base_calls) * shape_all) * scale
+        if composition is not None:
+            mix = np.asarray(composition, dtype=np.float64)
+            mix = mix / mix.sum()
+        else:
+            mix = self.popularity
+        lam = level[:, None] * mix[None, :]
         noise = 1.0 + cfg.noise * self._rng.standard_normal(size=lam.shape)
         counts = self._rng.poisson(np.maximum(lam * noise, 0.0)).astype(np.int64)
         return counts

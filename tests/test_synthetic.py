@@ -79,3 +79,33 @@ def test_anomaly_injection():
     assert (after[10:15] > before[10:15]).all()
     np.testing.assert_allclose(after[:10], before[:10])
     np.testing.assert_allclose(after[15:], before[15:])
+
+
+def test_traffic_scenarios():
+    app = small_app(windows_per_day=100, n_days=1)
+    base = app.traffic_plan()
+    flat = app.traffic_plan(shape="flat")
+    scaled = app.traffic_plan(scale=3.0)
+    comp = app.traffic_plan(composition=[1.0, 0.0, 0.0, 0.0])
+    # flat shape has much lower peak-to-trough ratio than waves
+    assert flat.sum(1).max() / max(flat.sum(1).min(), 1) < 3
+    # 3x users roughly triples volume
+    ratio = scaled.sum() / max(base.sum(), 1)
+    assert 2.0 < ratio < 4.5
+    # composition override routes all calls to api 0
+    assert comp[:, 1:].sum() == 0 and comp[:, 0].sum() > 0
+
+
+def test_plots_produce_files(tmp_path):
+    from deeprest_amd.engine.plots import plot_estimator_overlay, plot_learning_curves
+
+    import numpy as np
+    p1 = plot_learning_curves([1.0, 0.5, 0.3], [1.1, 0.6, 0.4],
+                              str(tmp_path / "lc.png"))
+    gt = np.sin(np.linspace(0, 6, 60)) + 2
+    p2 = plot_estimator_overlay(
+        "svc_cpu", gt,
+        {"resrc": gt * 1.1, "comp": gt * 0.9, "deepr": gt * 1.02},
+        str(tmp_path / "ov.png"), band=(gt * 0.8, gt * 1.2))
+    import os
+    assert os.path.getsize(p1) > 1000 and os.path.getsize(p2) > 1000
