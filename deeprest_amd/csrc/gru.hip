@@ -307,7 +307,8 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
     int b = (int)(rr / C), c = (int)(rr % C);
     hout_bc[u] = ((int64_t)b * TT + t_first) * C + c;
     hout_blk[u] = blk * 8;
-    hout_lds[u] = row * 256 + ((blk ^ (row & 15)) << 4);
+    hout_lds[u] = FP8 ? (row * 128 + ((blk ^ (row & 15)) << 3))
+                      : (row * 256 + ((blk ^ (row & 15)) << 4));
   }
   // per-lane-row indices for the saves stores (pi layout)
   int64_t sv_bc[4];
@@ -428,7 +429,18 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
     for (int u = 0; u < 4; ++u) {
       if (hout_live[u]) {
         T* dst = h_all + (hout_bc[u] + bc_toff) * H + hout_blk[u];
-        st_frag(dst, lds_read8(Hl, hout_lds[u]));
+        if constexpr (FP8) {
+          uint64_t raw = *reinterpret_cast<const uint64_t*>(Hl + hout_lds[u]);
+          const uint8_t* q = reinterpret_cast<const uint8_t*>(&raw);
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            __hip_fp8_e4m3 x;
+            x.__x = q[e];
+            stf(dst + e, float(x));
+          }
+        } else {
+          st_frag(dst, lds_read8(Hl, hout_lds[u]));
+        }
       }
     }
     stage_toff += stage_tstep;
