@@ -166,8 +166,11 @@ __device__ __forceinline__ bf16x8 ld_frag<float>(const float* src) {
 // GEMM runs on v_mfma_f32_16x16x32_fp8_fp8 (state stays fp32 in registers;
 // activations/outputs keep dtype T).  Inference-only (BASELINE config 5:
 // fp8 MFMA for the long-horizon path); training uses bf16.
-template <typename T, bool SAVE, bool FP8 = false>
-__global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
+// NSUB = 2: 8 waves / 128-row tile — 2 waves/SIMD (the kernel is
+// latency-stalled at 1 wave/SIMD); gamma/beta/b_hh then stream from L1/L2
+// instead of registers so each wave fits the 256-reg budget.
+template <typename T, bool SAVE, bool FP8 = false, int NSUB = 1>
+__global__ __launch_bounds__(THREADS * NSUB) void gru_fwd_kernel(
     const T* __restrict__ xg,      // (B, TT, 3H)
     const T* __restrict__ gamma,   // (C, 3H)
     const T* __restrict__ beta,    // (C, 3H)
@@ -179,19 +182,21 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
     int B, int TT, int C, int reverse) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   constexpr int ELT = FP8 ? 1 : 2;             // GEMM-tile element bytes
+  constexpr int BROWS = ROWS * NSUB;           // rows per block
+  constexpr int BTHREADS = THREADS * NSUB;
   char* Wl = smem;
   char* Hl = smem + G3H * H * ELT;
-  char* XGl = Hl + ROWS * H * ELT;
+  char* XGl = Hl + BROWS * H * ELT;
 
   const int tid = threadIdx.x;
   const int wv = tid / DR_WAVE;
   const int lane = tid % DR_WAVE;
   const int64_t R = (int64_t)B * C;
-  const int64_t r0 = (int64_t)blockIdx.x * ROWS;
+  const int64_t r0 = (int64_t)blockIdx.x * BROWS;
   const int b_lo = (int)(r0 / C);
 
   // ---- prologue: stage the W image into swizzled LDS ----
-  for (int id = tid; id < G3H * (H / 8); id += THREADS) {
+  for (int id = tid; id < G3H * (H / 8); id += BTHREADS) {
     int j = id / (H / 8);
     int blk = id % (H / 8);
     bf16x8 v = *reinterpret_cast<const bf16x8*>(w_gemm + (int64_t)j * H + blk * 8);
@@ -225,29 +230,36 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
   }
 
   // ---- preload T-invariant per-lane values ----
-  // gamma/beta packed as bf16 pairs in one u32 per (row, gate, tile): the
-  // LDS budget caps this kernel at 1 block/CU regardless, so registers for
-  // these are free and beat per-step L2 reloads.
-  uint32_t gb[4][3][8];
+  // NSUB==1: gamma/beta packed as bf16 pairs in one u32 per (row, gate,
+  // tile) — LDS caps this variant at 1 block/CU so the registers are free.
+  // NSUB==2: two waves/SIMD need the registers back; gamma/beta/b_hh are
+  // L1/L2-resident and reload per step (opaque pointers in the epilogue).
+  constexpr int GBN = (NSUB == 1) ? 8 : 1;
+  uint32_t gb[4][3][GBN];
+  float bh[3][GBN];
+  int gb_off[4];
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    const T* grow = gamma + (int64_t)comp_of[i] * G3H;
-    const T* brow = beta + (int64_t)comp_of[i] * G3H;
+  for (int i = 0; i < 4; ++i) gb_off[i] = comp_of[i] * G3H;
+  if constexpr (NSUB == 1) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const T* grow = gamma + gb_off[i];
+      const T* brow = beta + gb_off[i];
+#pragma unroll
+      for (int g = 0; g < 3; ++g)
+#pragma unroll
+        for (int nt = 0; nt < 8; ++nt) {
+          int col = g * H + nt * 16 + c_col;
+          uint16_t gv = live[i] ? f2bf(ldf(grow + col)) : 0;
+          uint16_t bv = live[i] ? f2bf(ldf(brow + col)) : 0;
+          gb[i][g][nt] = ((uint32_t)bv << 16) | gv;
+        }
+    }
 #pragma unroll
     for (int g = 0; g < 3; ++g)
 #pragma unroll
-      for (int nt = 0; nt < 8; ++nt) {
-        int col = g * H + nt * 16 + c_col;
-        uint16_t gv = live[i] ? f2bf(ldf(grow + col)) : 0;
-        uint16_t bv = live[i] ? f2bf(ldf(brow + col)) : 0;
-        gb[i][g][nt] = ((uint32_t)bv << 16) | gv;
-      }
+      for (int nt = 0; nt < 8; ++nt) bh[g][nt] = b_hh[g * H + nt * 16 + c_col];
   }
-  float bh[3][8];
-#pragma unroll
-  for (int g = 0; g < 3; ++g)
-#pragma unroll
-    for (int nt = 0; nt < 8; ++nt) bh[g][nt] = b_hh[g * H + nt * 16 + c_col];
 
   // ---- fp32 hidden state in registers + bf16 tile in LDS ----
   float h[4][8];
@@ -278,8 +290,8 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
   int stage_pi0[MAXCH];     // pi-layout LDS element offset of the chunk
   int n_stage = 0;
   {
-    int n_b = (int)(std::min<int64_t>(r0 + ROWS - 1, R - 1) / C) - b_lo + 1;
-    for (int id = tid; id < n_b * (G3H / 8) && n_stage < MAXCH; id += THREADS) {
+    int n_b = (int)(std::min<int64_t>(r0 + BROWS - 1, R - 1) / C) - b_lo + 1;
+    for (int id = tid; id < n_b * (G3H / 8) && n_stage < MAXCH; id += BTHREADS) {
       int slot = id / (G3H / 8);
       int blk = id % (G3H / 8);
       int j0 = blk * 8;
@@ -298,7 +310,7 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
   bool hout_live[4];
 #pragma unroll
   for (int u = 0; u < 4; ++u) {
-    int id = tid + u * THREADS;             // ROWS*(H/8) = 1024 = 4*THREADS
+    int id = tid + u * BTHREADS;            // BROWS*(H/8) = 4*BTHREADS
     int row = id / (H / 8);
     int blk = id % (H / 8);
     int64_t r = r0 + row;
@@ -386,8 +398,14 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
     }
 
     // ---- fused gate epilogue (vectorized pi-layout saves) ----
+    const T* gm_v = gamma;
+    const T* bt_v = beta;
+    const float* bh_v = b_hh;
+    if constexpr (NSUB == 2) asm volatile("" : "+v"(gm_v), "+v"(bt_v), "+v"(bh_v));
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
+      const T* grow = gm_v + gb_off[i];
+      const T* brow = bt_v + gb_off[i];
       float fr[8], fz[8], fn[8], fh[8];
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
@@ -397,12 +415,24 @@ __global__ __launch_bounds__(THREADS) void gru_fwd_kernel(
         float xz = bf2f(xg_rows[i][H + nt]);
         float xn = bf2f(xg_rows[i][2 * H + nt]);
         int col = nt * 16 + c_col;
-        float g_r = xr * bf2f((uint16_t)gb[i][0][nt]) + bf2f((uint16_t)(gb[i][0][nt] >> 16));
-        float g_z = xz * bf2f((uint16_t)gb[i][1][nt]) + bf2f((uint16_t)(gb[i][1][nt] >> 16));
-        float g_n = xn * bf2f((uint16_t)gb[i][2][nt]) + bf2f((uint16_t)(gb[i][2][nt] >> 16));
-        float rp = sigmoidf_(acc[nt][i] + bh[0][nt] + g_r);
-        float zp = sigmoidf_(acc[nt + 8][i] + bh[1][nt] + g_z);
-        float hn = acc[nt + 16][i] + bh[2][nt];
+        float gmr, gmz, gmn, btr, btz, btn, bhr, bhz, bhn;
+        if constexpr (NSUB == 1) {
+          gmr = bf2f((uint16_t)gb[i][0][nt]); btr = bf2f((uint16_t)(gb[i][0][nt] >> 16));
+          gmz = bf2f((uint16_t)gb[i][1][nt]); btz = bf2f((uint16_t)(gb[i][1][nt] >> 16));
+          gmn = bf2f((uint16_t)gb[i][2][nt]); btn = bf2f((uint16_t)(gb[i][2][nt] >> 16));
+          bhr = bh[0][nt]; bhz = bh[1][nt]; bhn = bh[2][nt];
+        } else {
+          gmr = ldf(grow + col); btr = ldf(brow + col);
+          gmz = ldf(grow + H + col); btz = ldf(brow + H + col);
+          gmn = ldf(grow + 2 * H + col); btn = ldf(brow + 2 * H + col);
+          bhr = bh_v[col]; bhz = bh_v[H + col]; bhn = bh_v[2 * H + col];
+        }
+        float g_r = xr * gmr + btr;
+        float g_z = xz * gmz + btz;
+        float g_n = xn * gmn + btn;
+        float rp = sigmoidf_(acc[nt][i] + bhr + g_r);
+        float zp = sigmoidf_(acc[nt + 8][i] + bhz + g_z);
+        float hn = acc[nt + 16][i] + bhn;
         float nn = tanhf_(g_n + rp * hn);
         float hnew = (1.f - zp) * nn + zp * h[i][nt];
         h[i][nt] = hnew;
@@ -715,6 +745,8 @@ __global__ void gru_dgamma_kernel(const T* __restrict__ dpre,  // (BT, C, 4H) pi
 // ------------------------------------------------------------- launchers
 // fp8 tiles halve the GEMM-side LDS: 48K W + 8K h + 24K xg
 constexpr int LDS_FWD_FP8 = G3H * H + ROWS * H + XG_SLOTS * G3H * 2;  // 81920 B
+// 8-wave variant: 96K W + 32K h(128 rows) + 24K xg
+constexpr int LDS_FWD_8W = G3H * H * 2 + 2 * ROWS * H * 2 + XG_SLOTS * G3H * 2;
 
 template <typename T>
 static void gru_fwd_launch_t(const void* xg, const void* gamma, const void* beta,
@@ -734,7 +766,30 @@ static void gru_fwd_launch_t(const void* xg, const void* gamma, const void* beta
     DR_HIP_CHECK(hipFuncSetAttribute(
         reinterpret_cast<const void*>(&gru_fwd_kernel<T, false, true>),
         hipFuncAttributeMaxDynamicSharedMemorySize, LDS_FWD_FP8));
+    DR_HIP_CHECK(hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&gru_fwd_kernel<T, true, false, 2>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, LDS_FWD_8W));
+    DR_HIP_CHECK(hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&gru_fwd_kernel<T, false, false, 2>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, LDS_FWD_8W));
     attr_set = true;
+  }
+  int tiles128 = (int)((R + 2 * ROWS - 1) / (2 * ROWS));
+  if (!fp8 && tiles128 >= 192 && C >= 5) {
+    // 8-wave 128-row variant: 2 waves/SIMD (needs enough tiles + xg slots)
+    if (save)
+      hipLaunchKernelGGL((gru_fwd_kernel<T, true, false, 2>), dim3(tiles128),
+                         dim3(2 * THREADS), LDS_FWD_8W, stream, (const T*)xg,
+                         (const T*)gamma, (const T*)beta, (const uint16_t*)w_gemm,
+                         b_hh, (const T*)h0, (T*)h_all, (T*)saves, B, TT, C,
+                         reverse);
+    else
+      hipLaunchKernelGGL((gru_fwd_kernel<T, false, false, 2>), dim3(tiles128),
+                         dim3(2 * THREADS), LDS_FWD_8W, stream, (const T*)xg,
+                         (const T*)gamma, (const T*)beta, (const uint16_t*)w_gemm,
+                         b_hh, (const T*)h0, (T*)h_all, (T*)saves, B, TT, C,
+                         reverse);
+    return;
   }
   if (fp8)
     hipLaunchKernelGGL((gru_fwd_kernel<T, false, true>), dim3(grid), dim3(THREADS),
