@@ -541,6 +541,7 @@ __global__ __launch_bounds__(NSUB * THREADS) void gru_bwd_kernel(
   for (int i = 0; i < 4; ++i)
 #pragma unroll
     for (int nt = 0; nt < 8; ++nt) dh_carry[i][nt] = 0.f;
+  if constexpr (NSUB == 2) __syncthreads();  // W staging visible to all waves
 
   // ---- compact addressing: per-row indices + ONE shared per-step offset ----
   // (pointer arrays spilled; full recompute burned VALU on divisions — this
@@ -609,7 +610,12 @@ __global__ __launch_bounds__(NSUB * THREADS) void gru_bwd_kernel(
       // iterations and the combined live ranges spill to scratch
       __builtin_amdgcn_sched_barrier(0);
     }
-    __syncthreads();  // all dpre rows visible (same-CU L2) before A-frag reads
+    // Every A-fragment row this wave reads was written by THIS wave's own
+    // lanes (rows wv*16..wv*16+15), so no cross-wave barrier is needed —
+    // only completion of our own stores (same-CU L1/L2 then serves the
+    // cross-lane reads).  Waves free-run, which is worth real latency
+    // hiding on this memory-bound kernel.
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
 
     // ---- MFMA: delta = dpre_pi (64 x 384 over m) @ W_pi, wave's 16 rows ----
     {
