@@ -114,8 +114,13 @@ class SyntheticApp:
                 for _ in range(config.shapes_per_api)
             ]
             self.shapes[api_id] = shapes
-            w = rng.dirichlet(np.ones(config.shapes_per_api) * 2.0)
-            self.shape_weights[api_id] = w
+            # the shape MIX drifts over the timeline (real apps: cache-hit
+            # ratios, request variants change) — a static component-count
+            # linear map cannot track the resulting cost drift, call-path
+            # features can
+            w0 = rng.dirichlet(np.ones(config.shapes_per_api) * 2.0)
+            w1 = rng.dirichlet(np.ones(config.shapes_per_api) * 2.0)
+            self.shape_weights[api_id] = (w0, w1)
 
         # API popularity (Zipf-ish) and diurnal peaks
         pop = 1.0 / (1.0 + np.arange(config.n_apis)) ** 0.8
@@ -148,12 +153,22 @@ class SyntheticApp:
             self._shape_vec[api] = vecs
             self._shape_comp[api] = comps
 
-        # ground-truth resource model parameters per (component, resource)
+        # ground-truth resource model parameters per (component, resource).
+        # The per-(api, shape) cost multipliers make utilization depend on the
+        # PATH MIX, not just total component invocations — the premise of
+        # call-path-aware estimation (a component-count linear model cannot
+        # resolve two shapes hitting the same component at different cost).
         C = len(all_components)
         R = len(config.resources)
         self._res_base = rng.uniform(5.0, 50.0, size=(C, R))
         self._res_gain = rng.uniform(0.2, 2.0, size=(C, R))
         self._res_ema = rng.uniform(0.0, 0.9, size=(C, R))  # memory-like persistence
+        # weighted per-shape component loads: shape s of api hits component c
+        # with cost count * cost_mult (cost_mult in [0.2, 3])
+        self._shape_comp_w: Dict[str, np.ndarray] = {}
+        for api in self.apis:
+            cost = rng.uniform(0.2, 3.0, size=self._shape_comp[api].shape)
+            self._shape_comp_w[api] = self._shape_comp[api] * cost
 
     # ------------------------------------------------------------------ traffic
     def traffic_plan(self, scale: float = 1.0, shape: str = "waves",
@@ -194,17 +209,18 @@ class SyntheticApp:
         return counts
 
     def _sample_shape_counts(self, api_calls: np.ndarray) -> Dict[str, np.ndarray]:
-        """Per api: (T, shapes_per_api) multinomial split of each window's calls."""
+        """Per api: (T, shapes_per_api) multinomial split of each window's
+        calls, with the mix drifting linearly from w0 to w1 over the run."""
         out = {}
+        T = api_calls.shape[0]
+        alpha = np.linspace(0.0, 1.0, max(T, 2))[:T]
         for a, api in enumerate(self.apis):
-            w = self.shape_weights[api]
+            w0, w1 = self.shape_weights[api]
             calls = api_calls[:, a]
-            picks = np.zeros((len(calls), len(w)), dtype=np.int64)
-            nz = calls > 0
-            if nz.any():
-                picks[nz] = np.stack(
-                    [self._rng.multinomial(int(n), w) for n in calls[nz]]
-                )
+            picks = np.zeros((T, len(w0)), dtype=np.int64)
+            for t in np.nonzero(calls > 0)[0]:
+                wt = (1.0 - alpha[t]) * w0 + alpha[t] * w1
+                picks[t] = self._rng.multinomial(int(calls[t]), wt)
             out[api] = picks
         return out
 
@@ -233,9 +249,11 @@ class SyntheticApp:
         C = len(self.all_components)
 
         inv = np.zeros((T, C), dtype=np.int64)
+        drive = np.zeros((T, C))
         for api in self.apis:
             inv += shape_counts[api] @ self._shape_comp[api]
-        res = self._resources_from_invocations(inv)
+            drive += shape_counts[api] @ self._shape_comp_w[api]
+        res = self._resources_from_invocations(drive)
 
         raw = []
         for t in range(T):
@@ -265,10 +283,12 @@ class SyntheticApp:
 
         traffic = np.zeros((T, P), dtype=np.int64)
         inv = np.zeros((T, C), dtype=np.int64)
+        drive = np.zeros((T, C))
         for api in self.apis:
             traffic += shape_counts[api] @ self._shape_vec[api]
             inv += shape_counts[api] @ self._shape_comp[api]
-        res = self._resources_from_invocations(inv)
+            drive += shape_counts[api] @ self._shape_comp_w[api]
+        res = self._resources_from_invocations(drive)
 
         resources = {}
         resource_components = {}
