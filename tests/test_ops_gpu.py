@@ -269,7 +269,9 @@ def test_predictor_hipgraph_capture(dev):
                            device=dev, graph_batch=16, use_graph=True)
     pred_eager = Predictor(model, x_scaler, y_scalers, data.metric_names,
                            device=dev, use_graph=False)
-    w = sliding_window(data.traffic.astype(np.float64), 30)[:40]  # 2.5 batches
+    # a small request (<= graph_batch) takes the replay path; check both the
+    # padded-replay correctness and the adaptive bulk fallthrough
+    w = sliding_window(data.traffic.astype(np.float64), 30)[:12]
     out_g = pred_graph.predict(w)
     out_e = pred_eager.predict(w)
     assert pred_graph._graph is not None, "hipGraph was not captured"
