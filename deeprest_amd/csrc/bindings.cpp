@@ -39,6 +39,10 @@ void dr_gru_bwd_reduce(const void* dpre, const void* gamma, const void* xg,
 void dr_mha_fwd(const void* q, const void* k, const void* v, void* o, float* lse,
                 int64_t BH, int T_len, int D, float scale, int is_bf16,
                 hipStream_t stream);
+void dr_mha_bwd(const void* q, const void* k, const void* v, const void* o,
+                const void* dout, const float* lse, float* dq, void* dk, void* dv,
+                int64_t BH, int T_len, int D, float scale, int is_bf16,
+                hipStream_t stream);
 }
 
 namespace {
@@ -253,6 +257,23 @@ std::vector<at::Tensor> mha_forward(at::Tensor q, at::Tensor k, at::Tensor v,
   return {o, lse};
 }
 
+std::vector<at::Tensor> mha_backward(at::Tensor q, at::Tensor k, at::Tensor v,
+                                     at::Tensor o, at::Tensor dout, at::Tensor lse,
+                                     double scale) {
+  const at::cuda::CUDAGuard guard(q.device());
+  int64_t B = q.size(0), NH = q.size(1);
+  int T_len = (int)q.size(2), D = (int)q.size(3);
+  TORCH_CHECK(dout.is_contiguous() && o.is_contiguous());
+  auto dq = at::zeros(q.sizes(), q.options().dtype(at::kFloat));
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  dr_mha_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+             dout.data_ptr(), lse.data_ptr<float>(), dq.data_ptr<float>(),
+             dk.data_ptr(), dv.data_ptr(), B * NH, T_len, D, (float)scale,
+             is_bf16(q), cur_stream());
+  return {dq, dk, dv};
+}
+
 }  // namespace
 
 void register_featurize(py::module_& m);
@@ -270,4 +291,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gru_seq_backward_kernel", &gru_seq_backward_kernel);
   m.def("gru_bwd_reduce", &gru_bwd_reduce);
   m.def("mha_forward", &mha_forward);
+  m.def("mha_backward", &mha_backward);
 }

@@ -42,25 +42,17 @@ class _MHAFunction(torch.autograd.Function):
     def forward(ctx, q, k, v, scale):
         ext = require_native("mha_forward")
         o, lse = ext.mha_forward(q, k, v, float(scale))
-        ctx.save_for_backward(q, k, v, lse)
+        ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale = float(scale)
         return o
 
     @staticmethod
     def backward(ctx, grad_o):
-        q, k, v, lse = ctx.saved_tensors
-        scale = ctx.scale
-        grad_o = grad_o.contiguous()
-        # recompute P in fp32 from the saved logsumexp
-        s = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
-        p = torch.exp(s - lse.unsqueeze(-1))                  # (B, H, T, T)
-        dv = torch.matmul(p.transpose(-1, -2), grad_o.float())
-        dp = torch.matmul(grad_o.float(), v.float().transpose(-1, -2))
-        row = (dp * p).sum(dim=-1, keepdim=True)
-        ds = p * (dp - row)
-        dq = torch.matmul(ds, k.float()) * scale
-        dk = torch.matmul(ds.transpose(-1, -2), q.float()) * scale
-        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None
+        ext = require_native("mha_backward")
+        q, k, v, o, lse = ctx.saved_tensors
+        dq, dk, dv = ext.mha_backward(q, k, v, o, grad_o.contiguous(), lse,
+                                      ctx.scale)
+        return dq.to(q.dtype), dk, dv, None
 
 
 def mha_forward(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
