@@ -18,18 +18,16 @@ from __future__ import annotations
 
 import argparse
 import json
-import os
 import sys
 import time
 
-import numpy as np
 import torch
 
 from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
 from deeprest_amd.engine.dataset import EstimationDataset
 from deeprest_amd.models.net import DeepRestNet, DeepRestNetConfig, build_model_spec
 from deeprest_amd.ops.adam import FusedAdam
-from deeprest_amd.parallel.dist import DistContext, init_distributed
+from deeprest_amd.parallel.dist import init_distributed
 
 
 def parse_args():

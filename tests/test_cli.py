@@ -1,8 +1,6 @@
-import json
 import subprocess
 import sys
 
-import numpy as np
 
 from deeprest_amd.data.contract import save_raw_data
 from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig

@@ -7,8 +7,6 @@ and a 2-rank training step producing identical models on both ranks.
 
 import os
 
-import numpy as np
-import pytest
 import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp

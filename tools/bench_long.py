@@ -15,7 +15,6 @@ import time
 
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
-import numpy as np
 import torch
 
 from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
