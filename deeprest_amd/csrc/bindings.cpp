@@ -30,9 +30,11 @@ void dr_gru_fwd(const void* xg, const void* gamma, const void* beta,
                 const void* w_hh, const float* b_hh, const void* h0, void* h_all,
                 void* saves, int B, int TT, int C, int reverse, int save,
                 int fp8, int is_bf16, hipStream_t stream);
-void dr_gru_bwd(const void* grad_h, const void* w_hh, const void* h0,
-                const void* h_all, const void* saves, void* dpre_x, float* dh0,
-                int B, int TT, int C, int reverse, int is_bf16, hipStream_t stream);
+void dr_gru_bwd(const void* grad_h, const void* w_img, const void* w_fwd,
+                const void* xg, const void* gamma, const void* beta,
+                const float* b_hh, const void* h0, const void* h_all,
+                const void* saves, void* dpre, float* dh0, int B, int TT, int C,
+                int reverse, int is_bf16, hipStream_t stream);
 void dr_gru_bwd_reduce(const void* dpre, const void* gamma, const void* xg,
                        void* dxg, float* dgamma, float* dbeta, int64_t BT, int C,
                        int is_bf16, hipStream_t stream);
@@ -184,7 +186,7 @@ std::vector<at::Tensor> gru_seq_forward(at::Tensor xg, at::Tensor w_hh,
   TORCH_CHECK(xg.numel() < (1LL << 31),
               "x_gates too large for 32-bit staging offsets");
   auto h_all = at::empty({B, TT, C, H}, xg.options());
-  auto saves = save ? at::empty({B, TT, C, 4 * H}, xg.options())
+  auto saves = save ? at::empty({B, TT, C, 2 * H}, xg.options())
                     : at::empty({0}, xg.options());
   // the GEMM streams a bf16 weight image from L2 regardless of T
   auto w_gemm = w_hh.scalar_type() == at::kBFloat16
@@ -198,26 +200,38 @@ std::vector<at::Tensor> gru_seq_forward(at::Tensor xg, at::Tensor w_hh,
 }
 
 std::vector<at::Tensor> gru_seq_backward_kernel(at::Tensor grad_h, at::Tensor w_img,
-                                                at::Tensor h0, at::Tensor h_all,
-                                                at::Tensor saves, bool reverse) {
+                                                at::Tensor w_fwd, at::Tensor xg,
+                                                at::Tensor gamma, at::Tensor beta,
+                                                at::Tensor b_hh, at::Tensor h0,
+                                                at::Tensor h_all, at::Tensor saves,
+                                                bool reverse) {
   const at::cuda::CUDAGuard guard(grad_h.device());
   int B = (int)grad_h.size(0), TT = (int)grad_h.size(1);
   int C = (int)grad_h.size(2), H = (int)grad_h.size(3);
   TORCH_CHECK(H == 128);
-  TORCH_CHECK(saves.numel() == (int64_t)B * TT * C * 4 * H,
+  TORCH_CHECK(saves.numel() == (int64_t)B * TT * C * 2 * H,
               "gru backward: saves tensor missing or wrong size (forward must "
               "run with save=true)");
   TORCH_CHECK(w_img.scalar_type() == at::kBFloat16 &&
                   w_img.sizes() == at::IntArrayRef({H, 3 * H}) &&
                   w_img.is_contiguous(),
               "w_img must be the (H, 3H) bf16 pi-permuted W image");
+  TORCH_CHECK(w_fwd.scalar_type() == at::kBFloat16 &&
+                  w_fwd.sizes() == at::IntArrayRef({3 * H, H}) &&
+                  w_fwd.is_contiguous(),
+              "w_fwd must be the (3H, H) bf16 weight image");
+  TORCH_CHECK(b_hh.scalar_type() == at::kFloat && b_hh.is_contiguous());
   auto dt = grad_h.scalar_type();
+  TORCH_CHECK(xg.scalar_type() == dt && gamma.scalar_type() == dt &&
+              beta.scalar_type() == dt);
   TORCH_CHECK(grad_h.is_contiguous() && h_all.is_contiguous() && saves.is_contiguous());
   auto dpre = at::empty({B, TT, C, 4 * H}, grad_h.options());
   auto dh0 = at::empty({B, C, H}, grad_h.options().dtype(at::kFloat));
-  dr_gru_bwd(grad_h.data_ptr(), w_img.data_ptr(), h0.data_ptr(), h_all.data_ptr(),
-             saves.data_ptr(), dpre.data_ptr(), dh0.data_ptr<float>(), B, TT, C,
-             reverse ? 1 : 0, dt == at::kBFloat16, cur_stream());
+  dr_gru_bwd(grad_h.data_ptr(), w_img.data_ptr(), w_fwd.data_ptr(), xg.data_ptr(),
+             gamma.data_ptr(), beta.data_ptr(), b_hh.data_ptr<float>(),
+             h0.data_ptr(), h_all.data_ptr(), saves.data_ptr(), dpre.data_ptr(),
+             dh0.data_ptr<float>(), B, TT, C, reverse ? 1 : 0,
+             dt == at::kBFloat16, cur_stream());
   return {dpre, dh0};
 }
 

@@ -406,7 +406,7 @@ __global__ __launch_bounds__(THREADS * NSUB) void gru_fwd_kernel(
     for (int i = 0; i < 4; ++i) {
       const T* grow = gm_v + gb_off[i];
       const T* brow = bt_v + gb_off[i];
-      float fr[8], fz[8], fn[8], fh[8];
+      float fr[8], fz[8], fn[8];
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
         // pi layout: per-gate values sit at CONSECUTIVE LDS addresses over
@@ -440,14 +440,14 @@ __global__ __launch_bounds__(THREADS * NSUB) void gru_fwd_kernel(
           *reinterpret_cast<uint8_t*>(Hl + swz8(row_of[i], col)) = f2fp8(hnew);
         else
           *reinterpret_cast<uint16_t*>(Hl + swz(row_of[i], col)) = f2bf(hnew);
-        fr[nt] = rp; fz[nt] = zp; fn[nt] = nn; fh[nt] = hn;
+        fr[nt] = rp; fz[nt] = zp; fn[nt] = nn; (void)fn;
       }
       if (SAVE && live[i]) {
-        T* sv = saves + (sv_bc[i] + bc_toff) * G4H + c_col * 8;
+        // only r and z are saved (2H); the backward recomputes n and hh_n
+        // (an extra H x H GEMM there is cheaper than 2H of HBM both ways)
+        T* sv = saves + (sv_bc[i] + bc_toff) * (2 * H) + c_col * 8;
         st8(sv, fr);
         st8(sv + H, fz);
-        st8(sv + 2 * H, fn);
-        st8(sv + 3 * H, fh);
       }
       // fence: stop the scheduler interleaving all 4 rows' live ranges
       __builtin_amdgcn_sched_barrier(0);
@@ -496,13 +496,19 @@ template <typename T, int NSUB>
 __global__ __launch_bounds__(NSUB * THREADS) void gru_bwd_kernel(
     const T* __restrict__ grad_h,   // (B, TT, C, H)
     const uint16_t* __restrict__ w_img,  // (H, 384) bf16: w_img[k][m] = W[natJ(m)][k]
+    const uint16_t* __restrict__ w_fwd,  // (3H, H) bf16 natural layout
+    const T* __restrict__ xg,       // (B, TT, 3H)
+    const T* __restrict__ gamma,    // (C, 3H)
+    const T* __restrict__ beta,     // (C, 3H)
+    const float* __restrict__ b_hh, // (3H,)
     const T* __restrict__ h0,       // (B, C, H)
     const T* __restrict__ h_all,    // (B, TT, C, H)
-    const T* __restrict__ saves,    // (B, TT, C, 4H) pi layout
+    const T* __restrict__ saves,    // (B, TT, C, 2H) pi layout: r|z
     T* __restrict__ dpre,           // (B, TT, C, 4H) pi: dr|dz|dn|d_hhn
     float* __restrict__ dh0,        // (B, C, H)
     int B, int TT, int C, int reverse) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* wn_lds = smem + H * G3H * 2;   // (NSUB==2) 128 x 128 bf16 swizzled
   const int tid = threadIdx.x;
   const int wv = tid / DR_WAVE;
   const int lane = tid % DR_WAVE;
@@ -510,13 +516,20 @@ __global__ __launch_bounds__(NSUB * THREADS) void gru_bwd_kernel(
   const int64_t r0 = (int64_t)blockIdx.x * (ROWS * NSUB);
 
   if constexpr (NSUB == 2) {
-    // stage the W image into LDS, swz768-swizzled rows [k_h][m]
+    // stage the pi W image (dh GEMM), swz768-swizzled rows [k_h][m]
     for (int id = tid; id < H * (G3H / 8); id += NSUB * THREADS) {
       int k = id / (G3H / 8);
       int mblk = id % (G3H / 8);
       bf16x8 v = *reinterpret_cast<const bf16x8*>(w_img + (int64_t)k * G3H + mblk * 8);
-      // swizzle per 16B block within the 768 B row
       *reinterpret_cast<bf16x8*>(smem + k * 768 + ((mblk ^ (k & 15)) << 4)) = v;
+    }
+    // stage W_hn rows (hh_n recompute GEMM), fwd-style [j][k] swizzled
+    for (int id = tid; id < H * (H / 8); id += NSUB * THREADS) {
+      int jl = id / (H / 8);
+      int blk = id % (H / 8);
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          w_fwd + (int64_t)(2 * H + jl) * H + blk * 8);
+      *reinterpret_cast<bf16x8*>(wn_lds + jl * 256 + ((blk ^ (jl & 15)) << 4)) = v;
     }
   }
 
@@ -559,42 +572,91 @@ __global__ __launch_bounds__(NSUB * THREADS) void gru_bwd_kernel(
   const int64_t ra = (r0 + arow < R) ? (r0 + arow) : (R - 1);
   const int ab = (int)(ra / C), ac = (int)(ra % C);
   const int64_t abc0 = ((int64_t)ab * TT + t_first) * C + ac;
+  const int64_t h0_off_a = ((int64_t)ab * C + ac) * H;
   const int k0 = (lane >> 4) * 8;
+  // per-row xg bases + t-invariant b_hh n-slice
+  int64_t xgb0[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+    xgb0[i] = ((int64_t)b_of[i] * TT + t_first) * G3H;
+  float bhn[8];
+#pragma unroll
+  for (int nt = 0; nt < 8; ++nt) bhn[nt] = b_hh[2 * H + nt * 16 + c_col];
 
   int64_t bc_toff = 0;      // += dirC per step
+  int64_t xg_toff = 0;      // += dir*G3H per step
+  const int64_t xg_step = (reverse ? 1 : -1) * (int64_t)G3H;
   for (int step = 0; step < TT; ++step) {
     // the FIRST processed step is the forward pass's LAST, whose h_prev is
     // h_all of the step before; h0 is the prev only at the LAST processed
     // step (= forward t boundary)
     const bool use_h0 = (step == TT - 1);
 
+    // ---- recompute hh_n = h_prev @ W_hn^T for this wave's 16 rows ----
+    // (saving it cost 2H of HBM each way; one extra HxH MFMA pass is cheaper)
+    f32x4 hhn_acc[8];
+    {
+      const uint16_t* wfv = w_fwd;
+      if constexpr (NSUB == 1) asm volatile("" : "+v"(wfv));
+      const T* hsrc_a = use_h0 ? (h0 + h0_off_a)
+                               : (h_all + (abc0 + bc_toff + dirC) * H);
+      bf16x8 hfrag[KT];
+#pragma unroll
+      for (int kt = 0; kt < KT; ++kt)
+        hfrag[kt] = ld_frag(hsrc_a + kt * 32 + k0);
+#pragma unroll
+      for (int nt = 0; nt < 8; ++nt) {
+        f32x4 a = {0.f, 0.f, 0.f, 0.f};
+        const int jl = nt * 16 + c_col;
+#pragma unroll
+        for (int kt = 0; kt < KT; ++kt) {
+          bf16x8 bfrag;
+          if constexpr (NSUB == 2)
+            bfrag = lds_read8(wn_lds, swz(jl, kt * 32 + k0));
+          else
+            bfrag = *reinterpret_cast<const bf16x8*>(
+                wfv + (int64_t)(2 * H + jl) * H + kt * 32 + k0);
+          a = __builtin_amdgcn_mfma_f32_16x16x32_bf16(hfrag[kt], bfrag, a, 0, 0, 0);
+        }
+        hhn_acc[nt] = a;
+      }
+    }
+
+    const T* gm_v = gamma;
+    const T* bt_v = beta;
+    asm volatile("" : "+v"(gm_v), "+v"(bt_v));
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       if (!live[i]) continue;
       const int64_t bc = bc0[i] + bc_toff;
-      const T* sv_p = saves + bc * G4H + c_col * 8;
-      float rp[8], zp[8], nn[8], hn[8];
+      const T* sv_p = saves + bc * (2 * H) + c_col * 8;
+      float rp[8], zp[8];
       ld8(sv_p, rp);
       ld8(sv_p + H, zp);
-      ld8(sv_p + 2 * H, nn);
-      ld8(sv_p + 3 * H, hn);
       float fdr[8], fdz[8], fdn[8], fdh[8];
       // h_prev sits at bc + dirC rows (tprev = t+dir); h0 at the boundary.
       // pointer select -> ONE load (the OOB address is never dereferenced)
       const T* hsrc = use_h0 ? (h0 + h0_off[i])
                              : (h_all + (bc + dirC) * H);
       const T* gr_p = grad_h + bc * H;
+      const T* xg_n = xg + xgb0[i] + xg_toff + 2 * H;
+      const T* gmn = gm_v + comp_of[i] * G3H + 2 * H;
+      const T* btn = bt_v + comp_of[i] * G3H + 2 * H;
 #pragma unroll
       for (int nt = 0; nt < 8; ++nt) {
         int col = nt * 16 + c_col;
         float hp = ldf(hsrc + col);
         float g = ldf(gr_p + col);
+        // recompute the n gate: hh_n from the GEMM above, g_n from xg/FiLM
+        float hn = hhn_acc[nt][i] + bhn[nt];
+        float g_n = ldf(xg_n + col) * ldf(gmn + col) + ldf(btn + col);
+        float nn = tanhf_(g_n + rp[nt] * hn);
         float dht = g + dh_carry[i][nt];
-        float dz = dht * (hp - nn[nt]);
+        float dz = dht * (hp - nn);
         float dn = dht * (1.f - zp[nt]);
-        float dnp = dn * (1.f - nn[nt] * nn[nt]);
+        float dnp = dn * (1.f - nn * nn);
         float dhhn = dnp * rp[nt];
-        float dr = dnp * hn[nt];
+        float dr = dnp * hn;
         fdr[nt] = dr * rp[nt] * (1.f - rp[nt]);
         fdz[nt] = dz * zp[nt] * (1.f - zp[nt]);
         fdn[nt] = dnp;
@@ -651,8 +713,9 @@ __global__ __launch_bounds__(NSUB * THREADS) void gru_bwd_kernel(
       }
     }
     bc_toff += dirC;
+    xg_toff += xg_step;
     // no end-of-step barrier: the next step's stores go to DIFFERENT dpre
-    // addresses (per-t storage); the W image is read-only.
+    // addresses (per-t storage); the W images are read-only.
   }
 
   // ---- dh0 = final carry ----
@@ -815,7 +878,9 @@ static void gru_fwd_launch_t(const void* xg, const void* gamma, const void* beta
 }
 
 template <typename T>
-static void gru_bwd_launch_t(const void* grad_h, const void* w_img, const void* h0,
+static void gru_bwd_launch_t(const void* grad_h, const void* w_img,
+                             const void* w_fwd, const void* xg, const void* gamma,
+                             const void* beta, const float* b_hh, const void* h0,
                              const void* h_all, const void* saves, void* dpre,
                              float* dh0, int B, int TT, int C, int reverse,
                              hipStream_t stream) {
@@ -823,7 +888,7 @@ static void gru_bwd_launch_t(const void* grad_h, const void* w_img, const void* 
   int tiles128 = (int)((R + 2 * ROWS - 1) / (2 * ROWS));
   if (tiles128 >= 192) {
     // enough 128-row tiles to fill the chip at 1 block/CU: LDS-W variant
-    constexpr int LDS_WB = H * G3H * 2;  // 96 KB
+    constexpr int LDS_WB = H * G3H * 2 + H * H * 2;  // 96 + 32 KB
     static bool attr_set = false;
     if (!attr_set) {
       DR_HIP_CHECK(hipFuncSetAttribute(
@@ -833,15 +898,17 @@ static void gru_bwd_launch_t(const void* grad_h, const void* w_img, const void* 
     }
     hipLaunchKernelGGL((gru_bwd_kernel<T, 2>), dim3(tiles128), dim3(2 * THREADS),
                        LDS_WB, stream, (const T*)grad_h, (const uint16_t*)w_img,
-                       (const T*)h0, (const T*)h_all, (const T*)saves, (T*)dpre,
-                       dh0, B, TT, C, reverse);
+                       (const uint16_t*)w_fwd, (const T*)xg, (const T*)gamma,
+                       (const T*)beta, b_hh, (const T*)h0, (const T*)h_all,
+                       (const T*)saves, (T*)dpre, dh0, B, TT, C, reverse);
     return;
   }
   int grid = (int)((R + ROWS - 1) / ROWS);
   hipLaunchKernelGGL((gru_bwd_kernel<T, 1>), dim3(grid), dim3(THREADS), 0, stream,
-                     (const T*)grad_h, (const uint16_t*)w_img, (const T*)h0,
-                     (const T*)h_all, (const T*)saves, (T*)dpre, dh0, B, TT, C,
-                     reverse);
+                     (const T*)grad_h, (const uint16_t*)w_img,
+                     (const uint16_t*)w_fwd, (const T*)xg, (const T*)gamma,
+                     (const T*)beta, b_hh, (const T*)h0, (const T*)h_all,
+                     (const T*)saves, (T*)dpre, dh0, B, TT, C, reverse);
 }
 
 template <typename T>
@@ -879,15 +946,19 @@ void dr_gru_fwd(const void* xg, const void* gamma, const void* beta,
                                 B, TT, C, reverse, save, fp8, stream);
 }
 
-void dr_gru_bwd(const void* grad_h, const void* w_hh, const void* h0,
-                const void* h_all, const void* saves, void* dpre, float* dh0,
-                int B, int TT, int C, int reverse, int is_bf16, hipStream_t stream) {
+void dr_gru_bwd(const void* grad_h, const void* w_img, const void* w_fwd,
+                const void* xg, const void* gamma, const void* beta,
+                const float* b_hh, const void* h0, const void* h_all,
+                const void* saves, void* dpre, float* dh0, int B, int TT, int C,
+                int reverse, int is_bf16, hipStream_t stream) {
   if (is_bf16)
-    dr::gru_bwd_launch_t<uint16_t>(grad_h, w_hh, h0, h_all, saves, dpre, dh0,
-                                   B, TT, C, reverse, stream);
+    dr::gru_bwd_launch_t<uint16_t>(grad_h, w_img, w_fwd, xg, gamma, beta, b_hh,
+                                   h0, h_all, saves, dpre, dh0, B, TT, C,
+                                   reverse, stream);
   else
-    dr::gru_bwd_launch_t<float>(grad_h, w_hh, h0, h_all, saves, dpre, dh0,
-                                B, TT, C, reverse, stream);
+    dr::gru_bwd_launch_t<float>(grad_h, w_img, w_fwd, xg, gamma, beta, b_hh,
+                                h0, h_all, saves, dpre, dh0, B, TT, C,
+                                reverse, stream);
 }
 
 void dr_gru_bwd_reduce(const void* dpre, const void* gamma, const void* xg,

@@ -81,16 +81,16 @@ class _FusedGRUSequence(torch.autograd.Function):
         h_all, saves = ext.gru_seq_forward(
             x_gates, w_hh, b_hh, h0, gamma, beta, bool(reverse), bool(need_grad)
         )
-        ctx.save_for_backward(x_gates, w_hh, h0, gamma, h_all, saves)
+        ctx.save_for_backward(x_gates, w_hh, b_hh, h0, gamma, beta, h_all, saves)
         ctx.reverse = bool(reverse)
         return h_all
 
     @staticmethod
     def backward(ctx, grad_h_all):
         ext = require_native("fused_gru_sequence")
-        x_gates, w_hh, h0, gamma, h_all, saves = ctx.saved_tensors
+        x_gates, w_hh, b_hh, h0, gamma, beta, h_all, saves = ctx.saved_tensors
         reverse = ctx.reverse
-        # pi-permuted W image (H, 3H) the kernel streams from L2: row k holds
+        # pi-permuted W image (H, 3H) for the dh GEMM: row k holds
         # W[natJ(m), k] over the MFMA K axis m (dr | dz | d_hhn regions)
         H = w_hh.shape[1]
         m = torch.arange(3 * H, device=w_hh.device)
@@ -99,9 +99,11 @@ class _FusedGRUSequence(torch.autograd.Function):
         col = (q % 8) * 16 + (q // 8)
         natj = torch.where(m < 2 * H, g * H + col, 2 * H + col)
         w_img = w_hh[natj, :].t().contiguous().to(torch.bfloat16)
+        w_fwd = w_hh.to(torch.bfloat16).contiguous()  # hh_n recompute GEMM
         # sequential chain (custom kernel): dpre = [dr_pre|dz_pre|dn_pre|d_hh_n]
         dpre, dh0 = ext.gru_seq_backward_kernel(
-            grad_h_all.contiguous(), w_img, h0, h_all, saves, reverse
+            grad_h_all.contiguous(), w_img, w_fwd, x_gates, gamma, beta, b_hh,
+            h0, h_all, saves, reverse
         )
         B, T, C, G4 = dpre.shape
         H = G4 // 4
