@@ -87,7 +87,7 @@ std::vector<at::Tensor> layer_norm_backward(at::Tensor dy, at::Tensor x, at::Ten
   TORCH_CHECK(dy.scalar_type() == x.scalar_type());
   int D = (int)x.size(-1);
   int64_t n_rows = x.numel() / D;
-  int n_blocks = (int)std::min<int64_t>((n_rows + 3) / 4, 512);
+  int n_blocks = (int)std::min<int64_t>((n_rows + 3) / 4, 1024);
   if (n_blocks == 0) n_blocks = 1;
   auto dx = at::empty_like(x);
   auto part = at::zeros({n_blocks, 2, D}, x.options().dtype(at::kFloat));
