@@ -25,6 +25,7 @@ import torch
 
 from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
 from deeprest_amd.engine.dataset import EstimationDataset
+from deeprest_amd.engine.graphstep import GraphedTrainStep
 from deeprest_amd.models.net import DeepRestNet, DeepRestNetConfig, build_model_spec
 from deeprest_amd.ops.adam import FusedAdam
 from deeprest_amd.parallel.dist import init_distributed
@@ -80,7 +81,7 @@ def main():
 
     torch.manual_seed(1234)  # identical init on all ranks (weak-scaled DP)
     model = DeepRestNet(spec, DeepRestNetConfig(dropout=0.0)).to(device)
-    opt = FusedAdam(model.parameters(), lr=1e-3)
+    opt = FusedAdam(model.parameters(), lr=1e-3, capturable=on_gpu)
 
     X = ds.X_train.to(device)
     y = ds.y_train.to(device)
@@ -92,9 +93,19 @@ def main():
     if n < B:
         raise RuntimeError(f"only {n} train windows for per-GPU batch {B}")
 
+    # single-GPU: capture the whole step in a hipGraph (None -> stay eager)
+    graphed = None
+    if on_gpu:
+        graphed = GraphedTrainStep.build(
+            model, opt, lambda o, t: model.loss(o.float(), t),
+            X[:B], y[:B],
+            autocast_dtype=autocast_dtype if use_autocast else None, warmup=2)
+
     def step(i: int):
         s = (i * B) % max(n - B, 1)
         xb, yb = X[s : s + B], y[s : s + B]
+        if graphed is not None:
+            return graphed.run(xb, yb)
         with torch.autocast(device_type="cuda", dtype=autocast_dtype,
                             enabled=use_autocast):
             out = model(xb)

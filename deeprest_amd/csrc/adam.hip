@@ -8,10 +8,20 @@
 
 namespace dr {
 
+// STEP_DEV = true: the step counter lives in device memory (incremented by a
+// captured device op between replays), so bias correction stays exact inside
+// a hipGraph-captured training step; host-arg bias factors would be frozen
+// at their capture-time values.
+template <bool STEP_DEV>
 __global__ void fused_adam_kernel(const int64_t* __restrict__ meta, int nt,
                                   int64_t total, float lr, float beta1, float beta2,
                                   float eps, float weight_decay, float bias_c1,
-                                  float bias_c2) {
+                                  float bias_c2, const int* __restrict__ step_ptr) {
+  if (STEP_DEV) {
+    const float s = (float)*step_ptr;
+    bias_c1 = 1.f - __powf(beta1, s);
+    bias_c2 = 1.f - __powf(beta2, s);
+  }
   const int64_t* p_ptrs = meta;
   const int64_t* g_ptrs = meta + nt;
   const int64_t* m_ptrs = meta + 2 * nt;
@@ -59,8 +69,20 @@ void dr_fused_adam(const int64_t* meta, int nt, int64_t total, float lr, float b
   if (grid == 0) grid = 1;
   float bias_c1 = 1.f - powf(beta1, (float)step);
   float bias_c2 = 1.f - powf(beta2, (float)step);
-  hipLaunchKernelGGL(dr::fused_adam_kernel, dim3(grid), dim3(block), 0, stream, meta,
-                     nt, total, lr, beta1, beta2, eps, weight_decay, bias_c1, bias_c2);
+  hipLaunchKernelGGL((dr::fused_adam_kernel<false>), dim3(grid), dim3(block), 0,
+                     stream, meta, nt, total, lr, beta1, beta2, eps, weight_decay,
+                     bias_c1, bias_c2, nullptr);
+}
+
+void dr_fused_adam_dev(const int64_t* meta, int nt, int64_t total, float lr,
+                       float beta1, float beta2, float eps, float weight_decay,
+                       const int* step_ptr, hipStream_t stream) {
+  const int block = 256;
+  int grid = (int)std::min<int64_t>((total + block - 1) / block, 4096);
+  if (grid == 0) grid = 1;
+  hipLaunchKernelGGL((dr::fused_adam_kernel<true>), dim3(grid), dim3(block), 0,
+                     stream, meta, nt, total, lr, beta1, beta2, eps, weight_decay,
+                     0.f, 0.f, step_ptr);
 }
 
 }  // extern "C"

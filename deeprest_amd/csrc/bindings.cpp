@@ -22,10 +22,14 @@ void dr_pinball_fwd(const float* out, const float* labels, const float* quantile
                     int Q, int64_t N, float inv_count, float* loss,
                     hipStream_t stream);
 void dr_pinball_bwd(const float* out, const float* labels, const float* quantiles,
-                    int Q, int64_t N, float gscale, float* dout, hipStream_t stream);
+                    int Q, int64_t N, const float* grad_loss, float inv_n,
+                    float* dout, hipStream_t stream);
 void dr_fused_adam(const int64_t* meta, int nt, int64_t total, float lr, float beta1,
                    float beta2, float eps, float weight_decay, int step,
                    hipStream_t stream);
+void dr_fused_adam_dev(const int64_t* meta, int nt, int64_t total, float lr,
+                       float beta1, float beta2, float eps, float weight_decay,
+                       const int* step_ptr, hipStream_t stream);
 void dr_gru_fwd(const void* xg, const void* gamma, const void* beta,
                 const void* w_hh, const float* b_hh, const void* h0, void* h_all,
                 void* saves, int B, int TT, int C, int reverse, int save,
@@ -123,11 +127,15 @@ at::Tensor pinball_backward(at::Tensor grad, at::Tensor outputs, at::Tensor labe
   int Q = (int)outputs.size(-1);
   int64_t N = outputs.numel() / Q;
   auto dout = at::empty_like(outputs);
-  float g = grad.item<float>();
-  float gscale = N > 0 ? g / (float)N : 0.f;
+  // keep the upstream grad on-device (no .item() host sync — the kernel reads
+  // it, so the whole loss backward is hipGraph-capturable)
+  TORCH_CHECK(grad.is_cuda() && grad.scalar_type() == at::kFloat && grad.numel() == 1,
+              "pinball grad must be a scalar f32 CUDA tensor");
+  float inv_n = N > 0 ? 1.f / (float)N : 0.f;
+  auto gc = grad.contiguous();
   dr_pinball_bwd(outputs.data_ptr<float>(), labels.data_ptr<float>(),
-                 q.data_ptr<float>(), Q, N, gscale, dout.data_ptr<float>(),
-                 cur_stream());
+                 q.data_ptr<float>(), Q, N, gc.data_ptr<float>(), inv_n,
+                 dout.data_ptr<float>(), cur_stream());
   return dout;
 }
 
@@ -157,6 +165,23 @@ void fused_adam(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
   dr_fused_adam(meta_t.data_ptr<int64_t>(), nt, total, (float)lr, (float)beta1,
                 (float)beta2, (float)eps, (float)weight_decay, (int)step,
                 cur_stream());
+}
+
+// hipGraph-capturable variant: the pointer table is prebuilt on device (ops/
+// adam.py caches it while pointers are stable) and the step counter is a
+// device int32 scalar, so a replayed graph keeps exact bias correction with
+// zero host work per step.
+void fused_adam_capturable(at::Tensor meta_dev, int64_t nt, int64_t total,
+                           at::Tensor step_t, double lr, double beta1, double beta2,
+                           double eps, double weight_decay) {
+  const at::cuda::CUDAGuard guard(meta_dev.device());
+  TORCH_CHECK(meta_dev.is_cuda() && meta_dev.scalar_type() == at::kLong &&
+              meta_dev.is_contiguous());
+  TORCH_CHECK(step_t.is_cuda() && step_t.scalar_type() == at::kInt &&
+              step_t.numel() == 1);
+  dr_fused_adam_dev(meta_dev.data_ptr<int64_t>(), (int)nt, total, (float)lr,
+                    (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+                    step_t.data_ptr<int>(), cur_stream());
 }
 
 // -------------------------------------------------------------------- gru
@@ -299,6 +324,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pinball_forward", &pinball_forward);
   m.def("pinball_backward", &pinball_backward);
   m.def("fused_adam", &fused_adam);
+  m.def("fused_adam_capturable", &fused_adam_capturable);
   m.def("gru_seq_forward", &gru_seq_forward, py::arg("xg"), py::arg("w_hh"),
         py::arg("b_hh"), py::arg("h0"), py::arg("gamma"), py::arg("beta"),
         py::arg("reverse"), py::arg("save"), py::arg("fp8") = false);

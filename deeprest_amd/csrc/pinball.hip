@@ -43,8 +43,10 @@ template <int QMAX>
 __global__ void pinball_bwd_kernel(const float* __restrict__ out,
                                    const float* __restrict__ labels,
                                    const float* __restrict__ quantiles, int Q,
-                                   int64_t N, float gscale,
-                                   float* __restrict__ dout) {
+                                   int64_t N, const float* __restrict__ grad_loss,
+                                   float inv_n, float* __restrict__ dout) {
+  // upstream grad read on-device (no host .item() sync -> hipGraph-capturable)
+  const float gscale = grad_loss[0] * inv_n;
   float q[QMAX];
   for (int i = 0; i < Q; ++i) q[i] = quantiles[i];
   for (int64_t n = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; n < N;
@@ -76,12 +78,13 @@ void dr_pinball_fwd(const float* out, const float* labels, const float* quantile
 }
 
 void dr_pinball_bwd(const float* out, const float* labels, const float* quantiles,
-                    int Q, int64_t N, float gscale, float* dout, hipStream_t stream) {
+                    int Q, int64_t N, const float* grad_loss, float inv_n,
+                    float* dout, hipStream_t stream) {
   const int block = 256;
   int grid = (int)std::min<int64_t>((N + block - 1) / block, 2048);
   if (grid == 0) grid = 1;
   hipLaunchKernelGGL((dr::pinball_bwd_kernel<8>), dim3(grid), dim3(block), 0, stream,
-                     out, labels, quantiles, Q, N, gscale, dout);
+                     out, labels, quantiles, Q, N, grad_loss, inv_n, dout);
 }
 
 }  // extern "C"

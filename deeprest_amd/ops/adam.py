@@ -53,12 +53,52 @@ def fused_adam_step(
 
 
 class FusedAdam(torch.optim.Optimizer):
-    """Optimizer wrapper over fused_adam_step (one kernel launch per step on GPU)."""
+    """Optimizer wrapper over fused_adam_step (one kernel launch per step on GPU).
+
+    ``capturable=True`` makes ``step()`` hipGraph-safe: the packed pointer
+    table is cached on device (rebuilt only if any pointer changes, which can
+    only happen outside capture) and the step counter is a device int32
+    scalar incremented by a captured add — so a replayed training-step graph
+    keeps exact Adam bias correction with zero host work.
+    """
 
     def __init__(self, params: Iterable[torch.Tensor], lr: float = 1e-3,
-                 betas=(0.9, 0.999), eps: float = 1e-8, weight_decay: float = 0.0):
+                 betas=(0.9, 0.999), eps: float = 1e-8, weight_decay: float = 0.0,
+                 capturable: bool = False):
         defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
         super().__init__(params, defaults)
+        if capturable and len(self.param_groups) > 1:
+            raise ValueError("capturable FusedAdam supports a single param group")
+        self.capturable = capturable
+        self._step_t: torch.Tensor | None = None   # device int32 scalar
+        self._meta: torch.Tensor | None = None     # device int64 pointer table
+        self._meta_key = None                      # pointer tuple behind _meta
+        self._meta_total = 0
+        self._meta_nt = 0
+
+    def _capturable_step(self, params, grads, m, v, group) -> None:
+        ext = require_native("fused_adam_step")
+        dev = params[0].device
+        key = tuple(t.data_ptr() for t in params + grads + m + v)
+        if self._meta is None or self._meta_key != key:
+            import itertools
+            nt = len(params)
+            ptrs = [t.data_ptr() for t in itertools.chain(params, grads, m, v)]
+            cum, total = [], 0
+            for p in params:
+                total += p.numel()
+                cum.append(total)
+            self._meta = torch.tensor(ptrs + cum, dtype=torch.int64, device=dev)
+            self._meta_key = key
+            self._meta_total = total
+            self._meta_nt = nt
+        if self._step_t is None:
+            self._step_t = torch.zeros(1, dtype=torch.int32, device=dev)
+        self._step_t += 1  # device add: captured, so replays keep counting
+        beta1, beta2 = group["betas"]
+        ext.fused_adam_capturable(self._meta, self._meta_nt, self._meta_total,
+                                  self._step_t, group["lr"], beta1, beta2,
+                                  group["eps"], group["weight_decay"])
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -82,6 +122,9 @@ class FusedAdam(torch.optim.Optimizer):
                 m.append(state["exp_avg"])
                 v.append(state["exp_avg_sq"])
             if not params:
+                continue
+            if self.capturable and params[0].is_cuda:
+                self._capturable_step(params, grads, m, v, group)
                 continue
             step = self.state[params[0]]["step"]
             beta1, beta2 = group["betas"]
