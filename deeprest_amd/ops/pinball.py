@@ -30,11 +30,26 @@ def reference_pinball_loss(outputs: torch.Tensor, labels: torch.Tensor,
     return l.sum(dim=-1).mean(dim=(0, 1)).mean()
 
 
+# device quantile tensors, cached per (device, quantiles): building one from a
+# python list is a pageable H2D copy, which is both per-step overhead and
+# forbidden inside hipGraph capture
+_Q_CACHE: dict = {}
+
+
+def _q_tensor(quantiles, device) -> torch.Tensor:
+    key = (str(device), tuple(quantiles))
+    q = _Q_CACHE.get(key)
+    if q is None:
+        q = torch.as_tensor(list(quantiles), dtype=torch.float32, device=device)
+        _Q_CACHE[key] = q
+    return q
+
+
 class _PinballLoss(torch.autograd.Function):
     @staticmethod
     def forward(ctx, outputs, labels, quantiles):
         ext = require_native("pinball_loss")
-        q = torch.as_tensor(list(quantiles), dtype=torch.float32, device=outputs.device)
+        q = _q_tensor(quantiles, outputs.device)
         loss = ext.pinball_forward(outputs, labels, q)
         ctx.save_for_backward(outputs, labels, q)
         return loss
