@@ -35,6 +35,29 @@ import torch
 
 from .native import require_native
 
+# constant pi-permutation index tensors, cached per device (rebuilding them
+# each backward call costs ~6 tiny kernel launches per direction per step)
+_IDX_CACHE: dict = {}
+
+
+def _pi_indices(device):
+    key = str(device)
+    got = _IDX_CACHE.get(key)
+    if got is None:
+        H = 128
+        m = torch.arange(3 * H, device=device)
+        g = m // H
+        q = m % H
+        col = (q % 8) * 16 + (q // 8)
+        natj = torch.where(m < 2 * H, g * H + col, 2 * H + col)
+        idx = torch.arange(4 * H, device=device)
+        gg = idx // H
+        qq = idx % H
+        nat4 = gg * H + (qq % 8) * 16 + (qq // 8)  # dpre-row unpermute, 4H
+        got = (natj, nat4)
+        _IDX_CACHE[key] = got
+    return got
+
 
 def reference_gru_sequence(
     x_gates: torch.Tensor,       # (B, T, 3H) precomputed input gates
@@ -91,15 +114,11 @@ class _FusedGRUSequence(torch.autograd.Function):
         x_gates, w_hh, b_hh, h0, gamma, beta, h_all, saves = ctx.saved_tensors
         reverse = ctx.reverse
         # pi-permuted W image (H, 3H) for the dh GEMM: row k holds
-        # W[natJ(m), k] over the MFMA K axis m (dr | dz | d_hhn regions)
-        H = w_hh.shape[1]
-        m = torch.arange(3 * H, device=w_hh.device)
-        g = m // H
-        q = m % H
-        col = (q % 8) * 16 + (q // 8)
-        natj = torch.where(m < 2 * H, g * H + col, 2 * H + col)
-        w_img = w_hh[natj, :].t().contiguous().to(torch.bfloat16)
+        # W[natJ(m), k] over the MFMA K axis m (dr | dz | d_hhn regions).
+        # Cast once, then gather/transpose in bf16 (fewer/smaller copies).
+        natj, _ = _pi_indices(w_hh.device)
         w_fwd = w_hh.to(torch.bfloat16).contiguous()  # hh_n recompute GEMM
+        w_img = w_fwd[natj, :].t().contiguous()
         # sequential chain (custom kernel): dpre = [dr_pre|dz_pre|dn_pre|d_hh_n]
         dpre, dh0 = ext.gru_seq_backward_kernel(
             grad_h_all.contiguous(), w_img, w_fwd, x_gates, gamma, beta, b_hh,
@@ -111,14 +130,11 @@ class _FusedGRUSequence(torch.autograd.Function):
         # dpre rows use the kernel's pi packing: position g*128 + cc*8 + nt
         # holds natural column g*128 + nt*16 + cc — unpermute dW rows after
         # the GEMM (tiny index_copy on a (K, H) matrix).
+        _, nat4 = _pi_indices(w_hh.device)
+
         def unpi(dw_pi):
-            K = dw_pi.shape[0]
-            idx = torch.arange(K, device=dw_pi.device)
-            g = idx // 128
-            q = idx % 128
-            nat = g * 128 + (q % 8) * 16 + (q // 8)
             out = torch.empty_like(dw_pi)
-            out.index_copy_(0, nat, dw_pi)
+            out.index_copy_(0, nat4[: dw_pi.shape[0]], dw_pi)
             return out
 
         # dW_hh = sum over (b,t,c) of [dr|dz|d_hhn]^T h_prev — batched strided
