@@ -117,7 +117,8 @@ def main():
             autocast_dtype=autocast_dtype if use_autocast else None, warmup=2)
 
     def step(i: int):
-        s = (i * B) % max(n - B, 1)
+        # each rank walks a different offset sequence (its DP shard)
+        s = ((i + 3 * rank) * B + rank * 17) % max(n - B, 1)
         xb, yb = X[s : s + B], y[s : s + B]
         if graphed is not None:
             return graphed.run(xb, yb)
