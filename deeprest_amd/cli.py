@@ -108,7 +108,7 @@ def cmd_serve(args):
 
     from .serve.api import create_app
 
-    app = create_app(checkpoint_path=args.checkpoint)
+    app = create_app(checkpoint_path=args.checkpoint, results_path=args.results)
     uvicorn.run(app, host=args.host, port=args.port)
     return 0
 
@@ -138,6 +138,7 @@ def main(argv=None):
 
     v = sub.add_parser("serve")
     v.add_argument("--checkpoint", default=None)
+    v.add_argument("--results", default=None, help="results.pkl to browse at /results")
     v.add_argument("--host", default="0.0.0.0")
     v.add_argument("--port", type=int, default=2021)
 

@@ -13,6 +13,9 @@ Endpoints (FastAPI; run with `uvicorn deeprest_amd.serve.api:create_app`):
 - POST /anomaly                   sanity-check measured series against the
                                   model's quantile band
 - GET  /apis                      known API endpoints (for what-if queries)
+- GET  /results[/{exp}[/{comp}/{metric}]]  browse a results.pkl store (the
+                                  reference web-demo's DataLoader surface,
+                                  web-demo/dataloader.py:30-167, as REST)
 
 The app holds an IngestStore plus an optional Predictor loaded from a
 checkpoint; everything is JSON-serializable.
@@ -53,14 +56,52 @@ class IngestStore:
         return data
 
 
-def create_app(checkpoint_path: Optional[str] = None, predictor: Optional[Predictor] = None):
+def create_app(checkpoint_path: Optional[str] = None, predictor: Optional[Predictor] = None,
+               results_path: Optional[str] = None):
     from fastapi import FastAPI, HTTPException
 
     app = FastAPI(title="deeprest-amd", version="0.1.0")
     store = IngestStore()
-    state = {"predictor": predictor}
+    state = {"predictor": predictor, "results": None}
     if checkpoint_path and predictor is None:
         state["predictor"] = Predictor.from_checkpoint(checkpoint_path)
+    if results_path:
+        from .results import ResultsStore
+
+        state["results"] = ResultsStore.load(results_path)
+
+    # ---- results browsing (the reference web-demo's DataLoader surface over
+    # results.pkl — web-demo/dataloader.py:30-167 — as REST) ----
+    def _results():
+        rs = state["results"]
+        if rs is None:
+            raise HTTPException(status_code=400, detail="no results loaded "
+                                "(serve with results_path=...)")
+        return rs
+
+    @app.get("/results")
+    def results_index():
+        return {"experiments": _results().experiments()}
+
+    @app.get("/results/{experiment}")
+    def results_experiment(experiment: str):
+        rs = _results()
+        if experiment not in rs.results:
+            raise HTTPException(status_code=404, detail="unknown experiment")
+        return {
+            comp: sorted(metrics.keys())
+            for comp, metrics in rs.results[experiment].items()
+        }
+
+    @app.get("/results/{experiment}/{component}/{metric}")
+    def results_entry(experiment: str, component: str, metric: str):
+        rs = _results()
+        try:
+            entry = rs.get(experiment, component, metric)
+        except KeyError:
+            raise HTTPException(status_code=404, detail="unknown entry")
+        return {k: (v.tolist() if isinstance(v, np.ndarray) else v)
+                for k, v in entry.items()}
 
     @app.get("/demo")
     def demo():

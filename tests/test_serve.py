@@ -80,6 +80,34 @@ def test_results_store_schema(tmp_path):
     assert loaded.get("exp1-waves_seen-1x", "frontend", "cpu")["scale_ours"] == entry["scale_ours"]
 
 
+def test_rest_results_browsing(tmp_path):
+    """GET /results mirrors the web-demo DataLoader surface over results.pkl."""
+    from starlette.testclient import TestClient
+
+    from deeprest_amd.serve.api import create_app
+
+    measurement = np.abs(np.random.default_rng(0).normal(50, 10, size=300))
+    preds = {est: np.abs(np.random.default_rng(1).normal(50, 10, size=(2, 60)))
+             for est in ("bl-resrc", "bl-api", "bl-trace", "ours")}
+    entry = build_results_entry(measurement, preds, calls=[np.arange(300)],
+                                train_len=180)
+    store = ResultsStore()
+    store.add("exp1-waves_seen-1x", "frontend", "cpu", entry)
+    p = str(tmp_path / "results.pkl")
+    store.save(p)
+
+    client = TestClient(create_app(results_path=p))
+    assert client.get("/results").json() == {"experiments": ["exp1-waves_seen-1x"]}
+    assert client.get("/results/exp1-waves_seen-1x").json() == {"frontend": ["cpu"]}
+    got = client.get("/results/exp1-waves_seen-1x/frontend/cpu").json()
+    assert "prediction_ours" in got and "scale_groundtruth" in got
+    assert len(got["measurement"]) == 300
+    assert client.get("/results/nope").status_code == 404
+    # without a results store the endpoints answer 400, not crash
+    bare = TestClient(create_app())
+    assert bare.get("/results").status_code == 400
+
+
 def test_anomaly_scorer_flags_injected_cpu_thief():
     T = 100
     rng = np.random.default_rng(2)
