@@ -41,6 +41,7 @@ class TrainConfig:
     baseline_epochs: int = 100            # baselines.py:57
     run_baselines: bool = True
     dtype: str = "bf16"                   # compute dtype on GPU
+    graph_step: bool = False              # hipGraph-capture the train step
     seed: int = 0
     checkpoint_path: Optional[str] = None
     resume: bool = False

@@ -69,7 +69,16 @@ class Trainer:
             spec = build_model_spec(data)
             model = DeepRestNet(spec, self.cfg.model)
         self.model = model.to(self.device)
-        self.optimizer = FusedAdam(self.model.parameters(), lr=self.cfg.train.lr)
+        # graph_step wants the capturable optimizer (device-side step counter);
+        # capturable mode is also correct for plain eager stepping
+        self._use_graph = bool(
+            self.cfg.train.graph_step
+            and self.device.type == "cuda"
+            and (dist_ctx is None or dist_ctx.world_size == 1)
+        )
+        self.optimizer = FusedAdam(self.model.parameters(), lr=self.cfg.train.lr,
+                                   capturable=self._use_graph)
+        self._graphed = None
         self.feature_space_state = (
             data.feature_space.state_dict() if data.feature_space is not None else None
         )
@@ -137,17 +146,35 @@ class Trainer:
             for s in range(0, len(perm), cfg.batch_size):
                 idx = perm[s : s + cfg.batch_size].to(self.device)
                 xb, yb = X_train[idx], y_train[idx]
-                with torch.autocast(
-                    device_type=self.device.type, dtype=self.autocast_dtype,
-                    enabled=(self.device.type == "cuda"),
-                ):
-                    out = self.model(xb)
-                    loss = self.model.loss(out.float(), yb)
-                self.optimizer.zero_grad(set_to_none=True)
-                loss.backward()
-                if self.dist is not None:
-                    self.dist.all_reduce_gradients(self.model)
-                self.optimizer.step()
+                full = xb.shape[0] == cfg.batch_size
+                if self._use_graph and self._graphed is None and full:
+                    from .graphstep import GraphedTrainStep
+
+                    # capture once on the first full batch (warmup replays are
+                    # real optimizer steps on that batch); tail batches and
+                    # capture failure fall back to eager below
+                    self._graphed = GraphedTrainStep.build(
+                        self.model, self.optimizer,
+                        lambda o, t: self.model.loss(o.float(), t), xb, yb,
+                        autocast_dtype=self.autocast_dtype)
+                    if self._graphed is None:
+                        self._use_graph = False
+                if self._graphed is not None and full:
+                    loss = self._graphed.run(xb, yb)
+                else:
+                    with torch.autocast(
+                        device_type=self.device.type, dtype=self.autocast_dtype,
+                        enabled=(self.device.type == "cuda"),
+                    ):
+                        out = self.model(xb)
+                        loss = self.model.loss(out.float(), yb)
+                    # with an active graph, grads must keep their captured
+                    # addresses: zero in place instead of dropping to None
+                    self.optimizer.zero_grad(set_to_none=self._graphed is None)
+                    loss.backward()
+                    if self.dist is not None:
+                        self.dist.all_reduce_gradients(self.model)
+                    self.optimizer.step()
                 losses.append(loss.item())
                 total_samples += xb.shape[0] * self.world_size
             result.train_losses.append(float(np.mean(losses)))

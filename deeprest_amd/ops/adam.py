@@ -93,12 +93,25 @@ class FusedAdam(torch.optim.Optimizer):
             self._meta_total = total
             self._meta_nt = nt
         if self._step_t is None:
-            self._step_t = torch.zeros(1, dtype=torch.int32, device=dev)
+            # resume-consistent: host bookkeeping was already advanced for
+            # this call, so seed the device counter one behind it
+            start = int(self.state[params[0]]["step"]) - 1
+            self._step_t = torch.full((1,), start, dtype=torch.int32, device=dev)
         self._step_t += 1  # device add: captured, so replays keep counting
         beta1, beta2 = group["betas"]
         ext.fused_adam_capturable(self._meta, self._meta_nt, self._meta_total,
                                   self._step_t, group["lr"], beta1, beta2,
                                   group["eps"], group["weight_decay"])
+
+    def state_dict(self):
+        # graph replays advance only the device counter; sync the host step
+        # bookkeeping before checkpointing
+        if self._step_t is not None:
+            s = int(self._step_t.item())
+            for st in self.state.values():
+                if "step" in st:
+                    st["step"] = s
+        return super().state_dict()
 
     @torch.no_grad()
     def step(self, closure=None):
