@@ -296,6 +296,33 @@ def test_graphed_train_step(dev):
     assert abs(graphed[-1] - eager[-1]) < 0.25 * abs(eager[0] - eager[-1]) + 1e-3
 
 
+def test_trainer_graph_step_gpu(dev):
+    """Trainer with graph_step=True captures and keeps training correctly."""
+    import numpy as np
+
+    from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+    from deeprest_amd.engine.config import EngineConfig
+    from deeprest_amd.engine.trainer import Trainer
+    from deeprest_amd.models.net import DeepRestNetConfig
+
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=6, n_components=6, windows_per_day=160, n_days=1, seed=3))
+    data = app.generate_featurized()
+    cfg = EngineConfig()
+    cfg.train.epochs = 2
+    cfg.train.batch_size = 16
+    cfg.train.graph_step = True
+    cfg.train.run_baselines = False
+    cfg.train.log_every = 0
+    cfg.model = DeepRestNetConfig(d_model=64, n_heads=2, n_layers=1, d_ff=128,
+                                  hidden=128, comp_dim=16, dropout=0.0)
+    tr = Trainer(data, cfg, device=dev)
+    res = tr.train()
+    assert tr._graphed is not None, "graph was not captured"
+    assert np.isfinite(res.train_losses).all()
+    assert res.train_losses[-1] < res.train_losses[0] * 1.5
+
+
 # --------------------------------------------------------------- predictor
 def test_predictor_hipgraph_capture(dev):
     import numpy as np
