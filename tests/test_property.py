@@ -1,0 +1,119 @@
+"""Property-based tests (hypothesis) for the data layer.
+
+Randomized span trees / series exercise invariants the example-based suites
+can't sweep: arbitrary trace shapes and depths for the call-path feature
+space, arbitrary series for windowing and normalization.
+"""
+
+import numpy as np
+from hypothesis import given, settings, strategies as st
+
+from deeprest_amd.data.featurize import FeatureSpace
+from deeprest_amd.data.windows import MinMaxScaler, sliding_window
+from deeprest_amd.ops.pinball import reference_pinball_loss
+
+import torch
+
+
+# ---- random span-tree strategy (bounded depth/fanout) ----
+def _span(depth):
+    base = st.fixed_dictionaries({
+        "component": st.sampled_from(["svc-a", "svc-b", "svc-c", "db"]),
+        "operation": st.sampled_from(["read", "write", "compose"]),
+    })
+    if depth == 0:
+        return base.map(lambda d: {**d, "children": []})
+    child = _span(depth - 1)
+    return st.tuples(base, st.lists(child, max_size=3)).map(
+        lambda t: {**t[0], "children": list(t[1])}
+    )
+
+
+def _all_prefix_paths(trace, prefix=()):
+    """Brute-force enumeration of every root-to-node call path."""
+    p = prefix + ((trace["component"], trace["operation"]),)
+    out = [p]
+    for ch in trace.get("children", []):
+        out.extend(_all_prefix_paths(ch, p))
+    return out
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(_span(3), min_size=1, max_size=6))
+def test_feature_space_indexes_every_prefix_path(traces):
+    fs = FeatureSpace()
+    for tr in traces:
+        fs.observe_trace(tr)
+    want = set()
+    for tr in traces:
+        want.update(_all_prefix_paths(tr))
+    assert len(fs) == len(want)
+
+    # counting: per trace, vector total == number of spans; per path, count ==
+    # number of occurrences of that exact prefix path
+    for tr in traces:
+        v = np.zeros(len(fs), dtype=np.int64)
+        fs.count_trace(tr, v)
+        paths = _all_prefix_paths(tr)
+        assert v.sum() == len(paths)
+        for p in set(paths):
+            assert v[fs.index_of(p)] == paths.count(p)
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.lists(_span(2), min_size=1, max_size=4), st.lists(_span(2), min_size=1, max_size=4))
+def test_frozen_feature_space_never_grows(seen, unseen):
+    fs = FeatureSpace()
+    for tr in seen:
+        fs.observe_trace(tr)
+    n = len(fs)
+    # counting unseen traces against the frozen space must not grow it and
+    # must only count paths that already exist (reference featurize.py:27-33
+    # drops unseen paths at extract time)
+    for tr in unseen:
+        v = np.zeros(n, dtype=np.int64)
+        fs.count_trace(tr, v)
+        seen_paths = {p for t in seen for p in _all_prefix_paths(t)}
+        hits = [p for p in _all_prefix_paths(tr) if p in seen_paths]
+        assert v.sum() == len(hits)
+    assert len(fs) == n
+
+
+@settings(max_examples=30, deadline=None)
+@given(
+    st.integers(min_value=2, max_value=40).flatmap(
+        lambda n: st.tuples(
+            st.just(n),
+            st.integers(min_value=1, max_value=n - 1),
+            st.lists(st.floats(-1e3, 1e3, allow_nan=False), min_size=n, max_size=n),
+        )
+    )
+)
+def test_sliding_window_matches_naive(args):
+    n, w, vals = args
+    ts = np.asarray(vals, dtype=np.float64)
+    got = sliding_window(ts, w)
+    assert got.shape == (n - w, w)
+    for i in range(n - w):
+        np.testing.assert_array_equal(got[i], ts[i : i + w])
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.lists(st.floats(0, 1e6, allow_nan=False), min_size=4, max_size=64))
+def test_minmax_roundtrip(vals):
+    M = np.asarray(vals, dtype=np.float64).reshape(-1, 1)
+    sc = MinMaxScaler().fit(M, split=max(2, len(vals) // 2))
+    z = sc.transform(M)
+    back = sc.inverse_transform(z)
+    np.testing.assert_allclose(back, M, rtol=1e-9, atol=1e-6)
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(0, 2**31 - 1))
+def test_pinball_zero_at_exact_prediction_and_nonnegative(seed):
+    rng = np.random.default_rng(seed)
+    y = torch.from_numpy(rng.normal(size=(2, 5, 3)).astype(np.float32))
+    exact = y.unsqueeze(-1).repeat(1, 1, 1, 3)
+    assert float(reference_pinball_loss(exact, y, (0.05, 0.5, 0.95))) == 0.0
+    off = exact + torch.from_numpy(rng.normal(size=exact.shape).astype(np.float32))
+    assert float(reference_pinball_loss(off, y, (0.05, 0.5, 0.95))) >= 0.0
