@@ -79,6 +79,32 @@ def test_frozen_feature_space_never_grows(seen, unseen):
     assert len(fs) == n
 
 
+@settings(max_examples=25, deadline=None)
+@given(st.lists(_span(3), min_size=1, max_size=5),
+       st.lists(_span(3), min_size=1, max_size=5))
+def test_native_featurizer_property_parity(train_traces, new_traces):
+    """The C++ trie featurizer must agree with the Python FeatureSpace on
+    arbitrary span trees — both while growing the space and frozen."""
+    from deeprest_amd.data.featurize import Featurizer, _native_featurize
+
+    if _native_featurize() is None:
+        return  # extension not built in this environment
+
+    def windows(traces):
+        return [{"metrics": [{"component": "svc-a", "resource": "cpu",
+                              "value": 1.0}],
+                 "traces": traces}]
+
+    train = windows(train_traces)
+    fresh = windows(new_traces)
+    py = Featurizer(use_native=False).fit(train)
+    nat = Featurizer(use_native=True).fit(train)
+    assert py.feature_space.paths == nat.feature_space.paths
+    dp = py.transform(train + fresh)   # frozen space: unseen paths dropped
+    dn = nat.transform(train + fresh)
+    np.testing.assert_array_equal(dp.traffic, dn.traffic)
+
+
 @settings(max_examples=30, deadline=None)
 @given(
     st.integers(min_value=2, max_value=40).flatmap(
