@@ -6,7 +6,11 @@ Run on a GPU box:  python tools/bench_serve.py [--windows 1000] [--iters 30]
 """
 import argparse
 import json
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 import numpy as np
 import torch
