@@ -22,7 +22,7 @@ Prometheus sample shape:
 from __future__ import annotations
 
 from collections import defaultdict
-from typing import Any, Dict, Iterable, List, Optional, Sequence, Tuple
+from typing import Any, Dict, Iterable, List, Optional, Tuple
 
 
 def span_tree_from_jaeger(trace: Dict[str, Any]) -> Optional[Dict[str, Any]]:

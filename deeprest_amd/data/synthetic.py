@@ -30,13 +30,13 @@ Two output paths:
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, Dict, List, Optional, Sequence, Tuple
 
 import numpy as np
 
 from .contract import validate_raw_data
-from .featurize import FeatureSpace, Featurizer, FeaturizedData
+from .featurize import FeatureSpace, FeaturizedData
 
 DEFAULT_RESOURCES = ("cpu", "memory", "write-iops")
 

@@ -14,7 +14,7 @@ Reproduces the reference's data preparation exactly
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import List
 
 import numpy as np
 import torch

@@ -16,11 +16,7 @@ import numpy as np
 import torch
 
 from ..data.featurize import FeaturizedData, Featurizer
-from ..models.baselines import (
-    ComponentAwareBaseline,
-    ResourceAwareBaseline,
-    TraceAwareBaseline,
-)
+from ..models.baselines import TraceAwareBaseline
 from ..serve.results import ResultsStore, build_results_entry
 from .config import EngineConfig
 from .trainer import Trainer

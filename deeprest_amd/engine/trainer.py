@@ -23,7 +23,7 @@ import torch
 
 from ..data.featurize import FeaturizedData
 from ..models.baselines import ComponentAwareBaseline, ResourceAwareBaseline
-from ..models.net import DeepRestNet, DeepRestNetConfig, build_model_spec
+from ..models.net import DeepRestNet, build_model_spec
 from ..ops.adam import FusedAdam
 from ..utils.errors import error_percentiles, format_error_table
 from .checkpoint import load_checkpoint, save_checkpoint

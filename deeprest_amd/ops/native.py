@@ -8,7 +8,7 @@ travels with the repo snapshot to GPU machines — no JIT cache dependence.
 from __future__ import annotations
 
 import importlib
-from typing import Optional
+
 
 _EXT = None
 _TRIED = False
