@@ -1,0 +1,42 @@
+"""bench.py driver-contract test.
+
+The round-end driver parses exactly one JSON line from rank 0 with a fixed
+key set (see the repo instructions); this test locks that contract on the
+CPU path so kernel/bench edits can't silently break it.
+"""
+
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_bench_json_contract():
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+         "--batch", "8", "--endpoints", "6", "--components", "5",
+         "--seq-len", "12"],
+        cwd=REPO, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, f"expected exactly one JSON line, got: {out.stdout!r}"
+    d = json.loads(lines[0])
+
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in d, f"missing contract key {key}"
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    assert d["steps"] == 2 and d["warmup"] == 1
+    assert d["higher_is_better"] is True
+    assert d["scaling"] == "weak"
+    assert d["data"] == "synthetic"
+    assert d["vs_baseline"] is None  # BASELINE.json publishes no number
+    cfg = d["config"]
+    for key in ("model", "global_batch", "seq_len", "parallelism"):
+        assert key in cfg
+    assert cfg["global_batch"] == 8
+    assert cfg["parallelism"] == "dp1"
