@@ -115,3 +115,45 @@ def test_forward_long_streaming_chunked_runs_and_is_finite():
     # recurrent state carries across chunks: chunked differs from
     # chunk-local-only encode but must stay in a sane range
     assert out.abs().max() < 1e3
+
+
+def test_gru_chunked_state_carry_exact_both_directions():
+    """forward_long's decoder recurrence is EXACT across chunk boundaries:
+    chunked sweeps with state carry-over reproduce the full-sequence GRU
+    bit-for-bit in both directions (models/net.py forward_long)."""
+    import torch
+
+    from deeprest_amd.ops.gru import fused_gru_sequence
+
+    torch.manual_seed(0)
+    B, T, C, H = 2, 37, 3, 16
+    xg = torch.randn(B, T, 3 * H)
+    w_hh = torch.randn(3 * H, H) * 0.2
+    b_hh = torch.randn(3 * H) * 0.1
+    h0 = torch.randn(B, C, H) * 0.5
+    gamma = torch.randn(C, 3 * H) * 0.3 + 1.0
+    beta = torch.randn(C, 3 * H) * 0.1
+
+    bounds = [0, 10, 20, 30, T]  # uneven chunks incl. a short tail
+    chunks = list(zip(bounds[:-1], bounds[1:]))
+
+    full_f = fused_gru_sequence(xg, w_hh, b_hh, h0, gamma, beta, reverse=False)
+    h = h0
+    parts = []
+    for s, e in chunks:
+        out = fused_gru_sequence(xg[:, s:e], w_hh, b_hh, h, gamma, beta,
+                                 reverse=False)
+        h = out[:, -1].contiguous()
+        parts.append(out)
+    assert torch.equal(torch.cat(parts, dim=1), full_f)
+
+    full_r = fused_gru_sequence(xg, w_hh, b_hh, h0, gamma, beta, reverse=True)
+    h = h0
+    parts_r = [None] * len(chunks)
+    for ci in range(len(chunks) - 1, -1, -1):
+        s, e = chunks[ci]
+        out = fused_gru_sequence(xg[:, s:e], w_hh, b_hh, h, gamma, beta,
+                                 reverse=True)
+        h = out[:, 0].contiguous()
+        parts_r[ci] = out
+    assert torch.equal(torch.cat(parts_r, dim=1), full_r)
