@@ -133,6 +133,10 @@ class Trainer:
         y_train = ds.y_train.to(self.device)
         n = X_train.shape[0]
         gen = torch.Generator().manual_seed(cfg.seed)
+        # resume determinism: replay the skipped epochs' permutation draws so
+        # a resumed run sees the same batch order as an uninterrupted one
+        for _ in range(self.start_epoch):
+            torch.randperm(n, generator=gen)
 
         total_samples = 0
         t_start = time.perf_counter()

@@ -108,3 +108,33 @@ def test_config_cli_overrides():
     out = apply_cli_overrides(cfg, ["train.epochs=9", "model.d_model=128"])
     assert out.train.epochs == 9
     assert out.model.d_model == 128
+
+
+def test_resume_reproduces_uninterrupted_run(tmp_path):
+    """3 epochs straight == 2 epochs + checkpoint + resume for 1 more:
+    identical final weights (the resumed run replays skipped permutation
+    draws, so batch order matches)."""
+    data = tiny_data()
+
+    (tmp_path / "a").mkdir()
+    (tmp_path / "b").mkdir()
+    cfg_a = tiny_config(tmp_path / "a", epochs=3)
+    cfg_a.train.run_baselines = False
+    torch.manual_seed(42)  # identical model init for both runs
+    ta = Trainer(data, cfg_a, device=torch.device("cpu"))
+    ta.train()
+
+    cfg_b1 = tiny_config(tmp_path / "b", epochs=2)
+    cfg_b1.train.run_baselines = False
+    torch.manual_seed(42)
+    tb1 = Trainer(data, cfg_b1, device=torch.device("cpu"))
+    tb1.train()
+    cfg_b2 = tiny_config(tmp_path / "b", epochs=3)
+    cfg_b2.train.run_baselines = False
+    cfg_b2.train.resume = True
+    tb2 = Trainer(data, cfg_b2, device=torch.device("cpu"))
+    tb2.train()
+
+    wa = torch.cat([p.detach().reshape(-1) for p in ta.model.parameters()])
+    wb = torch.cat([p.detach().reshape(-1) for p in tb2.model.parameters()])
+    assert torch.equal(wa, wb)
