@@ -135,22 +135,38 @@ class SyntheticApp:
         P = len(self.feature_space)
 
         # precompute per-shape path-count and component-count vectors
-        self._shape_vec: Dict[str, np.ndarray] = {}
+        # per-api path-count matrices are stored COMPACT: an api's shapes only
+        # touch its own ~dozen call paths, so (cols, (S, K) counts) instead of
+        # a dense (S, P) row keeps construction and the traffic accumulation
+        # linear in n_apis (dense was quadratic: P itself grows with n_apis —
+        # 20 GB of zeros and ~13 min of matmuls at the 4096-endpoint config)
+        self._shape_vec: Dict[str, tuple] = {}
         self._shape_comp: Dict[str, np.ndarray] = {}
         all_components = [self.frontend] + self.components
         self.component_index = {c: i for i, c in enumerate(all_components)}
         self.all_components = all_components
         for api in self.apis:
-            vecs = np.zeros((config.shapes_per_api, P), dtype=np.int64)
+            per_shape: List[Dict[int, int]] = []
             comps = np.zeros((config.shapes_per_api, len(all_components)), dtype=np.int64)
             for s, shape in enumerate(self.shapes[api]):
-                self.feature_space.count_trace(shape, vecs[s])
-                stack = [shape]
+                counts: Dict[int, int] = {}
+                stack = [(shape, ())]
                 while stack:
-                    node = stack.pop()
+                    node, prefix = stack.pop()
+                    path = prefix + ((node["component"], node["operation"]),)
+                    idx = self.feature_space.index_of(path)
+                    counts[idx] = counts.get(idx, 0) + 1
                     comps[s, self.component_index[node["component"]]] += 1
-                    stack.extend(node.get("children", []))
-            self._shape_vec[api] = vecs
+                    for child in node.get("children", []):
+                        stack.append((child, path))
+                per_shape.append(counts)
+            cols = sorted(set().union(*per_shape))
+            col_pos = {c: i for i, c in enumerate(cols)}
+            mat = np.zeros((config.shapes_per_api, len(cols)), dtype=np.int64)
+            for s, counts in enumerate(per_shape):
+                for c, v in counts.items():
+                    mat[s, col_pos[c]] = v
+            self._shape_vec[api] = (np.asarray(cols, dtype=np.int64), mat)
             self._shape_comp[api] = comps
 
         # ground-truth resource model parameters per (component, resource).
@@ -285,7 +301,8 @@ class SyntheticApp:
         inv = np.zeros((T, C), dtype=np.int64)
         drive = np.zeros((T, C))
         for api in self.apis:
-            traffic += shape_counts[api] @ self._shape_vec[api]
+            cols, mat = self._shape_vec[api]
+            traffic[:, cols] += shape_counts[api] @ mat
             inv += shape_counts[api] @ self._shape_comp[api]
             drive += shape_counts[api] @ self._shape_comp_w[api]
         res = self._resources_from_invocations(drive)
