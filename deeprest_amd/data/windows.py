@@ -26,14 +26,17 @@ def sliding_window(ts: np.ndarray, window_size: int) -> np.ndarray:
     len-window_size is excluded, matching reference utils.py:4-5).
     Returns a contiguous copy of shape (N, window_size, *ts.shape[1:]).
     """
-    ts = np.asarray(ts)
+    ts = np.ascontiguousarray(ts)
     n = len(ts) - window_size
     if n <= 0:
         return np.empty((0, window_size) + ts.shape[1:], dtype=ts.dtype)
-    view = np.lib.stride_tricks.sliding_window_view(ts, window_size, axis=0)
-    # view shape: (len-ws+1, *rest, ws) -> take first n, move window axis to 1
-    out = np.moveaxis(view[:n], -1, 1)
-    return np.ascontiguousarray(out)
+    # explicit per-window slice copies: each is one contiguous memcpy, so the
+    # copy runs at memory bandwidth regardless of how numpy would iterate the
+    # overlapping strided view
+    out = np.empty((n, window_size) + ts.shape[1:], dtype=ts.dtype)
+    for i in range(n):
+        out[i] = ts[i : i + window_size]
+    return out
 
 
 def minmax_fit(M: np.ndarray, split: int) -> Tuple[float, float]:
