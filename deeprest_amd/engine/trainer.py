@@ -179,9 +179,13 @@ class Trainer:
                     if self.dist is not None:
                         self.dist.all_reduce_gradients(self.model)
                     self.optimizer.step()
-                losses.append(loss.item())
+                # keep the loss on-device: a per-batch .item() is a host sync
+                # that drains the GPU pipeline every step
+                losses.append(loss.detach().clone())
                 total_samples += xb.shape[0] * self.world_size
-            result.train_losses.append(float(np.mean(losses)))
+            result.train_losses.append(
+                float(torch.stack(losses).mean().item()) if losses else float("nan")
+            )
 
             if self.rank == 0:
                 test_loss, tables = self.evaluate(baselines)
