@@ -40,3 +40,23 @@ def test_bench_json_contract():
         assert key in cfg
     assert cfg["global_batch"] == 8
     assert cfg["parallelism"] == "dp1"
+
+
+def test_bench_torchrun_dp2_contract():
+    """The driver's N>1 launch path: torchrun --nnodes=1 --nproc-per-node 2
+    with MASTER_ADDR=127.0.0.1 (gloo on CPU here, RCCL on the GPU node)."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29537", "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--batch", "8",
+         "--endpoints", "6", "--components", "5", "--seq-len", "12"],
+        cwd=REPO, capture_output=True, text=True, timeout=420,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, f"exactly one JSON line from rank 0, got {lines}"
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["config"]["global_batch"] == 16  # whole-job aggregate
