@@ -78,3 +78,39 @@ def test_collector_jaeger_process_map():
     }
     tree = span_tree_from_jaeger(tr)
     assert tree["component"] == "svc-via-process"
+
+
+def test_span_tree_child_listed_before_parent():
+    """Two-pass reconstruction is order-independent (Jaeger export order is
+    arbitrary)."""
+    trace = {
+        "spans": [
+            {"spanID": "c", "operationName": "leaf", "serviceName": "svc-b",
+             "references": [{"refType": "CHILD_OF", "spanID": "r"}],
+             "startTime": 2_000_000},
+            {"spanID": "r", "operationName": "root", "serviceName": "svc-a",
+             "references": [], "startTime": 1_000_000},
+        ]
+    }
+    tree = span_tree_from_jaeger(trace)
+    assert tree["component"] == "svc-a"
+    assert tree["children"][0]["component"] == "svc-b"
+    assert tree["children"][0]["operation"] == "leaf"
+
+
+def test_span_tree_orphan_reference_becomes_root():
+    """A span whose parent never arrived is treated as the root (partial
+    traces happen under sampling/drops); extra orphans are dropped."""
+    trace = {
+        "spans": [
+            {"spanID": "x", "operationName": "op", "serviceName": "svc",
+             "references": [{"refType": "CHILD_OF", "spanID": "missing"}],
+             "startTime": 0},
+        ]
+    }
+    tree = span_tree_from_jaeger(trace)
+    assert tree is not None and tree["component"] == "svc"
+
+
+def test_collector_empty_returns_no_windows():
+    assert Collector().windows() == []
