@@ -77,11 +77,24 @@ def cmd_train(args):
 
 
 def cmd_experiment(args):
-    from .engine.experiment import run_experiment
+    from .engine.experiment import run_experiment, run_scenario_suite
 
     cfg = _load_config(args)
-    data = _load_data(cfg)
-    store = run_experiment(data, args.name, config=cfg)
+    if args.suite:
+        # unseen-traffic scenario suite needs a traffic-plan-capable app,
+        # i.e. the synthetic generator (external data has no counterfactuals)
+        from .data.synthetic import SyntheticApp, SyntheticAppConfig
+
+        d = cfg.data
+        app = SyntheticApp(SyntheticAppConfig(
+            n_apis=d.synth_apis, n_components=d.synth_components,
+            windows_per_day=d.synth_windows_per_day, n_days=d.synth_days,
+            seed=d.synth_seed,
+        ))
+        store = run_scenario_suite(app, base_name=args.name, config=cfg)
+    else:
+        data = _load_data(cfg)
+        store = run_experiment(data, args.name, config=cfg)
     store.save(args.out)
     print(f"experiment '{args.name}' written to {args.out}")
     return 0
@@ -128,6 +141,8 @@ def main(argv=None):
         if name == "experiment":
             t.add_argument("--name", required=True)
             t.add_argument("--out", default="results.pkl")
+            t.add_argument("--suite", action="store_true",
+                           help="unseen-traffic scenario suite (synthetic app)")
 
     s = sub.add_parser("synthesize")
     s.add_argument("--raw", required=True)

@@ -256,10 +256,16 @@ class SyntheticApp:
         return np.maximum(vals, 0.0)
 
     # ------------------------------------------------------------------ outputs
-    def generate_raw(self) -> List[Dict[str, Any]]:
-        """Full contract-format raw_data with span trees."""
+    def generate_raw(self, plan: Optional[np.ndarray] = None) -> List[Dict[str, Any]]:
+        """Full contract-format raw_data with span trees.
+
+        ``plan``: optional (T, n_apis) expected-call matrix from
+        ``traffic_plan(...)`` — pass a scenario variant (unseen scale /
+        shape / composition) to generate query-period data for what-if
+        evaluation (reference: locustfile-{scale,shape,composition}.py).
+        """
         cfg = self.config
-        api_calls = self.traffic_plan()
+        api_calls = self.traffic_plan() if plan is None else np.asarray(plan)
         shape_counts = self._sample_shape_counts(api_calls)
         T = cfg.n_windows
         C = len(self.all_components)
@@ -288,10 +294,14 @@ class SyntheticApp:
         validate_raw_data(raw)
         return raw
 
-    def generate_featurized(self) -> FeaturizedData:
-        """Fast path: traffic matrix + resource series without building trees."""
+    def generate_featurized(self, plan: Optional[np.ndarray] = None) -> FeaturizedData:
+        """Fast path: traffic matrix + resource series without building trees.
+
+        ``plan`` as in :meth:`generate_raw` — a scenario traffic plan for
+        query-period/what-if data sharing this app's feature space.
+        """
         cfg = self.config
-        api_calls = self.traffic_plan()
+        api_calls = self.traffic_plan() if plan is None else np.asarray(plan)
         shape_counts = self._sample_shape_counts(api_calls)
         T = cfg.n_windows
         P = len(self.feature_space)

@@ -88,3 +88,100 @@ def run_experiment_from_raw(
 ) -> ResultsStore:
     data = Featurizer().fit_transform(raw_data)
     return run_experiment(data, experiment_name, config=config, **kw)
+
+
+# The reference's evaluation scenarios (one locustfile variant each:
+# locustfile-{normal,scale,shape,composition}.py) -> traffic_plan kwargs.
+DEFAULT_SCENARIOS = [
+    ("waves_seen-1x", {}),
+    ("waves_unseen-3x", {"scale": 3.0}),
+    ("flat_unseen-1x", {"shape": "flat"}),
+    ("waves_unseen_compositions-1x", {"composition": "unseen"}),
+]
+
+
+def run_scenario_suite(
+    app,
+    base_name: str = "synthetic",
+    config: Optional[EngineConfig] = None,
+    device: Optional[torch.device] = None,
+    scenarios: Optional[Sequence] = None,
+) -> ResultsStore:
+    """Train once on the app's normal traffic, then evaluate every estimator
+    on UNSEEN query scenarios — the reference's headline capability (estimate
+    for 3x users, unseen shapes, unseen compositions; web-demo experiment
+    names `...-waves_{shape}-{seen|unseen}_compositions-{N}x`).
+
+    ``app``: a SyntheticApp (anything with traffic_plan/generate_featurized
+    sharing one feature space). Returns a ResultsStore with one experiment
+    per scenario.
+    """
+    from ..data.windows import sliding_window
+    from ..models.baselines import ComponentAwareBaseline
+
+    cfg = config or EngineConfig()
+    base = app.generate_featurized()
+    trainer = Trainer(base, cfg, device=device)
+    trainer.train()
+    ds = trainer.dataset
+    step = ds.step_size
+    spec = trainer.model.spec
+    median_q = len(trainer.model.cfg.quantiles) // 2
+    bl = trainer.run_baselines()            # RESRC predictions (history-only)
+    X_np = ds.X.numpy()
+
+    # fit the frozen estimators once on the BASE train split
+    trace_bls, comp_bls = [], []
+    for m in range(len(ds.metric_names)):
+        trace_bls.append(TraceAwareBaseline(split=ds.split).fit(
+            X_np, ds.y_raw[:, :, m]))
+        comp = spec.components[spec.comp_of[m]]
+        comp_bls.append(ComponentAwareBaseline(
+            component=comp, invocations=base.invocations,
+            window=step, split=ds.split).fit(ds.y_raw[:, :, m]))
+
+    store = ResultsStore()
+    scenarios = DEFAULT_SCENARIOS if scenarios is None else scenarios
+    for scen_name, plan_kw in scenarios:
+        if plan_kw.get("composition") == "unseen":
+            # rotate API popularity: mass moves to the tail APIs
+            plan_kw = dict(plan_kw)
+            plan_kw["composition"] = np.roll(app.popularity,
+                                             len(app.popularity) // 2)
+        qdata = app.generate_featurized(plan=app.traffic_plan(**plan_kw))
+        Xq = sliding_window(np.asarray(qdata.traffic, dtype=np.float32), step)
+        eval_idx = list(range(0, len(Xq), step))
+        Xq_eval = Xq[eval_idx]
+        xb = torch.from_numpy(Xq_eval.copy())
+        if ds.x_scaler.scale != 0.0:
+            xb -= ds.x_scaler.min_val
+            xb /= ds.x_scaler.scale
+        trainer.model.eval()
+        with torch.no_grad():
+            out = trainer.model(xb.to(trainer.device)).float().cpu().numpy()
+
+        n_flat = len(qdata.traffic)
+        for m, name in enumerate(ds.metric_names):
+            comp = spec.components[spec.comp_of[m]]
+            resource = spec.resources[spec.res_of[m]]
+            measurement = np.asarray(qdata.resources[name])
+            comp_windows = comp_bls[m].estimate_series(
+                qdata.invocations.get(comp, qdata.invocations["general"]),
+                n_flat)[: len(Xq)]
+            preds = {
+                # RESRC is a single repeated history window by construction
+                "bl-resrc": np.tile(bl["resrc"][0, :, m], (len(eval_idx), 1)),
+                "bl-api": comp_windows[eval_idx],
+                "bl-trace": trace_bls[m].estimate(Xq_eval),
+                "ours": ds.denormalize_metric(out[:, :, m, median_q], m),
+            }
+            preds = {k: np.maximum(np.asarray(v, dtype=np.float64), 1e-6)
+                     for k, v in preds.items()}
+            entry = build_results_entry(
+                measurement=measurement,
+                predictions=preds,
+                calls=None,
+                train_len=min(ds.split, len(measurement) - 1),
+            )
+            store.add(f"{base_name}-{scen_name}", comp, resource, entry)
+    return store

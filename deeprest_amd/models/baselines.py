@@ -117,22 +117,34 @@ class TraceAwareBaseline:
     def __init__(self, split: int, ridge: float = 1e-3) -> None:
         self.split = split
         self.ridge = ridge
+        self._w: Optional[np.ndarray] = None
 
-    def fit_and_estimate(self, X_windows: np.ndarray, y_windows: np.ndarray) -> np.ndarray:
-        """X_windows: (N, T, P); y_windows: (N, T). Returns (N - split, T)."""
+    def fit(self, X_windows: np.ndarray, y_windows: np.ndarray) -> "TraceAwareBaseline":
         X = np.asarray(X_windows, dtype=np.float64)
         y = np.asarray(y_windows, dtype=np.float64)
-        N, T, P = X.shape
+        P = X.shape[-1]
         Xf = X[: self.split].reshape(-1, P)
         yf = y[: self.split].reshape(-1)
         # bias column + ridge-regularized normal equations
         Xf = np.concatenate([Xf, np.ones((len(Xf), 1))], axis=1)
         A = Xf.T @ Xf + self.ridge * np.eye(P + 1)
-        w = np.linalg.solve(A, Xf.T @ yf)
-        Xt = X[self.split :].reshape(-1, P)
+        self._w = np.linalg.solve(A, Xf.T @ yf)
+        return self
+
+    def estimate(self, X_windows: np.ndarray) -> np.ndarray:
+        """(N, T, P) windows (any traffic, incl. unseen scenarios) -> (N, T)."""
+        assert self._w is not None, "fit first"
+        X = np.asarray(X_windows, dtype=np.float64)
+        N, T, P = X.shape
+        Xt = X.reshape(-1, P)
         Xt = np.concatenate([Xt, np.ones((len(Xt), 1))], axis=1)
-        pred = (Xt @ w).reshape(-1, T)
+        pred = (Xt @ self._w).reshape(-1, T)
         return np.maximum(pred, 1e-6)
+
+    def fit_and_estimate(self, X_windows: np.ndarray, y_windows: np.ndarray) -> np.ndarray:
+        """X_windows: (N, T, P); y_windows: (N, T). Returns (N - split, T)."""
+        self.fit(X_windows, y_windows)
+        return self.estimate(np.asarray(X_windows)[self.split :])
 
 
 class ComponentAwareBaseline:
@@ -147,8 +159,8 @@ class ComponentAwareBaseline:
             inv if inv is not None else invocations["general"], dtype=np.float64
         )
 
-    def fit_and_estimate(self, y_windows: np.ndarray) -> np.ndarray:
-        """y_windows: (N, window). Returns (N - split, window) estimates."""
+    def fit(self, y_windows: np.ndarray) -> "ComponentAwareBaseline":
+        """Fit the min-max scaling map (w1..w4) on the train portion."""
         y = np.asarray(y_windows, dtype=np.float64)
         # reconstruct the flat series from stride-1 windows
         ts = np.concatenate([y[:-1, 0], y[-1]]) if len(y) > 1 else y[0]
@@ -156,19 +168,28 @@ class ComponentAwareBaseline:
 
         inv_train = self.invocation[:split_flat]
         met_train = ts[:split_flat]
-        w1 = float(np.min(inv_train))
-        w3 = float(np.max(inv_train) - w1)
-        w4 = float(np.min(met_train))
-        w2 = float(np.max(met_train) - w4)
+        self.w1 = float(np.min(inv_train))
+        self.w3 = float(np.max(inv_train) - self.w1)
+        self.w4 = float(np.min(met_train))
+        self.w2 = float(np.max(met_train) - self.w4)
+        self._n_flat = len(ts)
+        return self
 
-        if self.invocation.sum() > 0 and w3 > 0:
-            ts_hat = (self.invocation - w1) * w2 / w3 + w4
+    def estimate_series(self, invocation: np.ndarray, n_flat: int) -> np.ndarray:
+        """Apply the fitted map to ANY invocation series (e.g. an unseen
+        scenario's); returns stride-1 windows over its first n_flat steps."""
+        inv = np.asarray(invocation, dtype=np.float64)
+        if inv.sum() > 0 and self.w3 > 0:
+            ts_hat = (inv - self.w1) * self.w2 / self.w3 + self.w4
         else:
-            ts_hat = self.invocation.astype(np.float64)
+            ts_hat = inv.astype(np.float64)
         ts_hat = np.maximum(ts_hat, 1e-6)
-
-        n_flat = len(ts)
-        windows = np.stack(
+        return np.stack(
             [ts_hat[i - self.window : i] for i in range(self.window, n_flat + 1)]
         )
+
+    def fit_and_estimate(self, y_windows: np.ndarray) -> np.ndarray:
+        """y_windows: (N, window). Returns (N - split, window) estimates."""
+        self.fit(y_windows)
+        windows = self.estimate_series(self.invocation, self._n_flat)
         return windows[self.split :]

@@ -39,3 +39,47 @@ def test_run_experiment_produces_results_schema(tmp_path):
     store.save(p)
     loaded = ResultsStore.load(p)
     assert loaded.get("synthetic-waves_seen-1x", app.all_components[0], "cpu")
+
+
+def test_scenario_suite_unseen_traffic():
+    """Train once on normal traffic, query under unseen scale/shape/
+    composition — the reference's headline evaluation axes (locustfile
+    variants); one results.pkl experiment per scenario."""
+    import numpy as np
+
+    from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+    from deeprest_amd.engine.config import DataConfig, EngineConfig, TrainConfig
+    from deeprest_amd.engine.experiment import run_scenario_suite
+    from deeprest_amd.models.net import DeepRestNetConfig
+
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=6, n_components=6, windows_per_day=200, n_days=1, seed=9))
+    cfg = EngineConfig()
+    cfg.data = DataConfig(step_size=20, split=0.5)
+    cfg.train = TrainConfig(epochs=2, batch_size=16, baseline_epochs=3,
+                            eval_cycles=3, run_baselines=True, log_every=0)
+    cfg.model = DeepRestNetConfig(d_model=32, n_heads=4, n_layers=1, d_ff=64,
+                                  hidden=16, comp_dim=8, dropout=0.0)
+    store = run_scenario_suite(app, base_name="t", config=cfg,
+                               device=__import__("torch").device("cpu"))
+    exps = store.experiments()
+    assert sorted(exps) == sorted([
+        "t-waves_seen-1x", "t-waves_unseen-3x", "t-flat_unseen-1x",
+        "t-waves_unseen_compositions-1x",
+    ])
+    for exp in exps:
+        comps = store.results[exp]
+        assert comps, exp
+        some = next(iter(comps.values()))
+        entry = next(iter(some.values()))
+        for est in ("bl-resrc", "bl-api", "bl-trace", "ours"):
+            pred = np.asarray(entry[f"prediction_{est}"])
+            assert np.isfinite(pred).all()
+            assert (pred >= 0).all()
+        assert np.isfinite(entry["measurement"]).all()
+    # the 3x-scale scenario's ground truth really is bigger traffic
+    m_seen = np.mean(next(iter(next(iter(
+        store.results["t-waves_seen-1x"].values())).values()))["measurement"])
+    m_3x = np.mean(next(iter(next(iter(
+        store.results["t-waves_unseen-3x"].values())).values()))["measurement"])
+    assert m_3x > m_seen
