@@ -92,6 +92,14 @@ def cmd_experiment(args):
             seed=d.synth_seed,
         ))
         store = run_scenario_suite(app, base_name=args.name, config=cfg)
+        from .engine.experiment import scenario_error_tables
+
+        for exp, per_est in scenario_error_tables(store).items():
+            print(f"===== {exp} =====")
+            for est, t in per_est.items():
+                print(f"   {est:>9} => Median: {t['median']:.4f} | "
+                      f"95-th: {t['p95']:.4f} | 99-th: {t['p99']:.4f} | "
+                      f"Max: {t['max']:.4f}")
     else:
         data = _load_data(cfg)
         store = run_experiment(data, args.name, config=cfg)

@@ -73,6 +73,32 @@ def run_experiment(
     return store
 
 
+def scenario_error_tables(store: ResultsStore) -> Dict[str, Dict[str, Dict[str, float]]]:
+    """Per scenario, per estimator: Median/95th/99th/Max absolute error of
+    the flattened query-window predictions vs ground truth (the reference's
+    error-table format, estimate.py:112-122, aggregated over metrics)."""
+    from ..utils.errors import error_percentiles
+
+    tables: Dict[str, Dict[str, Dict[str, float]]] = {}
+    for exp in store.experiments():
+        errs: Dict[str, List[np.ndarray]] = {}
+        for comp, metrics in store.results[exp].items():
+            for metric, entry in metrics.items():
+                meas = np.asarray(entry["measurement"])
+                for est in ("bl-resrc", "bl-api", "bl-trace", "ours"):
+                    key = f"prediction_{est}"
+                    if key not in entry:
+                        continue
+                    pred = np.asarray(entry[key])
+                    n = min(len(pred), len(meas))
+                    errs.setdefault(est, []).append(
+                        np.abs(pred[:n] - meas[:n]))
+        tables[exp] = {
+            est: error_percentiles(np.concatenate(v)) for est, v in errs.items()
+        }
+    return tables
+
+
 def _flat_series(y_windows: np.ndarray) -> np.ndarray:
     """Reconstruct the flat series from stride-1 windows (N, W)."""
     if len(y_windows) == 1:
@@ -161,10 +187,12 @@ def run_scenario_suite(
             out = trainer.model(xb.to(trainer.device)).float().cpu().numpy()
 
         n_flat = len(qdata.traffic)
+        split_flat = ds.split + step - 1   # base learning period, flat steps
         for m, name in enumerate(ds.metric_names):
             comp = spec.components[spec.comp_of[m]]
             resource = spec.resources[spec.res_of[m]]
             measurement = np.asarray(qdata.resources[name])
+            base_peak = float(np.max(np.asarray(base.resources[name])[:split_flat]))
             comp_windows = comp_bls[m].estimate_series(
                 qdata.invocations.get(comp, qdata.invocations["general"]),
                 n_flat)[: len(Xq)]
@@ -181,7 +209,8 @@ def run_scenario_suite(
                 measurement=measurement,
                 predictions=preds,
                 calls=None,
-                train_len=min(ds.split, len(measurement) - 1),
+                train_len=0,               # query timeline starts at step 0
+                train_peak=base_peak,      # scale relative to learning period
             )
             store.add(f"{base_name}-{scen_name}", comp, resource, entry)
     return store

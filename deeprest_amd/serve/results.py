@@ -41,10 +41,15 @@ def build_results_entry(
     predictions: Dict[str, np.ndarray],          # est -> (K, W) query-window preds
     calls: Optional[Sequence[np.ndarray]] = None,  # per-API call series
     train_len: Optional[int] = None,             # learning-period length
+    train_peak: Optional[float] = None,          # scale anchor override: for
+    # scenario entries whose measurement is a SEPARATE query timeline, pass
+    # the base learning period's peak (the reference's scale semantics) and
+    # train_len=0 so predictions align from the series start
 ) -> Dict[str, object]:
     measurement = np.asarray(measurement, dtype=np.float64)
     t_train = train_len if train_len is not None else len(measurement) // 2
-    train_peak = float(np.max(measurement[:t_train])) if t_train > 0 else 1.0
+    if train_peak is None:
+        train_peak = float(np.max(measurement[:t_train])) if t_train > 0 else 1.0
 
     entry: Dict[str, object] = {
         "calls": [list(np.asarray(c, dtype=np.float64)) for c in (calls or [])],

@@ -83,3 +83,31 @@ def test_scenario_suite_unseen_traffic():
     m_3x = np.mean(next(iter(next(iter(
         store.results["t-waves_unseen-3x"].values())).values()))["measurement"])
     assert m_3x > m_seen
+
+
+def test_scenario_error_tables_shape():
+    import numpy as np
+
+    from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+    from deeprest_amd.engine.config import DataConfig, EngineConfig, TrainConfig
+    from deeprest_amd.engine.experiment import (run_scenario_suite,
+                                                scenario_error_tables)
+    from deeprest_amd.models.net import DeepRestNetConfig
+    import torch
+
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=5, n_components=5, windows_per_day=180, n_days=1, seed=4))
+    cfg = EngineConfig()
+    cfg.data = DataConfig(step_size=20, split=0.5)
+    cfg.train = TrainConfig(epochs=1, batch_size=16, baseline_epochs=2,
+                            eval_cycles=2, run_baselines=True, log_every=0)
+    cfg.model = DeepRestNetConfig(d_model=32, n_heads=4, n_layers=1, d_ff=64,
+                                  hidden=16, comp_dim=8, dropout=0.0)
+    store = run_scenario_suite(app, base_name="s", config=cfg,
+                               device=torch.device("cpu"))
+    tables = scenario_error_tables(store)
+    assert set(tables.keys()) == set(store.experiments())
+    for exp, per_est in tables.items():
+        assert set(per_est.keys()) == {"bl-resrc", "bl-api", "bl-trace", "ours"}
+        for est, t in per_est.items():
+            assert np.isfinite(t["median"]) and t["median"] >= 0
