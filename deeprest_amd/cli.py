@@ -145,7 +145,7 @@ def main(argv=None):
     for name in ("train", "experiment"):
         t = sub.add_parser(name)
         t.add_argument("--config", default=None)
-        t.add_argument("--set", nargs="*", default=[])
+        t.add_argument("--set", nargs="*", action="extend", default=[])
         if name == "experiment":
             t.add_argument("--name", required=True)
             t.add_argument("--out", default="results.pkl")
