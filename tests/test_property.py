@@ -134,6 +134,31 @@ def test_minmax_roundtrip(vals):
     np.testing.assert_allclose(back, M, rtol=1e-9, atol=1e-6)
 
 
+@settings(max_examples=20, deadline=None)
+@given(st.lists(_span(2), min_size=2, max_size=6), st.integers(0, 2**31 - 1),
+       st.integers(1, 30))
+def test_synthesizer_root_counts_and_conservation(traces, seed, count):
+    """Synthesized vectors are sums of observed whole-trace vectors: the
+    root-path count equals the requested calls, every count is achievable."""
+    from deeprest_amd.data.synthesizer import TraceSynthesizer
+
+    raw = [{"metrics": [{"component": "svc-a", "resource": "cpu", "value": 1.0}],
+            "traces": traces}]
+    syn = TraceSynthesizer().fit(raw)
+    fs = syn.feature_space
+    rng = np.random.default_rng(seed)
+    api = syn.apis[0]
+    vec = syn.synthesize({api: count}, rng=rng)
+    comp, op = api.split("_", 1)
+    root_idx = fs.index_of(((comp, op),))
+    # every sampled trace contributes exactly one root-path occurrence
+    assert vec[root_idx] == count
+    # total span count is between count * min-spans and count * max-spans
+    vecs, _ = syn.api2dist[api]
+    per_trace = vecs.sum(axis=1)
+    assert count * per_trace.min() <= vec.sum() <= count * per_trace.max()
+
+
 @settings(max_examples=25, deadline=None)
 @given(st.integers(0, 2**31 - 1))
 def test_pinball_zero_at_exact_prediction_and_nonnegative(seed):
