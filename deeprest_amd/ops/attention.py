@@ -7,11 +7,13 @@ MFMA with LDS-staged tiles").  The forward is ONE fused HIP kernel per
 softmax in registers, PV on MFMA — no S x S score tensor is materialized.
 It saves the per-row logsumexp so the backward can recompute P cheaply.
 
-The backward recomputes P = exp(QK^T * scale - lse) and chains plain GEMMs
-(rocBLAS) — dV = P^T dO, dP = dO V^T, dS = P*(dP - rowsum(dP*P)),
-dQ = dS K * scale, dK = dS^T Q * scale.  Sequence lengths here are short
-(T = 60..512), so recompute is cheap and keeps the fused-forward hot path
-simple.
+The backward is ONE fused kernel per (batch-head, 64-key tile) in a
+TRANSPOSED layout — keys as MFMA rows, queries as columns — so S^T, dP^T and
+dS^T share one C-fragment layout and the softmax-grad math is lane-local
+(per-column lse and D-row); dK/dV accumulate in registers with exclusive
+stores, dQ via fp32 atomics.  It recomputes P from the saved logsumexp
+instead of storing the S x S score tensor (replaced a rocBLAS recompute-GEMM
+chain: 1 ms -> 0.23 ms per step at the bench config).
 
 Replaces and upgrades the reference's per-metric feature-mask "attention"
 (reference: resource-estimation/qrnn.py:21-23,34).
