@@ -47,7 +47,19 @@ class IngestStore:
         self.windows.extend(windows)
         return len(self.windows)
 
-    def featurize(self):
+    def featurize(self, model_space=None):
+        """``model_space``: when a checkpointed model is loaded, its FROZEN
+        call-path feature space — synthesized what-if vectors must line up
+        with the model's input features, not with whatever subset the
+        ingested windows happen to cover (unseen paths are dropped, exactly
+        the reference's frozen-space extract semantics, featurize.py:27-33)."""
+        if model_space is not None:
+            self.featurizer = Featurizer(feature_space=model_space,
+                                         use_native=False)
+            data = self.featurizer.transform(self.windows)
+            self.synthesizer = TraceSynthesizer(
+                feature_space=model_space).fit(self.windows)
+            return data
         self.featurizer = Featurizer(use_native=False).fit(self.windows)
         data = self.featurizer.transform(self.windows)
         self.synthesizer = TraceSynthesizer(
@@ -145,11 +157,14 @@ def create_app(checkpoint_path: Optional[str] = None, predictor: Optional[Predic
     def featurize():
         if not store.windows:
             raise HTTPException(status_code=400, detail="no ingested windows")
-        data = store.featurize()
+        pred = state["predictor"]
+        model_space = pred.feature_space if pred is not None else None
+        data = store.featurize(model_space=model_space)
         return {
             "num_paths": data.num_paths,
             "num_windows": data.num_windows,
             "metrics": data.metric_names,
+            "frozen_to_model_space": model_space is not None,
         }
 
     @app.get("/apis")
@@ -169,6 +184,15 @@ def create_app(checkpoint_path: Optional[str] = None, predictor: Optional[Predic
         plan = payload.get("traffic_plan")
         if not plan:
             raise HTTPException(status_code=422, detail="traffic_plan required")
+        n_model = pred.model.spec.num_paths
+        n_syn = len(store.synthesizer.feature_space)
+        if n_syn != n_model:
+            raise HTTPException(
+                status_code=422,
+                detail=f"feature space mismatch: synthesizer has {n_syn} call "
+                       f"paths, model expects {n_model} — load the predictor "
+                       "with its checkpoint feature space (POST /featurize "
+                       "again after loading)")
         rng = np.random.default_rng(payload.get("seed"))
         try:
             out = pred.predict_what_if(

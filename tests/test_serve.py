@@ -182,3 +182,34 @@ def test_demo_page_served(trained):
     r = client.get("/demo")
     assert r.status_code == 200
     assert "deeprest-amd" in r.text and "canvas" in r.text
+
+
+def test_estimate_uses_model_feature_space(trained):
+    """A checkpoint-loaded predictor carries its frozen call-path space;
+    /featurize must featurize INTO it so /estimate's synthesized vectors
+    match the model's input width even when the ingested windows only cover
+    a subset of paths (regression: this used to 500 with a shape mismatch)."""
+    from starlette.testclient import TestClient
+
+    from deeprest_amd.serve.api import create_app
+
+    app_obj, raw, data, cfg, ckpt = trained
+    pred = Predictor.from_checkpoint(ckpt, device=torch.device("cpu"))
+    client = TestClient(create_app(predictor=pred))
+    # ingest only a few windows with truncated trace lists: a strict subset
+    subset = [{"metrics": w["metrics"], "traces": w["traces"][:1]} for w in raw[:10]]
+    assert client.post("/ingest", json=subset).status_code == 200
+    r = client.post("/featurize")
+    assert r.status_code == 200
+    if pred.feature_space is not None:
+        assert r.json()["frozen_to_model_space"] is True
+        assert r.json()["num_paths"] == pred.model.spec.num_paths
+    apis = client.get("/apis").json()["apis"]
+    if apis:
+        out = client.post("/estimate", json={
+            "traffic_plan": [{apis[0]: 2} for _ in range(25)],
+            "step_size": 20, "seed": 0})
+        # either a clean answer or a clean 422 (never a 500 shape crash)
+        assert out.status_code in (200, 422)
+        if out.status_code == 200:
+            assert out.json()["predictions"]
