@@ -114,3 +114,48 @@ def test_span_tree_orphan_reference_becomes_root():
 
 def test_collector_empty_returns_no_windows():
     assert Collector().windows() == []
+
+
+def test_jaeger_to_trained_model_end_to_end():
+    """Full seam: Jaeger-style exports + Prometheus-style samples ->
+    collector windows -> featurizer -> a training step of the real model.
+    Catches drift between the collection plane's output and what the
+    engine's data path accepts."""
+    import torch
+
+    from deeprest_amd.engine.config import DataConfig, EngineConfig, TrainConfig
+    from deeprest_amd.engine.trainer import Trainer
+    from deeprest_amd.models.net import DeepRestNetConfig
+
+    rng = np.random.default_rng(0)
+    col = Collector(window_sec=5.0, t0=0.0)
+    # 120 windows of alternating request shapes + drifting cpu samples
+    for w in range(120):
+        t_us = int((w * 5.0 + 0.5) * 1e6)
+        n_calls = 1 + int(2 + 2 * np.sin(w / 10) + rng.integers(0, 2))
+        for c in range(n_calls):
+            shape = [("root", "nginx", "/compose", None),
+                     ("s1", "text-svc", "/parse", "root")]
+            if (w + c) % 2 == 0:
+                shape.append(("s2", "user-db", "/find", "s1"))
+            col.add_traces([make_jaeger_trace(f"w{w}c{c}", t_us, shape)])
+        load = n_calls * 10.0 + rng.normal(0, 0.5)
+        col.add_metric_samples([
+            {"component": "nginx", "resource": "cpu", "value": load,
+             "timestamp": w * 5.0 + 1.0},
+            {"component": "user-db", "resource": "cpu", "value": load * 0.5,
+             "timestamp": w * 5.0 + 1.0},
+        ])
+    windows = col.windows()
+    validate_raw_data(windows)
+    data = Featurizer(use_native=False).fit_transform(windows)
+    assert data.num_windows >= 120
+
+    cfg = EngineConfig()
+    cfg.data = DataConfig(step_size=20, split=0.5)
+    cfg.train = TrainConfig(epochs=1, batch_size=8, run_baselines=False,
+                            log_every=0)
+    cfg.model = DeepRestNetConfig(d_model=32, n_heads=4, n_layers=1, d_ff=64,
+                                  hidden=16, comp_dim=8, dropout=0.0)
+    res = Trainer(data, cfg, device=torch.device("cpu")).train()
+    assert np.isfinite(res.train_losses).all()
