@@ -157,3 +157,25 @@ def test_gru_chunked_state_carry_exact_both_directions():
         h = out[:, 0].contiguous()
         parts_r[ci] = out
     assert torch.equal(torch.cat(parts_r, dim=1), full_r)
+
+
+def test_graph_propagation_respects_call_graph():
+    """Components connected in the call graph exchange information; isolated
+    components don't: perturbing one component's embedding moves its
+    neighbor's propagated embedding but not a disconnected one's."""
+    import numpy as np
+    import torch
+
+    from deeprest_amd.models.net import DeepRestNetConfig, _GraphPropagation
+
+    # 0-1 connected, 2 isolated (row-normalized with self-loops)
+    A = np.array([[.5, .5, 0.], [.5, .5, 0.], [0., 0., 1.]])
+    cfg = DeepRestNetConfig(comp_dim=8, prop_rounds=2)
+    torch.manual_seed(0)
+    gp = _GraphPropagation(cfg, 3, A)
+    base = gp().detach()
+    with torch.no_grad():
+        gp.emb[0] += 1.0
+    moved = gp().detach()
+    assert not torch.allclose(base[1], moved[1])   # neighbor sees the change
+    assert torch.allclose(base[2], moved[2])       # isolated does not
