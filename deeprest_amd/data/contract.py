@@ -45,11 +45,16 @@ class Span:
 
     @staticmethod
     def from_dict(d: Dict[str, Any]) -> "Span":
-        return Span(
-            component=d["component"],
-            operation=d["operation"],
-            children=[Span.from_dict(c) for c in d.get("children", [])],
-        )
+        """Iterative build (deep chains must not exhaust the Python stack)."""
+        root = Span(component=d["component"], operation=d["operation"])
+        stack = [(d, root)]
+        while stack:
+            src, dst = stack.pop()
+            for c in src.get("children", []):
+                child = Span(component=c["component"], operation=c["operation"])
+                dst.children.append(child)
+                stack.append((c, child))
+        return root
 
     def walk(self):
         """Depth-first pre-order traversal (iterative; deep trees safe)."""
@@ -87,16 +92,21 @@ class ContractError(ValueError):
 
 
 def _validate_span(node: Any, where: str) -> None:
-    if not isinstance(node, dict):
-        raise ContractError(f"{where}: span must be a dict, got {type(node).__name__}")
-    for key in ("component", "operation"):
-        if key not in node or not isinstance(node[key], str):
-            raise ContractError(f"{where}: span missing string field '{key}'")
-    children = node.get("children", [])
-    if not isinstance(children, list):
-        raise ContractError(f"{where}: span 'children' must be a list")
-    for i, child in enumerate(children):
-        _validate_span(child, f"{where}.children[{i}]")
+    # iterative: REST /ingest feeds arbitrary payloads here, and a deep
+    # span chain must produce a ContractError, not a RecursionError
+    stack = [(node, where)]
+    while stack:
+        cur, w = stack.pop()
+        if not isinstance(cur, dict):
+            raise ContractError(f"{w}: span must be a dict, got {type(cur).__name__}")
+        for key in ("component", "operation"):
+            if key not in cur or not isinstance(cur[key], str):
+                raise ContractError(f"{w}: span missing string field '{key}'")
+        children = cur.get("children", [])
+        if not isinstance(children, list):
+            raise ContractError(f"{w}: span 'children' must be a list")
+        for i, child in enumerate(children):
+            stack.append((child, f"{w}.children[{i}]"))
 
 
 def validate_raw_data(raw_data: Any) -> None:

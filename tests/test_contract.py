@@ -59,3 +59,25 @@ def test_span_walk_preorder():
     )
     order = [n.component for n in s.walk()]
     assert order == ["a", "b", "c", "d"]
+
+
+def test_validate_deep_span_chain_no_recursion_error():
+    """REST /ingest accepts arbitrary payloads: a 10k-deep span chain must
+    validate (or ContractError), never RecursionError."""
+    deep = {"component": "c", "operation": "o", "children": []}
+    node = deep
+    for _ in range(10_000):
+        child = {"component": "c", "operation": "o", "children": []}
+        node["children"].append(child)
+        node = child
+    validate_raw_data([{"metrics": [], "traces": [deep]}])
+    # typed view + featurizer also survive the same chain
+    from deeprest_amd.data.contract import Span
+    from deeprest_amd.data.featurize import Featurizer
+
+    s = Span.from_dict(deep)
+    assert sum(1 for _ in s.walk()) == 10_001
+    data = Featurizer(use_native=False).fit_transform(
+        [{"metrics": [{"component": "c", "resource": "cpu", "value": 1.0}],
+          "traces": [deep]}])
+    assert data.num_paths == 10_001
