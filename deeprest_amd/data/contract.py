@@ -37,11 +37,18 @@ class Span:
         return f"{self.component}_{self.operation}"
 
     def to_dict(self) -> Dict[str, Any]:
-        return {
-            "component": self.component,
-            "operation": self.operation,
-            "children": [c.to_dict() for c in self.children],
-        }
+        """Iterative (deep chains must not exhaust the Python stack)."""
+        out = {"component": self.component, "operation": self.operation,
+               "children": []}
+        stack = [(self, out)]
+        while stack:
+            src, dst = stack.pop()
+            for c in src.children:
+                cd = {"component": c.component, "operation": c.operation,
+                      "children": []}
+                dst["children"].append(cd)
+                stack.append((c, cd))
+        return out
 
     @staticmethod
     def from_dict(d: Dict[str, Any]) -> "Span":
