@@ -213,3 +213,27 @@ def test_estimate_uses_model_feature_space(trained):
         assert out.status_code in (200, 422)
         if out.status_code == 200:
             assert out.json()["predictions"]
+
+
+def test_cryptojacking_detection_end_to_end(trained):
+    """The reference's sanity-check story: inject a traffic-unjustified CPU
+    burst into the measured series and detect it against the TRAINED model's
+    own quantile band (reference locust/pow.py + README.md:3)."""
+    from deeprest_amd.data.windows import sliding_window
+
+    app_obj, raw, data, cfg, ckpt = trained
+    pred = Predictor.from_checkpoint(ckpt, device=torch.device("cpu"))
+    W = 20
+    windows = sliding_window(np.asarray(data.traffic, dtype=np.float64), W)[:1]
+    out = pred.predict(windows)                      # {metric: (1, W, 3)}
+    metric = data.metric_names[0]
+    q = out[metric][0]                               # (W, 3)
+    measured = q[:, 1].copy()                        # sits on the median
+    measured[8:14] += 50.0 * max(np.median(measured), 1.0)   # CPU thief
+
+    scorer = AnomalyScorer(threshold=1.0, min_run=3)
+    rep = scorer.score(measured, q[:, 0], q[:, 1], q[:, 2], metric=metric)
+    assert rep.is_anomalous
+    assert any(s >= 7 and e <= 15 for s, e in rep.windows)
+    # the untouched region stays quiet (measured == predicted median)
+    assert not rep.flags[:7].any() and not rep.flags[15:].any()
