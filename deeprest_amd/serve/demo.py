@@ -108,8 +108,7 @@ async function estimate() {
     drawBand(c, w.map(q => q[0]), w.map(q => q[1]), w.map(q => q[2]), name);
   }
   document.getElementById('status').textContent =
-    'estimated ' + names.length + ' metrics for mix ' + JSON.stringify(
-      Object.fromEntries(Object.entries(body.predictions).slice(0,0))) +
+    'estimated ' + names.length + ' metrics for mix ' + JSON.stringify(plan) +
     ' — showing first window quantile bands';
 }
 init();
