@@ -116,6 +116,10 @@ class Predictor:
         xn = self.x_scaler.transform(x)
         out = self.predict_normalized(torch.from_numpy(xn).float())
         out = out.float().cpu().numpy()            # (N, T, M, Q)
+        # quantile regression can emit crossed quantiles (q95 < q50) early in
+        # training; serving consumers (anomaly bands, demo plots) assume a
+        # monotone triple, so sort the Q axis — a no-op once calibrated
+        out = np.sort(out, axis=-1)
         preds = {}
         for m, name in enumerate(self.metric_names):
             preds[name] = np.maximum(
