@@ -182,6 +182,9 @@ def run_scenario_suite(
         if ds.x_scaler.scale != 0.0:
             xb -= ds.x_scaler.min_val
             xb /= ds.x_scaler.scale
+        # bl-trace was FIT on normalized windows (ds.X) — estimate on the
+        # same normalization, not raw counts
+        xq_norm = xb.numpy()
         trainer.model.eval()
         with torch.no_grad():
             out = trainer.model(xb.to(trainer.device)).float().cpu().numpy()
@@ -200,7 +203,7 @@ def run_scenario_suite(
                 # RESRC is a single repeated history window by construction
                 "bl-resrc": np.tile(bl["resrc"][0, :, m], (len(eval_idx), 1)),
                 "bl-api": comp_windows[eval_idx],
-                "bl-trace": trace_bls[m].estimate(Xq_eval),
+                "bl-trace": trace_bls[m].estimate(xq_norm),
                 "ours": ds.denormalize_metric(out[:, :, m, median_q], m),
             }
             preds = {k: np.maximum(np.asarray(v, dtype=np.float64), 1e-6)
