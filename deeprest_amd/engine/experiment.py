@@ -186,7 +186,9 @@ def run_scenario_suite(
         # same normalization, not raw counts
         xq_norm = xb.numpy()
         trainer.model.eval()
-        with torch.no_grad():
+        with torch.no_grad(), torch.autocast(
+                device_type=trainer.device.type, dtype=trainer.autocast_dtype,
+                enabled=(trainer.device.type == "cuda")):
             out = trainer.model(xb.to(trainer.device)).float().cpu().numpy()
 
         n_flat = len(qdata.traffic)
