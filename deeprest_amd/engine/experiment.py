@@ -39,7 +39,10 @@ def run_experiment(
 
     # model predictions for the non-overlapping query windows
     eval_idx = ds.eval_window_indices(max_cycles=10**9)
-    with torch.no_grad():
+    trainer.model.eval()
+    with torch.no_grad(), torch.autocast(
+            device_type=trainer.device.type, dtype=trainer.autocast_dtype,
+            enabled=(trainer.device.type == "cuda")):
         xb = ds.X_test[eval_idx].to(trainer.device)
         out = trainer.model(xb).float().cpu().numpy()        # (K, T, M, Q)
     median_q = len(trainer.model.cfg.quantiles) // 2
