@@ -95,7 +95,7 @@ def main():
         "small_request_windows": int(len(small)),
         "small_p50_ms": round(float(np.percentile(lat_s, 50)), 3),
         "small_p95_ms": round(float(np.percentile(lat_s, 95)), 3),
-        "hipgraph": not args.no_graph,
+        "hipgraph": pred.use_graph,
         "graph_batch": args.graph_batch,
         "endpoints": args.endpoints,
         "seq_len": args.seq_len,
