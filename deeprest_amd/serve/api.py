@@ -123,6 +123,16 @@ def create_app(checkpoint_path: Optional[str] = None, predictor: Optional[Predic
 
         return HTMLResponse(DEMO_HTML)
 
+    @app.get("/demo/results")
+    def demo_results():
+        """The reference web-demo's results browser (web-demo/app.py) over
+        the /results REST surface."""
+        from fastapi.responses import HTMLResponse
+
+        from .demo import RESULTS_HTML
+
+        return HTMLResponse(RESULTS_HTML)
+
     @app.get("/health")
     def health():
         from ..ops import native_available

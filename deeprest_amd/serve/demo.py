@@ -115,3 +115,121 @@ init();
 </script>
 </body>
 </html>"""
+
+
+# The reference web-demo's core function — browsing precomputed results.pkl
+# (web-demo/app.py:51-193: experiment/component/metric selection, ground
+# truth vs all four estimators, per-component scale factors) — as a second
+# self-contained page on the /results REST surface (route GET /demo/results).
+RESULTS_HTML = """<!DOCTYPE html>
+<html>
+<head>
+<meta charset="utf-8"/>
+<title>deeprest-amd results</title>
+<style>
+ body { font-family: system-ui, sans-serif; margin: 2rem; background:#fafafa; }
+ h1 { font-size: 1.3rem; }
+ .card { background: white; border: 1px solid #ddd; border-radius: 8px;
+         padding: 1rem; margin: .5rem 0; }
+ canvas { border: 1px solid #eee; background: white; }
+ select { padding: .3rem; margin-right: .8rem; }
+ #status { color: #666; font-size: .9rem; }
+ .legend span { margin-right: 1rem; font-size: .85rem; }
+</style>
+</head>
+<body>
+<h1>deeprest-amd &mdash; results browser</h1>
+<div id="status">loading&hellip;</div>
+<div class="card">
+  experiment <select id="exp"></select>
+  component <select id="comp"></select>
+  metric <select id="metric"></select>
+</div>
+<div class="card">
+  <div class="legend">
+    <span style="color:red">ground truth</span>
+    <span style="color:green">bl-resrc</span>
+    <span style="color:orange">bl-api</span>
+    <span style="color:purple">bl-trace</span>
+    <span style="color:mediumblue">ours</span>
+  </div>
+  <canvas id="chart" width="900" height="320"></canvas>
+  <div id="scales"></div>
+</div>
+<script>
+const COLORS = {'bl-resrc':'green','bl-api':'orange','bl-trace':'purple','ours':'mediumblue'};
+async function init() {
+  const r = await fetch('../results');
+  if (!r.ok) {
+    document.getElementById('status').textContent =
+      'no results loaded (serve with --results results.pkl)';
+    return;
+  }
+  const exps = (await r.json()).experiments;
+  fill('exp', exps);
+  await onExp();
+  document.getElementById('exp').onchange = onExp;
+  document.getElementById('comp').onchange = onComp;
+  document.getElementById('metric').onchange = draw;
+  document.getElementById('status').textContent = exps.length + ' experiments';
+}
+function fill(id, values) {
+  const s = document.getElementById(id);
+  s.innerHTML = '';
+  for (const v of values) {
+    const o = document.createElement('option'); o.value = v; o.textContent = v;
+    s.appendChild(o);
+  }
+}
+async function onExp() {
+  const exp = document.getElementById('exp').value;
+  const comps = await (await fetch('../results/' + encodeURIComponent(exp))).json();
+  window._comps = comps;
+  fill('comp', Object.keys(comps));
+  onComp();
+}
+function onComp() {
+  const comp = document.getElementById('comp').value;
+  fill('metric', window._comps[comp] || []);
+  draw();
+}
+async function draw() {
+  const exp = document.getElementById('exp').value;
+  const comp = document.getElementById('comp').value;
+  const met = document.getElementById('metric').value;
+  if (!exp || !comp || !met) return;
+  const e = await (await fetch('../results/' + encodeURIComponent(exp) + '/' +
+      encodeURIComponent(comp) + '/' + encodeURIComponent(met))).json();
+  const ctx = document.getElementById('chart').getContext('2d');
+  const W = 900, H = 320, pad = 30;
+  ctx.clearRect(0, 0, W, H);
+  const meas = e.measurement;
+  let all = meas.slice();
+  for (const est in COLORS)
+    if (e['prediction_' + est]) all = all.concat(e['prediction_' + est]);
+  const lo = Math.min(...all), hi = Math.max(...all) || 1;
+  const n = Math.max(meas.length,
+      ...Object.keys(COLORS).map(k => (e['prediction_' + k] || []).length));
+  const x = i => pad + (W - 2 * pad) * i / (n - 1 || 1);
+  const y = v => H - pad - (H - 2 * pad) * (v - lo) / (hi - lo || 1);
+  const line = (arr, color, dash) => {
+    if (!arr || !arr.length) return;
+    ctx.strokeStyle = color; ctx.setLineDash(dash || []);
+    ctx.beginPath(); ctx.moveTo(x(0), y(arr[0]));
+    for (let i = 1; i < arr.length; i++) ctx.lineTo(x(i), y(arr[i]));
+    ctx.stroke(); ctx.setLineDash([]);
+  };
+  line(meas, 'red', [5, 3]);
+  for (const est in COLORS) line(e['prediction_' + est], COLORS[est]);
+  const sc = [];
+  if (e.scale_groundtruth) sc.push('scale gt: ' +
+      e.scale_groundtruth.map(v => v.toFixed(2)).join(', '));
+  for (const est in COLORS)
+    if (e['scale_' + est]) sc.push('scale ' + est + ': ' +
+        e['scale_' + est].map(v => v.toFixed(2)).join(', '));
+  document.getElementById('scales').textContent = sc.join('  |  ');
+}
+init();
+</script>
+</body>
+</html>"""

@@ -101,6 +101,9 @@ def test_rest_results_browsing(tmp_path):
     store.save(p)
 
     client = TestClient(create_app(results_path=p))
+    # the results-browser page serves alongside the JSON surface
+    page = client.get("/demo/results")
+    assert page.status_code == 200 and "results browser" in page.text
     assert client.get("/results").json() == {"experiments": ["exp1-waves_seen-1x"]}
     assert client.get("/results/exp1-waves_seen-1x").json() == {"frontend": ["cpu"]}
     got = client.get("/results/exp1-waves_seen-1x/frontend/cpu").json()
