@@ -81,3 +81,27 @@ def test_validate_deep_span_chain_no_recursion_error():
         [{"metrics": [{"component": "c", "resource": "cpu", "value": 1.0}],
           "traces": [deep]}])
     assert data.num_paths == 10_001
+
+
+def test_shipped_fixtures_match_generator():
+    """examples/fixtures must stay in sync with the synthetic generator
+    (regenerate with examples/make_fixtures.py after generator changes)."""
+    import os
+
+    import numpy as np
+
+    from deeprest_amd.data.contract import load_raw_data
+    from deeprest_amd.data.featurize import FeaturizedData, Featurizer
+    from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+
+    root = os.path.join(os.path.dirname(__file__), "..", "examples", "fixtures")
+    raw = load_raw_data(os.path.join(root, "raw_data.pkl"))
+    data = FeaturizedData.load(os.path.join(root, "input.pkl"))
+
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=2, n_components=5, windows_per_day=3, n_days=1,
+        shapes_per_api=2, seed=1))
+    fresh_raw = app.generate_raw()
+    assert raw == fresh_raw, "fixtures stale: rerun examples/make_fixtures.py"
+    fresh = Featurizer().fit_transform(fresh_raw)
+    assert np.array_equal(fresh.traffic, data.traffic)
