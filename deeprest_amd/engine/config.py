@@ -8,7 +8,6 @@ CLI-overridable) preserving the same tunables.
 
 from __future__ import annotations
 
-import argparse
 import dataclasses
 from dataclasses import dataclass, field
 from typing import Any, Dict, Optional
@@ -83,12 +82,6 @@ class EngineConfig:
     def save(self, path: str) -> None:
         with open(path, "w") as f:
             yaml.safe_dump(self.to_dict(), f)
-
-
-def add_cli_overrides(parser: argparse.ArgumentParser) -> None:
-    parser.add_argument("--config", type=str, default=None, help="YAML config path")
-    parser.add_argument("--set", nargs="*", default=[], metavar="SECTION.KEY=VALUE",
-                        help="dot-path overrides, e.g. train.epochs=10")
 
 
 def apply_cli_overrides(cfg: EngineConfig, overrides) -> EngineConfig:
