@@ -111,3 +111,25 @@ def test_scenario_error_tables_shape():
         assert set(per_est.keys()) == {"bl-resrc", "bl-api", "bl-trace", "ours"}
         for est, t in per_est.items():
             assert np.isfinite(t["median"]) and t["median"] >= 0
+
+
+def test_run_experiment_from_raw_wrapper():
+    import torch
+
+    from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+    from deeprest_amd.engine.config import DataConfig, EngineConfig, TrainConfig
+    from deeprest_amd.engine.experiment import run_experiment_from_raw
+    from deeprest_amd.models.net import DeepRestNetConfig
+
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=3, n_components=4, windows_per_day=120, n_days=1, seed=6))
+    raw = app.generate_raw(plan=app.traffic_plan(scale=0.05))
+    cfg = EngineConfig(
+        data=DataConfig(step_size=20, split=0.5),
+        train=TrainConfig(epochs=1, batch_size=8, run_baselines=False,
+                          baseline_epochs=2, log_every=0),
+        model=DeepRestNetConfig(d_model=32, n_heads=4, n_layers=1, d_ff=64,
+                                hidden=16, comp_dim=8, dropout=0.0))
+    store = run_experiment_from_raw(raw, "from-raw", config=cfg,
+                                    device=torch.device("cpu"))
+    assert store.experiments() == ["from-raw"]
