@@ -8,8 +8,8 @@ designed for AMD Instinct MI355X (gfx950, CDNA4):
                 application generator.
 - ``models``:   the estimation engine (attention traffic encoder, call-graph
                 propagation, per-resource GRU decoders with quantile heads)
-                plus the two reference baselines (history-only ANN,
-                component-aware linear scaling).
+                plus the three comparison baselines (history-only ANN,
+                component-aware linear scaling, trace-level ridge).
 - ``ops``:      hand-written CDNA4 HIP kernels (fused GRU cell, MHA over the
                 endpoint x time window, LayerNorm, pinball loss, fused Adam)
                 with PyTorch autograd bindings; CPU fallbacks for testing.
