@@ -140,6 +140,9 @@ def main():
         torch.cuda.synchronize()
     if dist_ctx is not None:
         dist_ctx.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()   # NCCL barrier is stream-ordered: make sure
+                                   # it has RUN before any rank reads the clock
 
     # ---- timed region: exactly K steps ----
     t0 = time.perf_counter()
@@ -149,6 +152,8 @@ def main():
         torch.cuda.synchronize()
     if dist_ctx is not None:
         dist_ctx.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
 
     # max over ranks (slowest rank defines the job)
