@@ -245,15 +245,18 @@ class Featurizer:
                 for trace in window["traces"]:
                     fs.count_trace(trace, row)
 
-            # invocation counts for the component-aware baseline
+            # invocation counts for the component-aware baseline — include
+            # components seen in the DATA even when absent from a frozen
+            # feature space (matches the native walk's semantics)
             components = fs.components()
             invocations = {c: np.zeros(T, dtype=np.int64) for c in components}
             invocations["general"] = np.zeros(T, dtype=np.int64)
             for t, window in enumerate(raw_data):
                 counts = _count_invocations_window(window["traces"])
                 for comp, n in counts.items():
-                    if comp in invocations:
-                        invocations[comp][t] = n
+                    if comp not in invocations:
+                        invocations[comp] = np.zeros(T, dtype=np.int64)
+                    invocations[comp][t] = n
         return self._assemble(raw_data, traffic, invocations)
 
     def _assemble(self, raw_data, traffic, invocations) -> FeaturizedData:

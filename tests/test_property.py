@@ -103,6 +103,9 @@ def test_native_featurizer_property_parity(train_traces, new_traces):
     dp = py.transform(train + fresh)   # frozen space: unseen paths dropped
     dn = nat.transform(train + fresh)
     np.testing.assert_array_equal(dp.traffic, dn.traffic)
+    assert set(dp.invocations.keys()) == set(dn.invocations.keys())
+    for k in dp.invocations:
+        np.testing.assert_array_equal(dp.invocations[k], dn.invocations[k])
 
 
 @settings(max_examples=30, deadline=None)
