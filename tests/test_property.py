@@ -171,3 +171,17 @@ def test_pinball_zero_at_exact_prediction_and_nonnegative(seed):
     assert float(reference_pinball_loss(exact, y, (0.05, 0.5, 0.95))) == 0.0
     off = exact + torch.from_numpy(rng.normal(size=exact.shape).astype(np.float32))
     assert float(reference_pinball_loss(off, y, (0.05, 0.5, 0.95))) >= 0.0
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.lists(st.floats(-1e6, 1e6, allow_nan=False), min_size=4, max_size=40))
+def test_minmax_state_roundtrip_and_zero_range(vals):
+    M = np.asarray(vals, dtype=np.float64).reshape(-1, 1)
+    sc = MinMaxScaler().fit(M, split=max(2, len(vals) // 2))
+    sc2 = MinMaxScaler.from_state_dict(sc.state_dict())
+    np.testing.assert_array_equal(sc.transform(M), sc2.transform(M))
+    # degenerate constant series: transform/inverse must be total no-ops
+    C = np.full((6, 1), 3.25)
+    scc = MinMaxScaler().fit(C, split=3)
+    np.testing.assert_array_equal(scc.transform(C), C)
+    np.testing.assert_array_equal(scc.inverse_transform(C), C)
