@@ -216,7 +216,8 @@ def run_scenario_suite(
             entry = build_results_entry(
                 measurement=measurement,
                 predictions=preds,
-                calls=None,
+                # the query period's total call series (the demo's traffic view)
+                calls=[qdata.invocations["general"]],
                 train_len=0,               # query timeline starts at step 0
                 train_peak=base_peak,      # scale relative to learning period
             )
