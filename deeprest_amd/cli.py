@@ -162,7 +162,9 @@ def main(argv=None):
     v = sub.add_parser("serve")
     v.add_argument("--checkpoint", default=None)
     v.add_argument("--results", default=None, help="results.pkl to browse at /results")
-    v.add_argument("--host", default="0.0.0.0")
+    # 127.0.0.1 by default: the API is unauthenticated, so external
+    # binding must be an explicit opt-in (--host 0.0.0.0)
+    v.add_argument("--host", default="127.0.0.1")
     v.add_argument("--port", type=int, default=2021)
 
     args = p.parse_args(argv)

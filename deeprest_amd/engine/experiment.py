@@ -103,7 +103,13 @@ def scenario_error_tables(store: ResultsStore) -> Dict[str, Dict[str, Dict[str, 
 
 
 def _flat_series(y_windows: np.ndarray) -> np.ndarray:
-    """Reconstruct the flat series from stride-1 windows (N, W)."""
+    """Reconstruct the flat series from stride-1 windows (N, W).
+
+    Yields (N-1)+W = L-1 samples: the window construction itself drops the
+    final raw timestep (a length-L series makes only L-W stride-1 windows,
+    inherited from the reference's ``sliding_window``, estimate.py:26-27).
+    Consumers align by min-length truncation (``scenario_error_tables``).
+    """
     if len(y_windows) == 1:
         return y_windows[0]
     return np.concatenate([y_windows[:-1, 0], y_windows[-1]])

@@ -69,6 +69,10 @@ class Trainer:
             spec = build_model_spec(data)
             model = DeepRestNet(spec, self.cfg.model)
         self.model = model.to(self.device)
+        if dist_ctx is not None and dist_ctx.world_size > 1:
+            # replicas must start identical: model construction is unseeded,
+            # so sync rank-0 weights before the first averaged-gradient step
+            dist_ctx.broadcast_parameters(self.model)
         # graph_step wants the capturable optimizer (device-side step counter);
         # capturable mode is also correct for plain eager stepping
         self._use_graph = bool(
@@ -143,9 +147,13 @@ class Trainer:
         for epoch in range(self.start_epoch, cfg.epochs):
             self.model.train()
             perm = torch.randperm(n, generator=gen)
-            # data-parallel shard: contiguous slice of the permutation per rank
+            # data-parallel shard: equal-length strided slices — truncate to a
+            # multiple of world_size first, else ranks can issue different
+            # numbers of all_reduce calls per epoch (collectives pairing across
+            # batches = silently corrupted averaging, then a hang at the end)
             if self.world_size > 1:
-                perm = perm[self.rank :: self.world_size]
+                n_even = n - (n % self.world_size)
+                perm = perm[:n_even][self.rank :: self.world_size]
             losses = []
             for s in range(0, len(perm), cfg.batch_size):
                 idx = perm[s : s + cfg.batch_size].to(self.device)
