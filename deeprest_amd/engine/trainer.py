@@ -29,6 +29,7 @@ from ..utils.errors import error_percentiles, format_error_table
 from .checkpoint import load_checkpoint, save_checkpoint
 from .config import EngineConfig
 from .dataset import EstimationDataset
+from .step import TrainStep
 
 
 @dataclass
@@ -82,7 +83,6 @@ class Trainer:
         )
         self.optimizer = FusedAdam(self.model.parameters(), lr=self.cfg.train.lr,
                                    capturable=self._use_graph)
-        self._graphed = None
         self.feature_space_state = (
             data.feature_space.state_dict() if data.feature_space is not None else None
         )
@@ -90,6 +90,12 @@ class Trainer:
         self._baseline_preds: Optional[Dict[str, np.ndarray]] = None
         self.autocast_dtype = (
             torch.bfloat16 if self.cfg.train.dtype == "bf16" else torch.float32
+        )
+        # THE training step — shared verbatim with bench.py (engine/step.py)
+        self.step = TrainStep(
+            self.model, self.optimizer, dist_ctx=self.dist,
+            autocast_dtype=self.autocast_dtype
+            if self.device.type == "cuda" else None,
         )
 
     # ------------------------------------------------------------- baselines
@@ -158,35 +164,14 @@ class Trainer:
             for s in range(0, len(perm), cfg.batch_size):
                 idx = perm[s : s + cfg.batch_size].to(self.device)
                 xb, yb = X_train[idx], y_train[idx]
-                full = xb.shape[0] == cfg.batch_size
-                if self._use_graph and self._graphed is None and full:
-                    from .graphstep import GraphedTrainStep
-
-                    # capture once on the first full batch (warmup replays are
-                    # real optimizer steps on that batch); tail batches and
-                    # capture failure fall back to eager below
-                    self._graphed = GraphedTrainStep.build(
-                        self.model, self.optimizer,
-                        lambda o, t: self.model.loss(o.float(), t), xb, yb,
-                        autocast_dtype=self.autocast_dtype)
-                    if self._graphed is None:
+                if (self._use_graph and not self.step.graphed
+                        and xb.shape[0] == cfg.batch_size):
+                    # capture once on the first full batch (warmup replays
+                    # are real optimizer steps on that batch); tail batches
+                    # and capture failure fall back to eager inside TrainStep
+                    if not self.step.try_capture(xb, yb):
                         self._use_graph = False
-                if self._graphed is not None and full:
-                    loss = self._graphed.run(xb, yb)
-                else:
-                    with torch.autocast(
-                        device_type=self.device.type, dtype=self.autocast_dtype,
-                        enabled=(self.device.type == "cuda"),
-                    ):
-                        out = self.model(xb)
-                        loss = self.model.loss(out.float(), yb)
-                    # with an active graph, grads must keep their captured
-                    # addresses: zero in place instead of dropping to None
-                    self.optimizer.zero_grad(set_to_none=self._graphed is None)
-                    loss.backward()
-                    if self.dist is not None:
-                        self.dist.all_reduce_gradients(self.model)
-                    self.optimizer.step()
+                loss = self.step(xb, yb)
                 # keep the loss on-device: a per-batch .item() is a host sync
                 # that drains the GPU pipeline every step
                 losses.append(loss.detach().clone())

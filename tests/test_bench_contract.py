@@ -60,3 +60,33 @@ def test_bench_torchrun_dp2_contract():
     assert d["n_gpus"] == 2
     assert d["config"]["parallelism"] == "dp2"
     assert d["config"]["global_batch"] == 16  # whole-job aggregate
+
+
+def test_bench_accuracy_key_default_off_on_cpu():
+    """CPU default run: the accuracy key is present (contract) but null —
+    the probe only auto-runs on single-process GPU benches."""
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "1", "--warmup", "0",
+         "--batch", "8", "--endpoints", "6", "--components", "5",
+         "--seq-len", "12"],
+        cwd=REPO, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    d = json.loads([l for l in out.stdout.splitlines() if l.startswith("{")][0])
+    assert "accuracy" in d and d["accuracy"] is None
+
+
+def test_accuracy_probe_shape():
+    """The MAE probe emits the three-estimator aggregate the driver line
+    carries (tiny config so it runs on CPU in seconds)."""
+    sys.path.insert(0, REPO)
+    import torch
+    from bench import accuracy_probe
+
+    acc = accuracy_probe(1, torch.device("cpu"), n_apis=3, n_components=4,
+                         windows_per_day=40, n_days=2, step_size=10,
+                         baseline_epochs=2)
+    assert set(acc["mean_median_abs_err"]) == {"resrc", "comp", "deepr"}
+    assert acc["metrics"] == 5 * 3          # (4 comps + frontend) x 3 resources
+    assert 0 <= acc["deepr_beats_comp"] <= acc["metrics"]
+    assert acc["probe_seconds"] > 0

@@ -114,3 +114,40 @@ def test_parameter_broadcast():
 
 def test_dp_training_replicas_stay_identical():
     _spawn(_run_train_step, 29514)
+
+
+def _run_trainer_dp(rank, world_size, port, results):
+    """Full Trainer under DP: unseeded divergent construction must be healed
+    by the init broadcast, and an odd train-window count (65 with batch 32:
+    shards would be 33/32 without truncation -> 2 vs 1 all_reduce calls per
+    epoch -> corrupted averaging + hang) must stay in lockstep."""
+    from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+    from deeprest_amd.engine.config import DataConfig, EngineConfig, TrainConfig
+    from deeprest_amd.engine.trainer import Trainer
+    from deeprest_amd.models.net import DeepRestNetConfig
+
+    ctx = _init(rank, world_size, port)
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=4, n_components=5, windows_per_day=61, n_days=3, seed=21))
+    data = app.generate_featurized()
+    cfg = EngineConfig()
+    cfg.data = DataConfig(step_size=20, split=0.4)
+    cfg.train = TrainConfig(epochs=2, batch_size=32, baseline_epochs=2,
+                            eval_cycles=2, log_every=0, seed=0,
+                            run_baselines=False)
+    cfg.model = DeepRestNetConfig(d_model=32, n_heads=4, n_layers=1, d_ff=64,
+                                  hidden=16, comp_dim=8, dropout=0.0)
+    torch.manual_seed(rank * 31 + 5)   # deliberately divergent construction
+    trainer = Trainer(data, cfg, device=torch.device("cpu"), dist_ctx=ctx)
+    assert trainer.dataset.split == 65  # the uneven-shard regression shape
+    trainer.train()
+    flat = torch.cat([p.detach().reshape(-1)
+                      for p in trainer.model.parameters()])
+    gathered = [torch.zeros_like(flat) for _ in range(world_size)]
+    dist.all_gather(gathered, flat)
+    results[rank] = bool(torch.equal(gathered[0], gathered[1]))
+    dist.destroy_process_group()
+
+
+def test_trainer_dp_replicas_identical_uneven_windows():
+    _spawn(_run_trainer_dp, 29515)
