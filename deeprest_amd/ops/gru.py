@@ -177,6 +177,16 @@ class _FusedGRUSequence(torch.autograd.Function):
         )
 
 
+_WARNED_SHAPES = set()
+
+
+def gru_kernel_supports(H: int, C: int) -> bool:
+    """Shapes the fused CDNA4 kernel is built for: its LDS staging, MFMA
+    tiling and pi-permutation are specialized to hidden size 128, and the
+    64-row tile layout needs >= 3 components per batch row group."""
+    return H == 128 and C >= 3
+
+
 def fused_gru_sequence(
     x_gates: torch.Tensor,
     w_hh: torch.Tensor,
@@ -187,6 +197,24 @@ def fused_gru_sequence(
     reverse: bool = False,
     fp8: bool = False,
 ) -> torch.Tensor:
+    if x_gates.is_cuda and not gru_kernel_supports(h0.shape[2], h0.shape[1]):
+        # Defined degradation path (NOT a silent fallback for the flagship
+        # config — that path still requires the native kernel and fails
+        # loudly if the extension is missing): off-spec model configs
+        # (hidden != 128 or < 3 components) run the differentiable PyTorch
+        # composition on rocBLAS.  Warn once per shape so a user who meant
+        # to be on the hot path notices.
+        key = (int(h0.shape[2]), int(h0.shape[1]))
+        if key not in _WARNED_SHAPES:
+            _WARNED_SHAPES.add(key)
+            import warnings
+
+            warnings.warn(
+                f"fused GRU kernel supports hidden=128 and >=3 components; "
+                f"got hidden={key[0]}, components={key[1]} — using the "
+                f"(slower) composed rocBLAS path for this shape")
+        return reference_gru_sequence(
+            x_gates, w_hh, b_hh, h0, gamma, beta, reverse)
     if x_gates.is_cuda:
         dt = x_gates.dtype
         if gamma is None:

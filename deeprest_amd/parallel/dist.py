@@ -43,8 +43,11 @@ def init_distributed(backend: Optional[str] = None,
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
     if backend == "nccl":
-        torch.cuda.set_device(local_rank)
-        device = torch.device("cuda", local_rank)
+        # modulo: lets an N-rank launch smoke-test RCCL init + collectives
+        # on a box with fewer GPUs (e.g. 2 ranks on the 1-GPU lease)
+        dev_index = local_rank % max(torch.cuda.device_count(), 1)
+        torch.cuda.set_device(dev_index)
+        device = torch.device("cuda", dev_index)
     elif device is None:
         device = torch.device("cpu")
     if not dist.is_initialized():

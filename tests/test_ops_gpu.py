@@ -409,3 +409,90 @@ def test_gru_fp8_inference_path(dev):
     flatr = ref.flatten()
     corr = torch.corrcoef(torch.stack([flat8, flatr]))[0, 1].item()
     assert corr > 0.99, corr
+
+
+# ------------------------------------------------------- gru edge coverage
+@pytest.mark.parametrize("shape", [
+    (1, 1, 3, 128),     # absolute minimum: B=1, T=1, C=3
+    (1, 9, 3, 128),     # B=1
+    (6, 1, 7, 128),     # T=1
+    (64, 2, 3, 128),    # C=3 (launcher's minimum C) with a multi-tile grid
+])
+@pytest.mark.parametrize("reverse", [False, True])
+def test_gru_edge_shapes_fwd_bwd(dev, shape, reverse):
+    """Every edge the launcher branches on (csrc/gru.hip launchers): T=1,
+    B=1, C=3 — forward AND backward vs the fp32 oracle."""
+    from deeprest_amd.ops import fused_gru_sequence, reference_gru_sequence
+
+    B, T, C, H = shape
+    torch.manual_seed(11)
+    mk = lambda *s: torch.randn(*s, device=dev) * 0.4
+    xg, w, b, h0 = mk(B, T, 3 * H), mk(3 * H, H) / np.sqrt(H), mk(3 * H) * 0.2, mk(B, C, H)
+    g, be = 1.0 + 0.1 * mk(C, 3 * H), 0.1 * mk(C, 3 * H)
+    args_t = [t.detach().clone().requires_grad_(True) for t in (xg, w, b, h0, g, be)]
+    args_r = [t.detach().clone().requires_grad_(True) for t in (xg, w, b, h0, g, be)]
+    out_t = fused_gru_sequence(*args_t, reverse=reverse)
+    out_r = reference_gru_sequence(*args_r, reverse=reverse)
+    torch.testing.assert_close(out_t.float(), out_r, rtol=5e-2, atol=3e-2)
+    grad = torch.randn_like(out_r)
+    out_t.backward(grad)
+    out_r.backward(grad)
+    for name, at_, ar_ in zip(["xg", "w", "b", "h0", "gam", "bet"], args_t, args_r):
+        torch.testing.assert_close(
+            at_.grad.float(), ar_.grad.float(), rtol=8e-2, atol=5e-2,
+            msg=lambda m, n=name: f"{shape} grad mismatch for {n}: {m}")
+
+
+@pytest.mark.parametrize("B,C", [
+    (3056, 8),   # 191 x 128-row tiles: NSUB1 side of the boundary
+    (3072, 8),   # exactly 192 tiles: first NSUB2 launch
+    (3080, 8),   # 192+ tiles with a ragged tail row block
+    (6200, 4),   # big grid but C=4: fwd stays NSUB1 (C>=5 gate)
+])
+def test_gru_nsub_boundary(dev, B, C):
+    """The NSUB1/NSUB2 launcher boundary (tiles128 >= 192, fwd also C >= 5)
+    — both sides must agree with the oracle, fwd and bwd."""
+    from deeprest_amd.ops import fused_gru_sequence, reference_gru_sequence
+
+    T, H = 2, 128
+    torch.manual_seed(12)
+    mk = lambda *s: torch.randn(*s, device=dev) * 0.4
+    xg, w, b, h0 = mk(B, T, 3 * H), mk(3 * H, H) / np.sqrt(H), mk(3 * H) * 0.2, mk(B, C, H)
+    g, be = 1.0 + 0.1 * mk(C, 3 * H), 0.1 * mk(C, 3 * H)
+    args_t = [t.detach().clone().requires_grad_(True) for t in (xg, w, b, h0, g, be)]
+    args_r = [t.detach().clone().requires_grad_(True) for t in (xg, w, b, h0, g, be)]
+    out_t = fused_gru_sequence(*args_t)
+    out_r = reference_gru_sequence(*args_r)
+    torch.testing.assert_close(out_t.float(), out_r, rtol=5e-2, atol=3e-2)
+    grad = torch.randn_like(out_r)
+    out_t.backward(grad)
+    out_r.backward(grad)
+    for name, at_, ar_ in zip(["xg", "w", "b", "h0", "gam", "bet"], args_t, args_r):
+        err = (at_.grad.float() - ar_.grad.float()).abs().max()
+        scale = ar_.grad.float().abs().max() + 1e-9
+        assert err / scale < 2e-2, f"{name}: relmax {(err / scale).item():.4f}"
+
+
+def test_gru_offspec_shape_uses_composed_path(dev):
+    """hidden != 128 / C < 3 degrade to the differentiable rocBLAS
+    composition with a one-time warning — defined behavior, not a crash
+    and not a silent wrong answer."""
+    import warnings
+
+    from deeprest_amd.ops import fused_gru_sequence, reference_gru_sequence
+    from deeprest_amd.ops.gru import _WARNED_SHAPES
+
+    torch.manual_seed(13)
+    for (B, T, C, H) in [(4, 5, 6, 64), (4, 5, 2, 128)]:
+        _WARNED_SHAPES.clear()
+        mk = lambda *s: torch.randn(*s, device=dev) * 0.4
+        xg, w, b, h0 = mk(B, T, 3 * H), mk(3 * H, H) / np.sqrt(H), mk(3 * H) * 0.2, mk(B, C, H)
+        args = [t.detach().clone().requires_grad_(True) for t in (xg, w, b, h0)]
+        with warnings.catch_warnings(record=True) as rec:
+            warnings.simplefilter("always")
+            out = fused_gru_sequence(*args)
+        assert any("composed rocBLAS path" in str(r.message) for r in rec)
+        ref = reference_gru_sequence(xg, w, b, h0)
+        torch.testing.assert_close(out.float(), ref.float(), rtol=1e-4, atol=1e-4)
+        out.sum().backward()   # autograd must work on the degraded path
+        assert args[0].grad is not None and torch.isfinite(args[0].grad).all()
