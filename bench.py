@@ -142,10 +142,13 @@ def accuracy_probe(epochs: int, device: torch.device, n_apis: int = 13,
         data=DataConfig(step_size=step_size, split=0.40),
         train=TrainConfig(epochs=epochs, batch_size=32, lr=1e-3,
                           eval_cycles=9, baseline_epochs=baseline_epochs,
-                          log_every=0),
+                          log_every=0, eval_every=5, graph_step=True),
         model=DeepRestNetConfig(dropout=0.1),
     )
     trainer = Trainer(data, cfg, device=device)
+    t_setup = time.perf_counter() - t0
+    trainer.run_baselines()
+    t_baselines = time.perf_counter() - t0 - t_setup
     result = trainer.train()
 
     wins_resrc = wins_comp = total = 0
@@ -166,6 +169,8 @@ def accuracy_probe(epochs: int, device: torch.device, n_apis: int = 13,
         "config": "reference 50ep/b32/split.40/window60 config, "
                   "13-endpoint 12-component 8-day synthetic app",
         "probe_seconds": round(time.perf_counter() - t0, 1),
+        "setup_seconds": round(t_setup, 1),
+        "baseline_seconds": round(t_baselines, 1),
     }
 
 
@@ -179,7 +184,9 @@ def main():
     if args.device:
         device = torch.device(args.device)
     elif on_gpu:
-        device = torch.device("cuda", dist_ctx.local_rank if dist_ctx else 0)
+        # dist_ctx.device already maps local_rank modulo the visible GPU
+        # count (a 2-rank RCCL smoke on a 1-GPU box shares device 0)
+        device = dist_ctx.device if dist_ctx else torch.device("cuda", 0)
         torch.cuda.set_device(device)
     else:
         device = torch.device("cpu")

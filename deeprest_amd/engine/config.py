@@ -45,6 +45,8 @@ class TrainConfig:
     checkpoint_path: Optional[str] = None
     resume: bool = False
     log_every: int = 1
+    eval_every: int = 1                   # per-epoch eval cadence (rank 0);
+                                          # the final epoch always evaluates
 
 
 @dataclass

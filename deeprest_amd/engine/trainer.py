@@ -22,7 +22,7 @@ import numpy as np
 import torch
 
 from ..data.featurize import FeaturizedData
-from ..models.baselines import ComponentAwareBaseline, ResourceAwareBaseline
+from ..models.baselines import ComponentAwareBaseline
 from ..models.net import DeepRestNet, build_model_spec
 from ..ops.adam import FusedAdam
 from ..utils.errors import error_percentiles, format_error_table
@@ -105,22 +105,27 @@ class Trainer:
             return self._baseline_preds
         ds = self.dataset
         t_cfg = self.cfg.train
-        resrc_list, comp_list = [], []
+        # RESRC: every metric's history-MLP fit in ONE batched run (on the
+        # GPU when present; 39 sequential CPU fits took 74 s)
+        from ..models.baselines import ResourceAwareBatchBaseline
+
+        y_stack = np.ascontiguousarray(np.moveaxis(ds.y_raw, -1, 0))  # (M, N, T)
+        resrc_all = ResourceAwareBatchBaseline(
+            split=ds.split, window=ds.step_size,
+            epochs=t_cfg.baseline_epochs, seed=t_cfg.seed,
+            device=self.device if self.device.type == "cuda" else None,
+        ).fit_and_estimate(y_stack)                         # (M, N_test, T)
+        comp_list = []
         for idx, name in enumerate(ds.metric_names):
             yw = ds.y_raw[:, :, idx]
-            resrc = ResourceAwareBaseline(
-                split=ds.split, window=ds.step_size,
-                epochs=t_cfg.baseline_epochs, seed=t_cfg.seed,
-            ).fit_and_estimate(yw)
             comp_name = self.model.spec.components[self.model.spec.comp_of[idx]]
             comp = ComponentAwareBaseline(
                 component=comp_name, invocations=ds.data.invocations,
                 window=ds.step_size, split=ds.split,
             ).fit_and_estimate(yw)
-            resrc_list.append(resrc[:, :, None])
             comp_list.append(comp[:, :, None])
         self._baseline_preds = {
-            "resrc": np.concatenate(resrc_list, axis=-1),   # (N_test, T, M)
+            "resrc": np.moveaxis(resrc_all, 0, -1),         # (N_test, T, M)
             "comp": np.concatenate(comp_list, axis=-1),
         }
         return self._baseline_preds
@@ -180,7 +185,9 @@ class Trainer:
                 float(torch.stack(losses).mean().item()) if losses else float("nan")
             )
 
-            if self.rank == 0:
+            do_eval = ((epoch + 1) % max(cfg.eval_every, 1) == 0
+                       or epoch + 1 == cfg.epochs)
+            if self.rank == 0 and do_eval:
                 test_loss, tables = self.evaluate(baselines)
                 result.test_losses.append(test_loss)
                 result.error_tables = tables

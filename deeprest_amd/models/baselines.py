@@ -105,6 +105,112 @@ class ResourceAwareBaseline(nn.Module):
         return np.tile(pred, (n_test, 1))
 
 
+class ResourceAwareBatchBaseline:
+    """All metrics' history MLPs trained together as one batched model.
+
+    Semantics match running ResourceAwareBaseline per metric (reference:
+    baselines.py:34-77): independent per-metric weights (stacked along a
+    leading M axis, driven by bmm), the same shared init (the sequential
+    harness seeds every metric's MLP identically), per-metric min-max
+    normalization, and a per-metric-summed MSE so each metric's gradients
+    are exactly what its independent fit would see (Adam is elementwise).
+    ~M x fewer op launches: 39 sequential fits took 74 s on CPU; this runs
+    them in one loop (and on the GPU when one is present).
+    """
+
+    def __init__(
+        self,
+        split: int,
+        window: int,
+        offset: Optional[int] = None,
+        hidden: int = 128,
+        epochs: int = 100,
+        lr: float = 1e-3,
+        batch_size: int = 32,
+        device: Optional[torch.device] = None,
+        seed: int = 0,
+    ) -> None:
+        self.split = split
+        self.window = window
+        self.offset = offset if offset is not None else window - 1
+        self.hidden = hidden
+        self.epochs = epochs
+        self.lr = lr
+        self.batch_size = batch_size
+        self.seed = seed
+        self.device = device or torch.device("cpu")
+
+    def fit_and_estimate(self, y_stack: np.ndarray) -> np.ndarray:
+        """y_stack: (M, N, window) -> (M, N - split, window) predictions."""
+        y = np.asarray(y_stack, dtype=np.float64)
+        M, N, W = y.shape
+        train_n = self.split - self.offset
+        if train_n <= 0:
+            raise ValueError("split must exceed offset for the history baseline")
+
+        # per-metric min-max on the train split (minmax_fit/apply semantics,
+        # incl. the rng==0 "leave unchanged" edge case)
+        lo = y[:, : self.split].reshape(M, -1).min(axis=1)
+        hi = y[:, : self.split].reshape(M, -1).max(axis=1)
+        rng = hi - lo
+        div = np.where(rng != 0.0, rng, 1.0)
+        sub = np.where(rng != 0.0, lo, 0.0)
+        yn = (y - sub[:, None, None]) / div[:, None, None]
+
+        X = np.ascontiguousarray(yn[:, :train_n], dtype=np.float32)
+        # target = window `offset` steps ahead of the input window
+        Y = np.ascontiguousarray(
+            yn[:, self.offset : self.offset + train_n], dtype=np.float32)
+
+        dev = self.device
+        Xt = torch.from_numpy(X).to(dev)           # (M, n, W)
+        Yt = torch.from_numpy(Y).to(dev)
+        gen = torch.Generator().manual_seed(self.seed)
+        bound1 = 1.0 / np.sqrt(W)
+        bound2 = 1.0 / np.sqrt(self.hidden)
+        # one init shared by all metrics — the sequential harness seeds every
+        # metric's MLP with the same generator, so their inits are identical.
+        # Draw in nn.Linear's (out, in) layout so the generator stream maps
+        # to the same elements, then transpose for bmm.
+        w1 = torch.empty(self.hidden, W).uniform_(
+            -bound1, bound1, generator=gen).t().contiguous()
+        b1 = torch.empty(self.hidden).uniform_(-bound1, bound1, generator=gen)
+        w2 = torch.empty(W, self.hidden).uniform_(
+            -bound2, bound2, generator=gen).t().contiguous()
+        b2 = torch.empty(W).uniform_(-bound2, bound2, generator=gen)
+        w1 = w1[None].repeat(M, 1, 1).to(dev).requires_grad_(True)
+        b1 = b1[None, None].repeat(M, 1, 1).to(dev).requires_grad_(True)
+        w2 = w2[None].repeat(M, 1, 1).to(dev).requires_grad_(True)
+        b2 = b2[None, None].repeat(M, 1, 1).to(dev).requires_grad_(True)
+
+        opt = torch.optim.Adam([w1, b1, w2, b2], lr=self.lr)
+        n = Xt.shape[1]
+        perm_gen = torch.Generator().manual_seed(self.seed + 1)
+        for _ in range(self.epochs):
+            perm = torch.randperm(n, generator=perm_gen)
+            for s in range(0, n, self.batch_size):
+                idx = perm[s : s + self.batch_size].to(dev)
+                xb = Xt[:, idx]                     # (M, B, W)
+                yb = Yt[:, idx]
+                h = torch.relu(torch.bmm(xb, w1) + b1)
+                pred = torch.bmm(h, w2) + b2
+                # sum of per-metric means: each metric's grads match its
+                # independent fit exactly
+                loss = ((pred - yb) ** 2).mean(dim=(1, 2)).sum()
+                opt.zero_grad()
+                loss.backward()
+                opt.step()
+
+        with torch.no_grad():
+            pi = train_n - self.offset
+            probe = Xt[:, [pi if pi >= 0 else -1]]  # (M, 1, W)
+            h = torch.relu(torch.bmm(probe, w1) + b1)
+            pred = (torch.bmm(h, w2) + b2).squeeze(1).cpu().numpy()  # (M, W)
+        pred = np.maximum(pred * rng[:, None] + lo[:, None], 1e-6)
+        n_test = N - self.split
+        return np.tile(pred[:, None, :], (1, n_test, 1))
+
+
 class TraceAwareBaseline:
     """Ridge regression from call-path traffic features to the metric.
 

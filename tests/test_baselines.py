@@ -52,3 +52,27 @@ def test_resource_aware_shapes_and_repetition():
     # reference behavior: a single predicted window repeated for every test window
     assert np.allclose(pred[0], pred[-1])
     assert (pred >= 1e-6).all()
+
+
+def test_resource_aware_batch_matches_sequential():
+    """Batched RESRC == per-metric sequential fits when batch covers the
+    whole train set (full-batch gradient makes shuffling order irrelevant,
+    the only intended semantic difference)."""
+    import torch
+
+    from deeprest_amd.models.baselines import (ResourceAwareBaseline,
+                                               ResourceAwareBatchBaseline)
+
+    rng = np.random.default_rng(5)
+    M, N, W = 3, 60, 12
+    y = np.cumsum(rng.normal(1.0, 0.3, size=(M, N, W)), axis=1) + 10.0
+    split = 30
+    batch = ResourceAwareBatchBaseline(
+        split=split, window=W, epochs=4, batch_size=10**6, seed=3
+    ).fit_and_estimate(y)
+    for m in range(M):
+        torch.manual_seed(999)  # sequential perm draw is order-only: full batch
+        seq = ResourceAwareBaseline(
+            split=split, window=W, epochs=4, batch_size=10**6, seed=3
+        ).fit_and_estimate(y[m])
+        np.testing.assert_allclose(batch[m], seq, rtol=2e-4, atol=2e-4)

@@ -318,7 +318,7 @@ def test_trainer_graph_step_gpu(dev):
                                   hidden=128, comp_dim=16, dropout=0.0)
     tr = Trainer(data, cfg, device=dev)
     res = tr.train()
-    assert tr._graphed is not None, "graph was not captured"
+    assert tr.step.graphed, "graph was not captured"
     assert np.isfinite(res.train_losses).all()
     assert res.train_losses[-1] < res.train_losses[0] * 1.5
 
