@@ -41,10 +41,16 @@ def init_distributed(backend: Optional[str] = None,
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
-    if backend == "nccl":
-        # modulo: lets an N-rank launch smoke-test RCCL init + collectives
-        # on a box with fewer GPUs (e.g. 2 ranks on the 1-GPU lease)
+        # DEEPREST_DIST_BACKEND=gloo on a GPU box runs the whole distributed
+        # path (sharded data, CUDA-tensor all-reduce, max-elapsed reduction)
+        # through gloo — the only N>1 smoke possible on a 1-GPU lease, since
+        # RCCL (like NCCL) refuses two ranks on one device ("Duplicate GPU
+        # detected", measured: profiles/r02_dp_smoke.md)
+        backend = os.environ.get("DEEPREST_DIST_BACKEND") or (
+            "nccl" if torch.cuda.is_available() else "gloo")
+    if torch.cuda.is_available() and device is None:
+        # modulo lets an N-rank gloo smoke share the single GPU; with nccl
+        # each rank must own a distinct device (driver scale runs)
         dev_index = local_rank % max(torch.cuda.device_count(), 1)
         torch.cuda.set_device(dev_index)
         device = torch.device("cuda", dev_index)
