@@ -64,9 +64,15 @@ class GraphedTrainStep:
 
     def _one_step(self) -> torch.Tensor:
         enabled = self.autocast_dtype is not None
+        # cache_enabled=False is REQUIRED under graph capture: with the
+        # weight-cast cache on, captured replays hold references into cache
+        # entries that any later EAGER autocast step (e.g. a ragged tail
+        # batch) invalidates — measured as deterministic NaN corruption of
+        # every parameter a few epochs in (tools/probe_debug3.py: drop-last
+        # training was clean, any eager tail step poisoned the replays)
         with torch.autocast(device_type="cuda",
                             dtype=self.autocast_dtype or torch.bfloat16,
-                            enabled=enabled):
+                            enabled=enabled, cache_enabled=False):
             out = self.model(self.static_x)
             loss = self.loss_fn(out, self.static_y)
         # set_to_none=False: grads must stay at fixed addresses across replays
