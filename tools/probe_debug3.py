@@ -73,9 +73,11 @@ def run_variant(variant, data, device, epochs, seed):
     gen = torch.Generator().manual_seed(seed)
     g = None
     step_i = 0
+    sync = variant.startswith("sync") or variant == "graphsync"
     for epoch in range(epochs):
         perm = torch.randperm(n, generator=gen)
-        stop = n - (n % bs) if variant == "graphdrop" else n
+        drop = variant in ("graphdrop", "syncdrop")
+        stop = n - (n % bs) if drop else n
         losses, steps = [], []
         for s in range(0, stop, bs):
             idx = perm[s : s + bs].to(device)
@@ -91,13 +93,15 @@ def run_variant(variant, data, device, epochs, seed):
             if g is not None and xb.shape[0] == bs:
                 loss = g.run(xb, yb)
             else:  # eager tail
-                with torch.autocast("cuda", torch.bfloat16):
+                cache = variant != "synccachefree"
+                with torch.autocast("cuda", torch.bfloat16,
+                                    cache_enabled=cache):
                     out = model(xb)
                     loss = model.loss(out.float(), yb)
                 opt.zero_grad(set_to_none=False)
                 loss.backward()
                 opt.step()
-            if variant == "graphsync":
+            if sync:
                 torch.cuda.synchronize()
             losses.append(loss.detach().clone())
             steps.append((step_i, xb.shape[0]))
