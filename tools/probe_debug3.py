@@ -64,7 +64,8 @@ def run_variant(variant, data, device, epochs, seed):
     ds = EstimationDataset(data, step_size=60, split_fraction=0.40)
     spec = build_model_spec(data)
     model = DeepRestNet(spec, DeepRestNetConfig(dropout=0.0)).to(device)
-    opt = FusedAdam(model.parameters(), lr=1e-3, capturable=True)
+    opt = FusedAdam(model.parameters(), lr=1e-3,
+                    capturable=variant != "syncplainadam")
     loss_fn = lambda o, t: model.loss(o, t)
     X = ds.X_train.to(device)
     y = ds.y_train.to(device)
@@ -76,14 +77,14 @@ def run_variant(variant, data, device, epochs, seed):
     sync = variant.startswith("sync") or variant == "graphsync"
     for epoch in range(epochs):
         perm = torch.randperm(n, generator=gen)
-        drop = variant in ("graphdrop", "syncdrop")
+        drop = variant in ("graphdrop", "syncdrop", "syncfb", "syncplainadam")
         stop = n - (n % bs) if drop else n
         losses, steps = [], []
         for s in range(0, stop, bs):
             idx = perm[s : s + bs].to(device)
             xb, yb = X[idx], y[idx]
             if g is None and xb.shape[0] == bs:
-                if variant == "graphfb":
+                if variant in ("graphfb", "syncfb", "syncplainadam"):
                     g = FwdBwdGraph(model, opt, loss_fn, xb, yb)
                 else:
                     g = GraphedTrainStep(
