@@ -50,7 +50,12 @@ class GraphedTrainStep:
 
         # warmup on a side stream: allocates grads, Adam state, the fused
         # all-reduce bucket, autograd workspace — everything whose pointers
-        # the captured graph will bake in
+        # the captured graph will bake in.  Capture happens on the SAME
+        # stream: autograd binds each parameter's AccumulateGrad node to the
+        # stream it first ran on, so warming up on one stream and capturing
+        # on another records cross-stream edges that replay unsafely
+        # (measured: replays went NaN when the host synchronized between
+        # replays — tools/probe_debug3.py)
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
@@ -59,7 +64,7 @@ class GraphedTrainStep:
         torch.cuda.current_stream().wait_stream(side)
 
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
+        with torch.cuda.graph(self.graph, stream=side):
             self.static_loss = self._one_step()
 
     def _one_step(self) -> torch.Tensor:

@@ -40,7 +40,7 @@ class FwdBwdGraph:
                 opt.step()
         torch.cuda.current_stream().wait_stream(side)
         self.g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g):
+        with torch.cuda.graph(self.g, stream=side):
             self.static_loss = self._one()
 
     def _one(self):
