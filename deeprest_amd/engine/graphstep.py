@@ -18,6 +18,12 @@ Capture-safety in this codebase:
 Use ``GraphedTrainStep.build(...)``; it returns None (caller stays eager)
 when capture is unsupported or fails — capture is an optimization, never a
 correctness requirement.
+
+Contract for callers: do NOT host-synchronize between consecutive replays
+(a `torch.cuda.synchronize()` after every replay corrupts training on this
+stack a few dozen replays in — measured and isolated in
+profiles/r02_graph_capture_notes.md; epoch-granularity syncs are fine and
+are what Trainer/bench do).
 """
 
 from __future__ import annotations
