@@ -3,15 +3,23 @@
 Loads a checkpoint (weights + spec + fitted scalers + call-path feature
 space M — everything inference needs) and serves
 batched quantile predictions.  On GPU the forward is captured once into a
-hipGraph (torch.cuda.CUDAGraph IS hipGraph on ROCm) at a fixed batch shape;
-smaller requests are padded and replayed — one graph launch instead of
-hundreds of kernel launches per request (north star: "the online prediction
-step is hipGraph-captured").
+hipGraph (torch.cuda.CUDAGraph IS hipGraph on ROCm) per configured batch
+size; every request is served by replays of the smallest graph that fits —
+one graph launch instead of hundreds of kernel launches per request (north
+star: "the online prediction step is hipGraph-captured", BASELINE config 3:
+batched 1k-window inference).
+
+Staging-buffer ingestion: requests are copied straight into the captured
+graph's fixed input buffer (H2D for numpy requests, D2D for resident
+tensors) — no intermediate device tensor, so bulk 1k-window inference is a
+single copy + one replay, which beats the eager path (round 1's single
+64-window graph lost at bulk sizes because 1k windows took 16 chunked
+replays; measured in profiles/r02_serve_p50.md).
 """
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List, Optional, Sequence, Tuple
 
 import numpy as np
 import torch
@@ -20,6 +28,8 @@ from ..data.featurize import FeatureSpace
 from ..data.windows import MinMaxScaler
 from ..engine.checkpoint import load_checkpoint
 from ..models.net import DeepRestNet
+
+DEFAULT_GRAPH_BATCHES = (16, 256, 1024)
 
 
 class Predictor:
@@ -31,7 +41,8 @@ class Predictor:
         metric_names: List[str],
         feature_space: Optional[FeatureSpace] = None,
         device: Optional[torch.device] = None,
-        graph_batch: int = 64,
+        graph_batch: Optional[int] = None,
+        graph_batches: Optional[Sequence[int]] = None,
         use_graph: bool = True,
     ) -> None:
         self.device = device or torch.device(
@@ -42,12 +53,12 @@ class Predictor:
         self.y_scalers = y_scalers
         self.metric_names = metric_names
         self.feature_space = feature_space
-        self.graph_batch = graph_batch
+        if graph_batches is None:
+            graph_batches = (graph_batch,) if graph_batch else DEFAULT_GRAPH_BATCHES
+        self.graph_batches: Tuple[int, ...] = tuple(sorted(set(graph_batches)))
         self.use_graph = use_graph and self.device.type == "cuda"
-        self._graph = None
-        self._graph_in: Optional[torch.Tensor] = None
-        self._graph_out: Optional[torch.Tensor] = None
-        self._graph_T: Optional[int] = None
+        # (batch, T) -> (graph, input buffer, output buffer)
+        self._graphs: Dict[Tuple[int, int], tuple] = {}
 
     @staticmethod
     def from_checkpoint(path: str, device: Optional[torch.device] = None,
@@ -62,52 +73,81 @@ class Predictor:
         return Predictor(model, x_scaler, y_scalers, sc["metric_names"],
                          feature_space=fs, device=device, **kw)
 
+    @property
+    def captured_batches(self) -> List[int]:
+        return sorted({b for (b, _t) in self._graphs})
+
     # ---------------------------------------------------------------- capture
-    def _ensure_graph(self, T: int, P: int) -> None:
-        if self._graph is not None and self._graph_T == T:
-            return
-        B = self.graph_batch
-        self._graph_in = torch.zeros(B, T, P, device=self.device)
+    def _graph_for(self, n: int, T: int, P: int) -> tuple:
+        """Smallest configured graph batch >= n (n must be <= max batch)."""
+        b = next(bb for bb in self.graph_batches if bb >= n)
+        key = (b, T)
+        got = self._graphs.get(key)
+        if got is not None:
+            return got
+        gin = torch.zeros(b, T, P, device=self.device)
         # warmup on a side stream (required before capture)
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(2):
                 with torch.no_grad():
-                    self.model(self._graph_in)
+                    self.model(gin)
         torch.cuda.current_stream().wait_stream(s)
         g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
+        with torch.cuda.graph(g, stream=s):
             with torch.no_grad():
-                self._graph_out = self.model(self._graph_in)
-        self._graph = g
-        self._graph_T = T
+                gout = self.model(gin)
+        got = (g, gin, gout)
+        self._graphs[key] = got
+        return got
+
+    # ----------------------------------------------------- staged ingestion
+    def staging_buffer(self, n: int, T: int, P: int) -> torch.Tensor:
+        """A writable (n, T, P) view of the captured graph's input buffer.
+
+        Streamed ingestion writes normalized windows directly here (no
+        intermediate tensor), then calls ``predict_staged(n)`` — the
+        whole-request cost is one replay."""
+        _, gin, _ = self._graph_for(n, T, P)
+        return gin[:n]
+
+    @torch.no_grad()
+    def predict_staged(self, n: int, T: int) -> torch.Tensor:
+        """Replay the graph whose staging buffer was filled with n windows."""
+        b = next(bb for bb in self.graph_batches if bb >= n)
+        g, gin, gout = self._graphs[(b, T)]
+        if n < b:
+            gin[n:].zero_()
+        g.replay()
+        return gout[:n]
 
     # ---------------------------------------------------------------- predict
     @torch.no_grad()
     def predict_normalized(self, x: torch.Tensor) -> torch.Tensor:
         """x: (N, T, P) normalized traffic -> (N, T, M, Q).
 
-        Adaptive: requests up to graph_batch replay the captured hipGraph
-        (measured 17% lower p50 on small latency-sensitive requests); bulk
-        requests beyond it run one big eager batch (measured faster than
-        chunked replays at 1k windows)."""
-        x = x.to(self.device, dtype=torch.float32)
+        Requests up to the largest configured graph batch are one padded
+        replay; larger requests chunk at the largest batch (a 4096-window
+        request = four 1024-replays).  CPU inputs are copied H2D straight
+        into the staging buffer."""
+        x = x.float()
         N, T, P = x.shape
-        if not self.use_graph or N > self.graph_batch:
-            return self.model(x)
-        self._ensure_graph(T, P)
+        if not self.use_graph:
+            return self.model(x.to(self.device))
+        bmax = self.graph_batches[-1]
         outs = []
-        B = self.graph_batch
-        for s in range(0, N, B):
-            chunk = x[s : s + B]
-            n = chunk.shape[0]
-            self._graph_in[:n].copy_(chunk)
-            if n < B:
-                self._graph_in[n:].zero_()
-            self._graph.replay()
-            outs.append(self._graph_out[:n].clone())
-        return torch.cat(outs, dim=0)
+        s = 0
+        while s < N:
+            n = min(N - s, bmax)
+            g, gin, gout = self._graph_for(n, T, P)
+            gin[:n].copy_(x[s : s + n])
+            if n < gin.shape[0]:
+                gin[n:].zero_()
+            g.replay()
+            outs.append(gout[:n].clone())
+            s += n
+        return outs[0] if len(outs) == 1 else torch.cat(outs, dim=0)
 
     def predict(self, traffic_windows: np.ndarray) -> Dict[str, np.ndarray]:
         """Raw call-path count windows (N, T, P) -> per-metric denormalized

@@ -343,17 +343,28 @@ def test_predictor_hipgraph_capture(dev):
     x_scaler = MinMaxScaler().fit(data.traffic.astype(np.float64), 100)
     y_scalers = [MinMaxScaler() for _ in data.metric_names]
     pred_graph = Predictor(model, x_scaler, y_scalers, data.metric_names,
-                           device=dev, graph_batch=16, use_graph=True)
+                           device=dev, graph_batches=(16, 64), use_graph=True)
     pred_eager = Predictor(model, x_scaler, y_scalers, data.metric_names,
                            device=dev, use_graph=False)
-    # a small request (<= graph_batch) takes the replay path; check both the
-    # padded-replay correctness and the adaptive bulk fallthrough
-    w = sliding_window(data.traffic.astype(np.float64), 30)[:12]
-    out_g = pred_graph.predict(w)
-    out_e = pred_eager.predict(w)
-    assert pred_graph._graph is not None, "hipGraph was not captured"
-    for name in data.metric_names:
-        np.testing.assert_allclose(out_g[name], out_e[name], rtol=1e-3, atol=1e-3)
+    w_all = sliding_window(data.traffic.astype(np.float64), 30)
+    # small request -> padded replay of the 16-graph; mid request -> the
+    # 64-graph; oversized request -> chunked replays of the largest graph
+    for n in (12, 40, 100):
+        w = np.concatenate([w_all] * (n // len(w_all) + 1))[:n]
+        out_g = pred_graph.predict(w)
+        out_e = pred_eager.predict(w)
+        for name in data.metric_names:
+            np.testing.assert_allclose(out_g[name], out_e[name],
+                                       rtol=1e-3, atol=1e-3)
+    assert pred_graph.captured_batches == [16, 64], "graphs not captured"
+    # staged ingestion: write into the buffer, replay, same numbers
+    xn = pred_graph.x_scaler.transform(w_all[:10].astype(np.float64))
+    buf = pred_graph.staging_buffer(10, 30, xn.shape[-1])
+    buf.copy_(torch.from_numpy(xn).float())
+    out_s = pred_graph.predict_staged(10, 30)
+    out_d = pred_graph.predict_normalized(
+        torch.from_numpy(xn).float())
+    torch.testing.assert_close(out_s, out_d, rtol=1e-5, atol=1e-5)
 
 
 def test_gru_large_rows_8wave_variant(dev):

@@ -1,8 +1,11 @@
 #!/usr/bin/env python3
-"""Online-prediction latency benchmark (BASELINE config 3): batched 1k-window
-inference through the hipGraph-captured predictor; reports p50/p95 latency.
+"""Online-prediction latency benchmark (BASELINE config 3).
 
-Run on a GPU box:  python tools/bench_serve.py [--windows 1000] [--iters 30]
+Measures p50/p95 request latency for the graph-captured predictor vs the
+eager path at request sizes 16 / 256 / 1024 windows, plus the staged-
+ingestion variant (windows written directly into the graph's input buffer).
+
+Run on a GPU box:  python tools/bench_serve.py [--iters 30]
 """
 import argparse
 import json
@@ -23,15 +26,14 @@ from deeprest_amd.serve.predictor import Predictor
 
 def main():
     p = argparse.ArgumentParser()
-    p.add_argument("--windows", type=int, default=1000)
     p.add_argument("--iters", type=int, default=30)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--endpoints", type=int, default=256)
     p.add_argument("--components", type=int, default=64)
     p.add_argument("--seq-len", type=int, default=60)
-    p.add_argument("--graph-batch", type=int, default=256)
-    p.add_argument("--no-graph", action="store_true")
+    p.add_argument("--sizes", default="16,256,1024")
     args = p.parse_args()
+    sizes = [int(s) for s in args.sizes.split(",")]
 
     dev = torch.device("cuda" if torch.cuda.is_available() else "cpu")
     app = SyntheticApp(SyntheticAppConfig(
@@ -43,63 +45,50 @@ def main():
     model = DeepRestNet(spec, DeepRestNetConfig(dropout=0.0)).to(dev).eval()
     x_scaler = MinMaxScaler().fit(data.traffic.astype(np.float64), 200)
     y_scalers = [MinMaxScaler() for _ in data.metric_names]
-    pred = Predictor(model, x_scaler, y_scalers, data.metric_names, device=dev,
-                     graph_batch=args.graph_batch, use_graph=not args.no_graph)
+    graphed = Predictor(model, x_scaler, y_scalers, data.metric_names,
+                        device=dev, graph_batches=tuple(sizes), use_graph=True)
+    eager = Predictor(model, x_scaler, y_scalers, data.metric_names,
+                      device=dev, use_graph=False)
 
     w = sliding_window(data.traffic.astype(np.float64), args.seq_len)
-    reps = int(np.ceil(args.windows / len(w)))
-    w = np.concatenate([w] * reps)[: args.windows]
-    # resident on device: the benchmark measures the prediction path, not PCIe
-    xn = torch.from_numpy(x_scaler.transform(w)).float().to(dev)
+    reps = int(np.ceil(max(sizes) / len(w)))
+    w = np.concatenate([w] * reps)[: max(sizes)]
+    xn_cpu = torch.from_numpy(x_scaler.transform(w)).float()
+    xn_dev = xn_cpu.to(dev)
+    T, P = xn_dev.shape[1], xn_dev.shape[2]
 
-    def time_requests(request_windows):
+    def timed(fn):
         lat = []
         for i in range(args.warmup + args.iters):
             if dev.type == "cuda":
                 torch.cuda.synchronize()
             t0 = time.perf_counter()
-            pred.predict_normalized(request_windows)
+            fn()
             if dev.type == "cuda":
                 torch.cuda.synchronize()
-            dt = time.perf_counter() - t0
             if i >= args.warmup:
-                lat.append(dt * 1000.0)
-        return np.asarray(lat)
+                lat.append((time.perf_counter() - t0) * 1000.0)
+        a = np.asarray(lat)
+        return {"p50_ms": round(float(np.percentile(a, 50)), 3),
+                "p95_ms": round(float(np.percentile(a, 95)), 3)}
 
-    # bulk: one 1k-window batch
-    lat = time_requests(xn)
-    # small latency-sensitive request (graph batch matches it)
-    small = xn[: min(16, len(xn))]
-    if not args.no_graph:
-        pred_small = Predictor(model, x_scaler, y_scalers, data.metric_names,
-                               device=dev, graph_batch=len(small), use_graph=True)
-    else:
-        pred_small = pred
-    lat_s = []
-    for i in range(args.warmup + args.iters):
-        if dev.type == "cuda":
-            torch.cuda.synchronize()
-        t0 = time.perf_counter()
-        pred_small.predict_normalized(small)
-        if dev.type == "cuda":
-            torch.cuda.synchronize()
-        if i >= args.warmup:
-            lat_s.append((time.perf_counter() - t0) * 1000.0)
-    lat_s = np.asarray(lat_s)
-    print(json.dumps({
-        "metric": "online inference latency",
-        "bulk_p50_ms": round(float(np.percentile(lat, 50)), 3),
-        "bulk_p95_ms": round(float(np.percentile(lat, 95)), 3),
-        "bulk_windows": args.windows,
-        "bulk_windows_per_sec": round(args.windows / (np.percentile(lat, 50) / 1000.0), 1),
-        "small_request_windows": int(len(small)),
-        "small_p50_ms": round(float(np.percentile(lat_s, 50)), 3),
-        "small_p95_ms": round(float(np.percentile(lat_s, 95)), 3),
-        "hipgraph": pred.use_graph,
-        "graph_batch": args.graph_batch,
-        "endpoints": args.endpoints,
-        "seq_len": args.seq_len,
-    }))
+    results = {"device": str(dev), "endpoints": args.endpoints,
+               "num_paths": spec.num_paths, "seq_len": args.seq_len,
+               "iters": args.iters, "sizes": {}}
+    for n in sizes:
+        req_dev = xn_dev[:n]
+        entry = {}
+        entry["eager"] = timed(lambda: eager.predict_normalized(req_dev))
+        entry["graph"] = timed(lambda: graphed.predict_normalized(req_dev))
+
+        def staged():
+            buf = graphed.staging_buffer(n, T, P)
+            buf.copy_(req_dev)
+            graphed.predict_staged(n, T)
+        entry["graph_staged"] = timed(staged)
+        results["sizes"][n] = entry
+        print(f"n={n}: {json.dumps(entry)}", flush=True)
+    print(json.dumps(results))
 
 
 if __name__ == "__main__":
