@@ -71,6 +71,9 @@ def run_experiment(
             predictions=preds,
             calls=calls_series,
             train_len=ds.split,
+            # monotone metrics: scale factors compare growth relative to the
+            # learning period's end (reference dataloader.py:143-156)
+            reanchor=resource in ("memory", "usage"),
         )
         store.add(experiment_name, comp, resource, entry)
     return store
@@ -206,7 +209,8 @@ def run_scenario_suite(
             comp = spec.components[spec.comp_of[m]]
             resource = spec.resources[spec.res_of[m]]
             measurement = np.asarray(qdata.resources[name])
-            base_peak = float(np.max(np.asarray(base.resources[name])[:split_flat]))
+            base_series = np.asarray(base.resources[name])[:split_flat]
+            base_peak = float(np.max(base_series))
             comp_windows = comp_bls[m].estimate_series(
                 qdata.invocations.get(comp, qdata.invocations["general"]),
                 n_flat)[: len(Xq)]
@@ -226,6 +230,10 @@ def run_scenario_suite(
                 calls=[qdata.invocations["general"]],
                 train_len=0,               # query timeline starts at step 0
                 train_peak=base_peak,      # scale relative to learning period
+                reanchor=resource in ("memory", "usage"),
+                # the query timeline is separate: anchor = where the BASE
+                # learning period ended
+                anchor_value=float(base_series[-1]),
             )
             store.add(f"{base_name}-{scen_name}", comp, resource, entry)
     return store

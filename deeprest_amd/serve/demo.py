@@ -144,6 +144,7 @@ RESULTS_HTML = """<!DOCTYPE html>
   experiment <select id="exp"></select>
   component <select id="comp"></select>
   metric <select id="metric"></select>
+  query window <select id="qwin"></select>
 </div>
 <div class="card">
   <div class="legend">
@@ -155,6 +156,12 @@ RESULTS_HTML = """<!DOCTYPE html>
   </div>
   <canvas id="chart" width="900" height="320"></canvas>
   <div id="scales"></div>
+</div>
+<div class="card">
+  <b>Scaling factors</b> &mdash; all 4 estimators vs ground truth
+  (magenta dashed), every metric of this component
+  <!-- the reference's per-component bar view, web-demo/app.py:150-180 -->
+  <div><canvas id="bars" width="900" height="220"></canvas></div>
 </div>
 <script>
 const COLORS = {'bl-resrc':'green','bl-api':'orange','bl-trace':'purple','ours':'mediumblue'};
@@ -192,6 +199,67 @@ function onComp() {
   const comp = document.getElementById('comp').value;
   fill('metric', window._comps[comp] || []);
   draw();
+  drawBars();
+}
+async function fetchEntry(exp, comp, met) {
+  return await (await fetch('../results/' + encodeURIComponent(exp) + '/' +
+      encodeURIComponent(comp) + '/' + encodeURIComponent(met))).json();
+}
+// per-component grouped scaling-factor bars: 4 estimators per metric, with
+// the ground-truth scale as a magenta dashed line per metric group
+// (the reference's view, web-demo/app.py:150-180)
+async function drawBars() {
+  const exp = document.getElementById('exp').value;
+  const comp = document.getElementById('comp').value;
+  const mets = window._comps[comp] || [];
+  if (!exp || !comp || !mets.length) return;
+  const entries = {};
+  for (const m of mets) entries[m] = await fetchEntry(exp, comp, m);
+  const nwin = (entries[mets[0]].scale_groundtruth || []).length;
+  const qsel = document.getElementById('qwin');
+  if (qsel.options.length !== nwin) {
+    fill('qwin', Array.from({length: nwin}, (_, i) => i));
+    qsel.onchange = drawBars;
+  }
+  const qi = parseInt(qsel.value) || 0;
+  const ests = ['bl-resrc', 'bl-api', 'bl-trace', 'ours'];
+  const ctx = document.getElementById('bars').getContext('2d');
+  const W = 900, H = 220, pad = 34;
+  ctx.clearRect(0, 0, W, H);
+  let hi = 1.0;
+  for (const m of mets) {
+    for (const est of ests) {
+      const s = entries[m]['scale_' + est];
+      if (s && s[qi] !== undefined) hi = Math.max(hi, s[qi]);
+    }
+    const g = entries[m].scale_groundtruth;
+    if (g && g[qi] !== undefined) hi = Math.max(hi, g[qi]);
+  }
+  const y = v => H - pad - (H - 2 * pad) * v / hi;
+  const groupW = (W - 2 * pad) / mets.length;
+  const barW = groupW / (ests.length + 1.5);
+  mets.forEach((m, gi) => {
+    const x0 = pad + gi * groupW;
+    ests.forEach((est, bi) => {
+      const s = entries[m]['scale_' + est];
+      const v = (s && s[qi] !== undefined) ? s[qi] : 0;
+      ctx.fillStyle = COLORS[est];
+      ctx.fillRect(x0 + bi * barW, y(v), barW - 2, H - pad - y(v));
+    });
+    const g = entries[m].scale_groundtruth;
+    if (g && g[qi] !== undefined) {
+      ctx.strokeStyle = 'magenta'; ctx.setLineDash([4, 3]); ctx.lineWidth = 2;
+      ctx.beginPath();
+      ctx.moveTo(x0 - 2, y(g[qi]));
+      ctx.lineTo(x0 + ests.length * barW + 2, y(g[qi]));
+      ctx.stroke(); ctx.setLineDash([]); ctx.lineWidth = 1;
+    }
+    ctx.fillStyle = '#333'; ctx.font = '11px sans-serif';
+    ctx.fillText(m, x0, H - pad + 14);
+  });
+  ctx.fillStyle = '#333'; ctx.font = '11px sans-serif';
+  ctx.fillText(hi.toFixed(2) + 'x', 2, pad);
+  ctx.fillText('0', 2, H - pad);
 }
 async function draw() {
   const exp = document.getElementById('exp').value;

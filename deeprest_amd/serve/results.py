@@ -30,10 +30,27 @@ import numpy as np
 ESTIMATOR_KEYS = ("bl-resrc", "bl-api", "bl-trace", "ours")
 
 
-def _scales(pred_windows: np.ndarray, train_peak: float) -> List[float]:
-    """Per query window: peak prediction / learning-period peak."""
+def _scales(pred_windows: np.ndarray, train_peak: float,
+            anchor: Optional[float] = None) -> List[float]:
+    """Per query window: peak prediction / learning-period peak.
+
+    ``anchor`` (monotone metrics: memory, disk usage) re-anchors each window
+    to the last learning-period value before taking the peak — the
+    reference's semantics (reference: web-demo/dataloader.py:143-156:
+    ``pred - pred[0] + gt_offset``): for metrics that only ever grow, the
+    absolute level is history, and only growth relative to where the
+    learning period ended is a meaningful scale comparison."""
     train_peak = max(float(train_peak), 1e-9)
-    return [float(np.max(w)) / train_peak for w in pred_windows]
+    out = []
+    for w in pred_windows:
+        if len(w) == 0:
+            out.append(0.0)
+            continue
+        w = np.asarray(w, dtype=np.float64)
+        if anchor is not None:
+            w = w - w[0] + anchor
+        out.append(float(np.max(w)) / train_peak)
+    return out
 
 
 def build_results_entry(
@@ -45,11 +62,21 @@ def build_results_entry(
     # scenario entries whose measurement is a SEPARATE query timeline, pass
     # the base learning period's peak (the reference's scale semantics) and
     # train_len=0 so predictions align from the series start
+    reanchor: bool = False,                      # monotone metric (memory/
+    # usage): re-anchor windows to the learning period's last value before
+    # computing scale factors (dataloader.py:143-156)
+    anchor_value: Optional[float] = None,        # the learning period's last
+    # value when the measurement series here is a separate query timeline
 ) -> Dict[str, object]:
     measurement = np.asarray(measurement, dtype=np.float64)
     t_train = train_len if train_len is not None else len(measurement) // 2
     if train_peak is None:
         train_peak = float(np.max(measurement[:t_train])) if t_train > 0 else 1.0
+    anchor = None
+    if reanchor:
+        anchor = anchor_value if anchor_value is not None else (
+            float(measurement[t_train - 1]) if t_train > 0 else
+            float(measurement[0]))
 
     entry: Dict[str, object] = {
         "calls": [list(np.asarray(c, dtype=np.float64)) for c in (calls or [])],
@@ -63,14 +90,17 @@ def build_results_entry(
         if K is None:
             K, W = pw.shape
         entry[f"prediction_{est}"] = list(pw.reshape(-1))
-        entry[f"scale_{est}"] = _scales(pw, train_peak)
+        entry[f"scale_{est}"] = _scales(pw, train_peak, anchor)
     if K is not None:
         gt_scales = []
         for ki in range(K):
             seg = measurement[t_train + ki * W : t_train + (ki + 1) * W]
-            gt_scales.append(
-                float(np.max(seg)) / max(train_peak, 1e-9) if len(seg) else 0.0
-            )
+            if len(seg) == 0:
+                gt_scales.append(0.0)
+                continue
+            if anchor is not None:
+                seg = seg - seg[0] + anchor
+            gt_scales.append(float(np.max(seg)) / max(train_peak, 1e-9))
         entry["scale_groundtruth"] = gt_scales
     return entry
 

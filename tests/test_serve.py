@@ -247,3 +247,32 @@ def test_cryptojacking_detection_end_to_end(trained):
     assert any(s >= 7 and e <= 15 for s, e in rep.windows)
     # the untouched region stays quiet (measured == predicted median)
     assert not rep.flags[:7].any() and not rep.flags[15:].any()
+
+
+def test_results_entry_reanchors_monotone_metrics():
+    """Memory/usage scale factors re-anchor each query window to the last
+    learning-period value (reference: web-demo/dataloader.py:143-156) —
+    without it a monotonically growing metric shows absurd scales."""
+    from deeprest_amd.serve.results import build_results_entry
+
+    # monotone growth: learning period ends at 100, query continues to 160
+    meas = np.arange(1.0, 161.0)          # 160 steps
+    t_train = 100
+    pred = np.array([[130.0, 135.0, 140.0]])   # one query window, offset high
+    plain = build_results_entry(meas, {"ours": pred}, train_len=t_train)
+    re = build_results_entry(meas, {"ours": pred}, train_len=t_train,
+                             reanchor=True)
+    train_peak = 100.0
+    # plain: raw peak / train peak
+    assert plain["scale_ours"][0] == 140.0 / train_peak
+    # re-anchored: window shifted to start at meas[99] = 100 -> peak 110
+    assert re["scale_ours"][0] == (140.0 - 130.0 + 100.0) / train_peak
+    # ground-truth scale re-anchors the query segment the same way
+    seg = meas[100:103]
+    expected_gt = (seg[-1] - seg[0] + 100.0) / train_peak
+    assert abs(re["scale_groundtruth"][0] - expected_gt) < 1e-12
+    # separate-timeline form: anchor passed explicitly
+    re2 = build_results_entry(meas[100:], {"ours": pred}, train_len=0,
+                              train_peak=train_peak, reanchor=True,
+                              anchor_value=100.0)
+    assert re2["scale_ours"][0] == (140.0 - 130.0 + 100.0) / train_peak
