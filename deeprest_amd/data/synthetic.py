@@ -39,6 +39,11 @@ from .contract import validate_raw_data
 from .featurize import FeatureSpace, FeaturizedData
 
 DEFAULT_RESOURCES = ("cpu", "memory", "write-iops")
+# the reference estimates five resource types (reference:
+# resource-estimation/utils.py:8-26); pass resources=ALL_RESOURCES for the
+# full set — write-tp couples to write-iops via a per-component KB/op
+# factor, and disk usage integrates write throughput (monotone growth)
+ALL_RESOURCES = ("cpu", "memory", "write-iops", "write-tp", "usage")
 
 
 @dataclass
@@ -186,6 +191,17 @@ class SyntheticApp:
             cost = rng.uniform(0.2, 3.0, size=self._shape_comp[api].shape)
             self._shape_comp_w[api] = self._shape_comp[api] * cost
 
+        # five-resource-type dynamics, drawn from a SEPARATE derived stream
+        # so 3-resource configs keep their round-1 RNG sequence
+        # byte-identical (fixtures, bench configs):
+        # - write-tp = write-iops x per-component KB/op (correlated columns,
+        #   like a block device whose request size is a component property)
+        # - usage integrates write throughput: monotone disk growth
+        self._res_index = {r: i for i, r in enumerate(config.resources)}
+        rng5 = np.random.default_rng(config.seed + 90001)
+        self._kb_per_op = rng5.uniform(4.0, 64.0, size=C)
+        self._usage_rate = rng5.uniform(0.002, 0.02, size=C)
+
     # ------------------------------------------------------------------ traffic
     def traffic_plan(self, scale: float = 1.0, shape: str = "waves",
                      composition: Optional[Sequence[float]] = None) -> np.ndarray:
@@ -252,7 +268,25 @@ class SyntheticApp:
             state = self._res_ema * state + (1.0 - self._res_ema) * drive
             obs = self._res_base + state
             vals[t] = obs
+        ri = self._res_index
+        if "write-tp" in ri and "write-iops" in ri:
+            # throughput in KB = IOps x per-component request size
+            vals[:, :, ri["write-tp"]] = (
+                vals[:, :, ri["write-iops"]] * self._kb_per_op[None, :])
+        if "usage" in ri:
+            # disk usage integrates write throughput (falls back to its own
+            # affine response when no write-tp column exists)
+            growth_src = (vals[:, :, ri["write-tp"]] if "write-tp" in ri
+                          else vals[:, :, ri["usage"]])
+            increments = self._usage_rate[None, :] * growth_src
+            vals[:, :, ri["usage"]] = (
+                self._res_base[None, :, ri["usage"]]
+                + np.cumsum(increments, axis=0))
         vals *= 1.0 + cfg.resource_noise * self._rng.standard_normal(size=vals.shape)
+        if "usage" in ri:
+            # observation noise must not break monotonicity of disk usage
+            vals[:, :, ri["usage"]] = np.maximum.accumulate(
+                vals[:, :, ri["usage"]], axis=0)
         return np.maximum(vals, 0.0)
 
     # ------------------------------------------------------------------ outputs

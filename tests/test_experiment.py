@@ -133,3 +133,36 @@ def test_run_experiment_from_raw_wrapper():
     store = run_experiment_from_raw(raw, "from-raw", config=cfg,
                                     device=torch.device("cpu"))
     assert store.experiments() == ["from-raw"]
+
+
+def test_scenario_suite_five_resources_smoke():
+    """Scenario suite end-to-end with the reference's full resource set;
+    usage entries carry re-anchored scale factors."""
+    import torch
+
+    from deeprest_amd.data.synthetic import ALL_RESOURCES, SyntheticApp, SyntheticAppConfig
+    from deeprest_amd.engine.config import DataConfig, EngineConfig, TrainConfig
+    from deeprest_amd.engine.experiment import run_scenario_suite, scenario_error_tables
+    from deeprest_amd.models.net import DeepRestNetConfig
+
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=3, n_components=4, resources=ALL_RESOURCES,
+        windows_per_day=60, n_days=2, seed=31))
+    cfg = EngineConfig(
+        data=DataConfig(step_size=20, split=0.4),
+        train=TrainConfig(epochs=1, batch_size=16, baseline_epochs=2,
+                          log_every=0),
+        model=DeepRestNetConfig(d_model=32, n_heads=4, n_layers=1, d_ff=64,
+                                hidden=16, comp_dim=8, dropout=0.0))
+    store = run_scenario_suite(app, config=cfg, device=torch.device("cpu"),
+                               scenarios=[("waves_unseen-3x", {"scale": 3.0})])
+    exp = store.experiments()[0]
+    comps = store.results[exp]
+    # every component has entries for all five resource types
+    for comp, metrics in comps.items():
+        assert set(metrics.keys()) == set(ALL_RESOURCES)
+        e = metrics["usage"]
+        assert "scale_ours" in e and "scale_groundtruth" in e
+        assert all(np.isfinite(e["scale_ours"]))
+    tables = scenario_error_tables(store)
+    assert set(tables[exp].keys()) == {"bl-resrc", "bl-api", "bl-trace", "ours"}

@@ -33,11 +33,16 @@ def main():
     ap.add_argument("--windows-per-day", type=int, default=240)
     ap.add_argument("--seed", type=int, default=77)
     ap.add_argument("--out", default="results_scenarios.pkl")
+    ap.add_argument("--resources", type=int, choices=[3, 5], default=5,
+                    help="3 = round-1 default, 5 = the reference's full set")
     args = ap.parse_args()
+
+    from deeprest_amd.data.synthetic import ALL_RESOURCES, DEFAULT_RESOURCES
 
     app = SyntheticApp(SyntheticAppConfig(
         n_apis=args.apis, n_components=args.components,
         windows_per_day=args.windows_per_day, n_days=args.days,
+        resources=ALL_RESOURCES if args.resources == 5 else DEFAULT_RESOURCES,
         resource_noise=0.03, seed=args.seed))
 
     cfg = EngineConfig(
