@@ -201,6 +201,9 @@ class SyntheticApp:
         rng5 = np.random.default_rng(config.seed + 90001)
         self._kb_per_op = rng5.uniform(4.0, 64.0, size=C)
         self._usage_rate = rng5.uniform(0.002, 0.02, size=C)
+        # (EMA state, usage level) at the end of the last from-scratch
+        # generation — scenario query periods continue from here
+        self._base_carry: Optional[tuple] = None
 
     # ------------------------------------------------------------------ traffic
     def traffic_plan(self, scale: float = 1.0, shape: str = "waves",
@@ -256,19 +259,34 @@ class SyntheticApp:
             out[api] = picks
         return out
 
-    def _resources_from_invocations(self, inv: np.ndarray) -> np.ndarray:
-        """(T, C, R) ground-truth utilization from (T, C) invocation counts."""
+    def _resources_from_invocations(self, inv: np.ndarray,
+                                    continue_state: bool = False) -> np.ndarray:
+        """(T, C, R) ground-truth utilization from (T, C) invocation counts.
+
+        ``continue_state=True`` starts from where the most recent
+        from-scratch generation ended (EMA state + disk-usage level): a
+        scenario's query period is the SAME deployment continuing, not a
+        fresh one — restarting monotone metrics at zero would make query
+        ground truth incomparable to levels learned in training.
+        """
         cfg = self.config
         T, C = inv.shape
         R = len(cfg.resources)
+        ri = self._res_index
         vals = np.zeros((T, C, R))
-        state = np.zeros((C, R))
+        if continue_state and self._base_carry is not None:
+            state = self._base_carry[0].copy()
+            u = self._base_carry[1]
+            usage0 = u.copy() if u is not None else None
+        else:
+            state = np.zeros((C, R))
+            usage0 = (self._res_base[:, ri["usage"]].copy()
+                      if "usage" in ri else None)
         for t in range(T):
             drive = inv[t][:, None] * self._res_gain  # (C, R)
             state = self._res_ema * state + (1.0 - self._res_ema) * drive
             obs = self._res_base + state
             vals[t] = obs
-        ri = self._res_index
         if "write-tp" in ri and "write-iops" in ri:
             # throughput in KB = IOps x per-component request size
             vals[:, :, ri["write-tp"]] = (
@@ -279,9 +297,12 @@ class SyntheticApp:
             growth_src = (vals[:, :, ri["write-tp"]] if "write-tp" in ri
                           else vals[:, :, ri["usage"]])
             increments = self._usage_rate[None, :] * growth_src
-            vals[:, :, ri["usage"]] = (
-                self._res_base[None, :, ri["usage"]]
-                + np.cumsum(increments, axis=0))
+            vals[:, :, ri["usage"]] = usage0[None, :] + np.cumsum(
+                increments, axis=0)
+        if not continue_state:
+            self._base_carry = (
+                state.copy(),
+                vals[-1, :, ri["usage"]].copy() if "usage" in ri else None)
         vals *= 1.0 + cfg.resource_noise * self._rng.standard_normal(size=vals.shape)
         if "usage" in ri:
             # observation noise must not break monotonicity of disk usage
@@ -290,13 +311,16 @@ class SyntheticApp:
         return np.maximum(vals, 0.0)
 
     # ------------------------------------------------------------------ outputs
-    def generate_raw(self, plan: Optional[np.ndarray] = None) -> List[Dict[str, Any]]:
+    def generate_raw(self, plan: Optional[np.ndarray] = None,
+                     continue_state: bool = False) -> List[Dict[str, Any]]:
         """Full contract-format raw_data with span trees.
 
         ``plan``: optional (T, n_apis) expected-call matrix from
         ``traffic_plan(...)`` — pass a scenario variant (unseen scale /
         shape / composition) to generate query-period data for what-if
         evaluation (reference: locustfile-{scale,shape,composition}.py).
+        ``continue_state``: the plan continues the last from-scratch run's
+        deployment (EMA + disk-usage state carry over).
         """
         cfg = self.config
         api_calls = self.traffic_plan() if plan is None else np.asarray(plan)
@@ -309,7 +333,7 @@ class SyntheticApp:
         for api in self.apis:
             inv += shape_counts[api] @ self._shape_comp[api]
             drive += shape_counts[api] @ self._shape_comp_w[api]
-        res = self._resources_from_invocations(drive)
+        res = self._resources_from_invocations(drive, continue_state)
 
         raw = []
         for t in range(T):
@@ -328,11 +352,13 @@ class SyntheticApp:
         validate_raw_data(raw)
         return raw
 
-    def generate_featurized(self, plan: Optional[np.ndarray] = None) -> FeaturizedData:
+    def generate_featurized(self, plan: Optional[np.ndarray] = None,
+                            continue_state: bool = False) -> FeaturizedData:
         """Fast path: traffic matrix + resource series without building trees.
 
         ``plan`` as in :meth:`generate_raw` — a scenario traffic plan for
-        query-period/what-if data sharing this app's feature space.
+        query-period/what-if data sharing this app's feature space;
+        ``continue_state`` continues the last from-scratch deployment.
         """
         cfg = self.config
         api_calls = self.traffic_plan() if plan is None else np.asarray(plan)
@@ -349,7 +375,7 @@ class SyntheticApp:
             traffic[:, cols] += shape_counts[api] @ mat
             inv += shape_counts[api] @ self._shape_comp[api]
             drive += shape_counts[api] @ self._shape_comp_w[api]
-        res = self._resources_from_invocations(drive)
+        res = self._resources_from_invocations(drive, continue_state)
 
         resources = {}
         resource_components = {}

@@ -105,6 +105,40 @@ def scenario_error_tables(store: ResultsStore) -> Dict[str, Dict[str, Dict[str, 
     return tables
 
 
+def scenario_error_tables_by_resource(
+    store: ResultsStore,
+) -> Dict[str, Dict[str, Dict[str, Dict[str, float]]]]:
+    """Like scenario_error_tables but split per resource type:
+    tables[scenario][resource][estimator] -> percentile stats.  Resource
+    types have wildly different magnitudes (disk usage integrates to
+    thousands of MB while CPU sits at hundreds of millicores), so the
+    aggregate table's upper percentiles are dominated by the largest-scale
+    metric; this is the faithful per-type view (the reference prints
+    per-metric tables, estimate.py:112-122)."""
+    from ..utils.errors import error_percentiles
+
+    tables: Dict[str, Dict[str, Dict[str, Dict[str, float]]]] = {}
+    for exp in store.experiments():
+        errs: Dict[str, Dict[str, List[np.ndarray]]] = {}
+        for comp, metrics in store.results[exp].items():
+            for metric, entry in metrics.items():
+                meas = np.asarray(entry["measurement"])
+                for est in ("bl-resrc", "bl-api", "bl-trace", "ours"):
+                    key = f"prediction_{est}"
+                    if key not in entry:
+                        continue
+                    pred = np.asarray(entry[key])
+                    n = min(len(pred), len(meas))
+                    errs.setdefault(metric, {}).setdefault(est, []).append(
+                        np.abs(pred[:n] - meas[:n]))
+        tables[exp] = {
+            res: {est: error_percentiles(np.concatenate(v))
+                  for est, v in per_est.items()}
+            for res, per_est in errs.items()
+        }
+    return tables
+
+
 def _flat_series(y_windows: np.ndarray) -> np.ndarray:
     """Reconstruct the flat series from stride-1 windows (N, W).
 
@@ -186,7 +220,11 @@ def run_scenario_suite(
             plan_kw = dict(plan_kw)
             plan_kw["composition"] = np.roll(app.popularity,
                                              len(app.popularity) // 2)
-        qdata = app.generate_featurized(plan=app.traffic_plan(**plan_kw))
+        # the query period CONTINUES the learning deployment (EMA + disk
+        # usage state carry over) — all scenarios branch from the same
+        # end-of-training state, like the reference's day-8 query window
+        qdata = app.generate_featurized(plan=app.traffic_plan(**plan_kw),
+                                        continue_state=True)
         Xq = sliding_window(np.asarray(qdata.traffic, dtype=np.float32), step)
         eval_idx = list(range(0, len(Xq), step))
         Xq_eval = Xq[eval_idx]

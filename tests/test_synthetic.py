@@ -149,3 +149,29 @@ def test_three_resource_stream_unchanged_by_five_resource_support():
     first = np.asarray(data.resources[list(data.resources)[0]])
     assert first.shape == (40,)
     assert np.isfinite(first).all()
+
+
+def test_scenario_query_continues_deployment_state():
+    """A query-period generation with continue_state=True starts disk usage
+    where the learning period ended (not back at base)."""
+    from deeprest_amd.data.synthetic import ALL_RESOURCES
+
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=3, n_components=4, resources=ALL_RESOURCES,
+        windows_per_day=60, n_days=2, seed=9))
+    base = app.generate_featurized()
+    comp = app.all_components[0]
+    base_usage = np.asarray(base.resources[f"{comp}_usage"])
+    q = app.generate_featurized(plan=app.traffic_plan(scale=2.0),
+                                continue_state=True)
+    q_usage = np.asarray(q.resources[f"{comp}_usage"])
+    # continues near the base end (within noise), far above the base start
+    assert q_usage[0] > base_usage[-1] * 0.9
+    assert q_usage[0] > base_usage[0] * 1.5
+    assert np.all(np.diff(q_usage) >= 0)
+    # a second continued scenario branches from the SAME end-of-training
+    # state, not from the previous scenario's end
+    q2 = app.generate_featurized(plan=app.traffic_plan(shape="flat"),
+                                 continue_state=True)
+    q2_usage = np.asarray(q2.resources[f"{comp}_usage"])
+    assert abs(q2_usage[0] - q_usage[0]) < 0.25 * q_usage[0]

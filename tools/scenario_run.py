@@ -20,7 +20,8 @@ import torch
 from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
 from deeprest_amd.engine.config import DataConfig, EngineConfig, TrainConfig
 from deeprest_amd.engine.experiment import (run_scenario_suite,
-                                            scenario_error_tables)
+                                            scenario_error_tables,
+                                            scenario_error_tables_by_resource)
 from deeprest_amd.models.net import DeepRestNetConfig
 
 
@@ -56,16 +57,24 @@ def main():
     store = run_scenario_suite(app, base_name="synthetic", config=cfg,
                                device=dev)
     store.save(args.out)
-    tables = scenario_error_tables(store)
-    for exp, per_est in tables.items():
-        print(f"===== {exp} =====")
+
+    def show(per_est, indent=""):
         for est in ("bl-resrc", "bl-api", "bl-trace", "ours"):
             t = per_est.get(est)
             if t is None:
                 continue
-            print(f"   {est:>9} => Median: {t['median']:.4f} | "
+            print(f"{indent}   {est:>9} => Median: {t['median']:.4f} | "
                   f"95-th: {t['p95']:.4f} | 99-th: {t['p99']:.4f} | "
                   f"Max: {t['max']:.4f}")
+
+    tables = scenario_error_tables(store)
+    by_res = scenario_error_tables_by_resource(store)
+    for exp, per_est in tables.items():
+        print(f"===== {exp} (all metrics aggregated) =====")
+        show(per_est)
+        for res, res_est in by_res[exp].items():
+            print(f"  --- {res} ---")
+            show(res_est, "  ")
     print(f"saved {args.out}")
 
 
