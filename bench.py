@@ -219,16 +219,36 @@ def main():
 
     step = TrainStep(model, opt, dist_ctx=dist_ctx,
                      autocast_dtype=autocast_dtype if use_autocast else None)
-    # Whole-step hipGraph capture is available (DEEPREST_GRAPH_STEP=1) but OFF
-    # by default: replay requires copying each batch into the graph's static
-    # input buffer (~1.5 GB/step at this config) while the eager path feeds
-    # zero-copy views, and the measured copy cost exceeds the launch-gap
-    # savings (37.7k vs 38.1k windows/s).  It pays only when inputs already
-    # arrive in a fixed staging buffer (e.g. streamed ingestion).
-    if on_gpu and os.environ.get("DEEPREST_GRAPH_STEP", "0") == "1":
-        step.try_capture(X[:B], y[:B], warmup=2)
+    # hipGraph training (DEEPREST_GRAPH_STEP=1, single-process): capture K
+    # graphs over K persistent SLICES of the resident training set, sharing
+    # one capture pool, and cycle replays — zero input copies (round 1's
+    # single-graph variant lost because every replay paid a ~1.5 GB copy
+    # into its static buffer) and no per-launch gaps.
+    graph_cycle = None
+    if (on_gpu and world == 1
+            and os.environ.get("DEEPREST_GRAPH_STEP", "0") == "1"):
+        from deeprest_amd.engine.graphstep import GraphedTrainStep
+
+        K = int(os.environ.get("DEEPREST_GRAPH_SLICES", "4"))
+        offs = sorted({((k * B + 37 * k) % max(n - B, 1)) for k in range(K)})
+        cycle = []
+        pool = None
+        try:
+            for o in offs:
+                g = GraphedTrainStep(
+                    model, opt, lambda out, t: model.loss(out.float(), t),
+                    X[o : o + B], y[o : o + B],
+                    autocast_dtype=autocast_dtype if use_autocast else None,
+                    warmup=2, static_inputs=True, pool=pool)
+                pool = pool or g.graph.pool()
+                cycle.append(g)
+            graph_cycle = cycle
+        except Exception as exc:
+            print(f"graph cycle capture failed ({exc}); eager", file=sys.stderr)
 
     def run_step(i: int):
+        if graph_cycle is not None:
+            return graph_cycle[i % len(graph_cycle)].replay()
         # each rank walks a different offset sequence (its DP shard)
         s = ((i + 3 * rank) * B + rank * 17) % max(n - B, 1)
         return step(X[s : s + B], y[s : s + B])

@@ -46,13 +46,19 @@ class GraphedTrainStep:
                  loss_fn: Callable[[torch.Tensor, torch.Tensor], torch.Tensor],
                  x_example: torch.Tensor, y_example: torch.Tensor,
                  autocast_dtype: Optional[torch.dtype] = torch.bfloat16,
-                 warmup: int = 3):
+                 warmup: int = 3, static_inputs: bool = False, pool=None):
+        """``static_inputs=True``: capture directly over the PASSED tensors
+        (caller keeps them alive and stable) — a replay then reads them
+        zero-copy, e.g. a persistent slice of the resident training set.
+        ``pool``: share one capture memory pool across several graphs
+        (multi-offset cycling) instead of one activation arena each."""
         self.model = model
         self.optimizer = optimizer
         self.loss_fn = loss_fn
         self.autocast_dtype = autocast_dtype
-        self.static_x = x_example.clone()
-        self.static_y = y_example.clone()
+        self.static_x = x_example if static_inputs else x_example.clone()
+        self.static_y = y_example if static_inputs else y_example.clone()
+        self._pool = pool
 
         # warmup on a side stream: allocates grads, Adam state, the fused
         # all-reduce bucket, autograd workspace — everything whose pointers
@@ -70,7 +76,7 @@ class GraphedTrainStep:
         torch.cuda.current_stream().wait_stream(side)
 
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph, stream=side):
+        with torch.cuda.graph(self.graph, stream=side, pool=self._pool):
             self.static_loss = self._one_step()
 
     def _one_step(self) -> torch.Tensor:
@@ -93,8 +99,15 @@ class GraphedTrainStep:
         return loss
 
     def run(self, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
-        self.static_x.copy_(x, non_blocking=True)
-        self.static_y.copy_(y, non_blocking=True)
+        if x.data_ptr() != self.static_x.data_ptr():
+            self.static_x.copy_(x, non_blocking=True)
+        if y.data_ptr() != self.static_y.data_ptr():
+            self.static_y.copy_(y, non_blocking=True)
+        self.graph.replay()
+        return self.static_loss
+
+    def replay(self) -> torch.Tensor:
+        """Zero-copy replay over the captured (static) inputs."""
         self.graph.replay()
         return self.static_loss
 
