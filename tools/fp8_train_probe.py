@@ -32,9 +32,12 @@ class _FakeQuant(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):
         # per-tensor dynamic scale to the e4m3 range, like an fp8 GEMM's
-        # scaling factor would apply
+        # scaling factor would apply.  NOTE: the ROCm variant is
+        # e4m3fnuz whose max finite value is 240 (not OCP e4m3's 448) —
+        # scaling to 448 overflowed to NaN
+        fmax = float(torch.finfo(FP8).max) * 0.98
         amax = x.detach().abs().amax().clamp(min=1e-8)
-        scale = 448.0 / amax
+        scale = fmax / amax
         return (x * scale).to(FP8).to(x.dtype) / scale
 
     @staticmethod
