@@ -124,6 +124,39 @@ def cmd_synthesize(args):
     return 0
 
 
+def cmd_collect(args):
+    from .data.collector import Collector
+    from .data.contract import save_raw_data
+    from .data.poller import CollectionPoller
+
+    queries = {}
+    for spec in args.query or []:
+        resource, _, promql = spec.partition("=")
+        if not promql:
+            raise SystemExit(f"--query needs resource=promql, got {spec!r}")
+        queries[resource] = promql
+    collector = Collector(window_sec=args.interval)
+    poller = CollectionPoller(
+        collector,
+        jaeger_url=args.jaeger,
+        prometheus_url=args.prometheus,
+        services=args.services.split(","),
+        queries=queries,
+        component_label=args.component_label,
+        interval_sec=args.interval,
+    )
+    n = poller.run(
+        max_polls=args.max_polls,
+        duration_sec=args.duration,
+        on_poll=lambda s: print(f"poll: +{s['traces']} traces, "
+                                f"+{s['samples']} samples"),
+    )
+    windows = collector.windows()
+    save_raw_data(windows, args.out)
+    print(f"{n} polls -> {len(windows)} contract windows -> {args.out}")
+    return 0
+
+
 def cmd_serve(args):
     import uvicorn
 
@@ -159,6 +192,22 @@ def main(argv=None):
     s.add_argument("--seed", type=int, default=0)
     s.add_argument("--list-apis", action="store_true")
 
+    c = sub.add_parser("collect", help="live Jaeger/Prometheus polling loop"
+                       " -> raw_data.pkl (the reference's L3 collection plane)")
+    c.add_argument("--jaeger", required=True, help="Jaeger query base URL")
+    c.add_argument("--prometheus", required=True, help="Prometheus base URL")
+    c.add_argument("--services", required=True,
+                   help="comma-separated Jaeger service names")
+    c.add_argument("--query", action="append", metavar="RESOURCE=PROMQL",
+                   help="repeatable, e.g. cpu=rate(container_cpu_usage"
+                        "_seconds_total[1m])*1000")
+    c.add_argument("--component-label", default="component")
+    c.add_argument("--interval", type=float, default=5.0,
+                   help="poll + window seconds (reference scrape: 5 s)")
+    c.add_argument("--duration", type=float, default=None)
+    c.add_argument("--max-polls", type=int, default=None)
+    c.add_argument("--out", default="raw_data.pkl")
+
     v = sub.add_parser("serve")
     v.add_argument("--checkpoint", default=None)
     v.add_argument("--results", default=None, help="results.pkl to browse at /results")
@@ -173,6 +222,7 @@ def main(argv=None):
         "train": cmd_train,
         "experiment": cmd_experiment,
         "synthesize": cmd_synthesize,
+        "collect": cmd_collect,
         "serve": cmd_serve,
     }[args.cmd](args)
 
