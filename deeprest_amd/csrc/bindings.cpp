@@ -40,7 +40,7 @@ void dr_gru_bwd(const void* grad_h, const void* w_img, const void* w_fwd,
                 const void* saves, void* dpre, float* dh0, int B, int TT, int C,
                 int reverse, int is_bf16, hipStream_t stream);
 void dr_gru_bwd_reduce(const void* dpre, const void* gamma, const void* xg,
-                       void* dxg, float* dgamma, float* dbeta, int64_t BT, int C,
+                       float* dxg, float* dgamma, float* dbeta, int64_t BT, int C,
                        int is_bf16, hipStream_t stream);
 void dr_mha_fwd(const void* q, const void* k, const void* v, void* o, float* lse,
                 int64_t BH, int T_len, int D, float scale, int is_bf16,
@@ -268,14 +268,18 @@ std::vector<at::Tensor> gru_bwd_reduce(at::Tensor dpre, at::Tensor gamma,
   int64_t BT = dpre.size(0) * dpre.size(1);
   int C = (int)dpre.size(2);
   TORCH_CHECK(xg.is_contiguous() && gamma.is_contiguous() && dpre.is_contiguous());
-  auto dxg = at::empty({dpre.size(0), dpre.size(1), 384}, dpre.options());
+  // fused kernel accumulates dxg in f32 (multi-c-block atomics); cast back
+  // to the compute dtype afterwards (one cheap 94->47 MB pass)
+  auto dxg = at::zeros({dpre.size(0), dpre.size(1), 384},
+                       dpre.options().dtype(at::kFloat));
   auto dgamma = at::zeros({C, 384}, dpre.options().dtype(at::kFloat));
   auto dbeta = at::zeros({C, 512}, dpre.options().dtype(at::kFloat));
   dr_gru_bwd_reduce(dpre.data_ptr(), gamma.data_ptr(), xg.data_ptr(),
-                    dxg.data_ptr(), dgamma.data_ptr<float>(),
+                    dxg.data_ptr<float>(), dgamma.data_ptr<float>(),
                     dbeta.data_ptr<float>(), BT, C,
                     dpre.scalar_type() == at::kBFloat16, cur_stream());
-  return {dxg, dgamma, dbeta};
+  auto dxg_out = dpre.scalar_type() == at::kFloat ? dxg : dxg.to(dpre.scalar_type());
+  return {dxg_out, dgamma, dbeta};
 }
 
 // -------------------------------------------------------------------- mha
