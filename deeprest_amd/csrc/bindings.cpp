@@ -40,8 +40,9 @@ void dr_gru_bwd(const void* grad_h, const void* w_img, const void* w_fwd,
                 const void* saves, void* dpre, float* dh0, int B, int TT, int C,
                 int reverse, int is_bf16, hipStream_t stream);
 void dr_gru_bwd_reduce(const void* dpre, const void* gamma, const void* xg,
-                       float* dxg, float* dgamma, float* dbeta, int64_t BT, int C,
-                       int is_bf16, hipStream_t stream);
+                       void* dxg, float* dgamma, float* dbeta, void* gamma_pi,
+                       void* xg_pi, int64_t BT, int C, int is_bf16,
+                       hipStream_t stream);
 void dr_mha_fwd(const void* q, const void* k, const void* v, void* o, float* lse,
                 int64_t BH, int T_len, int D, float scale, int is_bf16,
                 hipStream_t stream);
@@ -268,18 +269,18 @@ std::vector<at::Tensor> gru_bwd_reduce(at::Tensor dpre, at::Tensor gamma,
   int64_t BT = dpre.size(0) * dpre.size(1);
   int C = (int)dpre.size(2);
   TORCH_CHECK(xg.is_contiguous() && gamma.is_contiguous() && dpre.is_contiguous());
-  // fused kernel accumulates dxg in f32 (multi-c-block atomics); cast back
-  // to the compute dtype afterwards (one cheap 94->47 MB pass)
-  auto dxg = at::zeros({dpre.size(0), dpre.size(1), 384},
-                       dpre.options().dtype(at::kFloat));
+  auto dxg = at::empty({dpre.size(0), dpre.size(1), 384}, dpre.options());
   auto dgamma = at::zeros({C, 384}, dpre.options().dtype(at::kFloat));
   auto dbeta = at::zeros({C, 512}, dpre.options().dtype(at::kFloat));
+  // pi-layout staging images (see pi_permute_rows_kernel)
+  auto gamma_pi = at::empty_like(gamma);
+  auto xg_pi = at::empty_like(xg);
   dr_gru_bwd_reduce(dpre.data_ptr(), gamma.data_ptr(), xg.data_ptr(),
-                    dxg.data_ptr<float>(), dgamma.data_ptr<float>(),
-                    dbeta.data_ptr<float>(), BT, C,
+                    dxg.data_ptr(), dgamma.data_ptr<float>(),
+                    dbeta.data_ptr<float>(), gamma_pi.data_ptr(),
+                    xg_pi.data_ptr(), BT, C,
                     dpre.scalar_type() == at::kBFloat16, cur_stream());
-  auto dxg_out = dpre.scalar_type() == at::kFloat ? dxg : dxg.to(dpre.scalar_type());
-  return {dxg_out, dgamma, dbeta};
+  return {dxg, dgamma, dbeta};
 }
 
 // -------------------------------------------------------------------- mha

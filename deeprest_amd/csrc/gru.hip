@@ -735,12 +735,35 @@ __global__ __launch_bounds__(NSUB * THREADS) void gru_bwd_kernel(
 // dominated the profiled step).  Both kernels understand the pi layout and
 // emit NATURAL-order outputs.
 
+// natural (r, 3H) -> pi-layout copy: position g*H + cc*8 + e holds natural
+// column g*H + cc + 16e.  Run ONCE per tensor so the reduction kernels'
+// inner loops replace 8 scalar loads with one 16-byte vector load (the
+// scalar-load issue rate, not HBM bandwidth, bounded the old versions).
+template <typename T>
+__global__ void pi_permute_rows_kernel(const T* __restrict__ in,
+                                       T* __restrict__ out, int64_t N) {
+  const int64_t n_chunks = N * 48;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < n_chunks; idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t r = idx / 48;
+    const int rem = (int)(idx % 48);
+    const int g = rem / 16;
+    const int cc = rem % 16;
+    const T* src = in + r * G3H + g * H + cc;
+    T* dst = out + r * G3H + g * H + cc * 8;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) dst[e] = src[e * 16];
+  }
+}
+
 // dxg[bt, j] = sum_c dpre[bt, c, pi(j)] * gamma[c, j]   (j in [0, 3H))
 // thread chunk = (bt, gate, c_col): 8 pi-contiguous dpre values per c.
+// gamma_pi is the pi-permuted gamma image: one ld8 per c instead of 8
+// scalar loads (tiny tensor, L1/L2-resident across all bt chunks).
 template <typename T>
-__global__ void gru_dxg_kernel(const T* __restrict__ dpre,   // (BT, C, 4H) pi
-                               const T* __restrict__ gamma,  // (C, 3H)
-                               T* __restrict__ dxg,          // (BT, 3H)
+__global__ void gru_dxg_kernel(const T* __restrict__ dpre,     // (BT, C, 4H) pi
+                               const T* __restrict__ gamma_pi, // (C, 3H) pi
+                               T* __restrict__ dxg,            // (BT, 3H)
                                int64_t BT, int C) {
   const int64_t n_chunks = BT * 48;      // 3 gates x 16 c_col groups
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < n_chunks;
@@ -755,11 +778,11 @@ __global__ void gru_dxg_kernel(const T* __restrict__ dpre,   // (BT, C, 4H) pi
     for (int e = 0; e < 8; ++e) acc[e] = 0.f;
     const T* base = dpre + bt * C * G4H + pi0;
     for (int c = 0; c < C; ++c) {
-      float d[8];
+      float d[8], gm[8];
       ld8(base + (int64_t)c * G4H, d);
-      const T* gm = gamma + (int64_t)c * G3H + g * H + cc;
+      ld8(gamma_pi + (int64_t)c * G3H + pi0, gm);
 #pragma unroll
-      for (int e = 0; e < 8; ++e) acc[e] += d[e] * ldf(gm + e * 16);
+      for (int e = 0; e < 8; ++e) acc[e] += d[e] * gm[e];
     }
     T* out = dxg + bt * G3H + g * H + cc;
 #pragma unroll
@@ -770,11 +793,12 @@ __global__ void gru_dxg_kernel(const T* __restrict__ dpre,   // (BT, C, 4H) pi
 // dgamma[c, j<3H] = sum_bt dpre[bt, c, pi(j)] * xg[bt, j]
 // dbeta4[c, j<4H] = sum_bt dpre[bt, c, pi(j)]   (4th slice -> db_hh_n)
 // grid.y slices BT; f32 atomics finalize.  Natural-order outputs.
+// xg_pi is the pi-permuted xg image: one ld8 per bt instead of 8 scalars.
 template <typename T>
-__global__ void gru_dgamma_kernel(const T* __restrict__ dpre,  // (BT, C, 4H) pi
-                                  const T* __restrict__ xg,    // (BT, 3H)
-                                  float* __restrict__ dgamma,  // (C, 3H) zeroed
-                                  float* __restrict__ dbeta4,  // (C, 4H) zeroed
+__global__ void gru_dgamma_kernel(const T* __restrict__ dpre,   // (BT, C, 4H) pi
+                                  const T* __restrict__ xg_pi,  // (BT, 3H) pi
+                                  float* __restrict__ dgamma,   // (C, 3H) zeroed
+                                  float* __restrict__ dbeta4,   // (C, 4H) zeroed
                                   int64_t BT, int C) {
   const int n_threads_needed = C * 64;   // 4 gates x 16 c_col groups
   const int tid_g = blockIdx.x * blockDim.x + threadIdx.x;
@@ -794,9 +818,10 @@ __global__ void gru_dgamma_kernel(const T* __restrict__ dpre,  // (BT, C, 4H) pi
     float d[8];
     ld8(dpre + (bt * C + c) * G4H + pi0, d);
     if (has_x) {
-      const T* x = xg + bt * G3H + g * H + cc;
+      float xv[8];
+      ld8(xg_pi + bt * G3H + pi0, xv);
 #pragma unroll
-      for (int e = 0; e < 8; ++e) accg[e] += d[e] * ldf(x + e * 16);
+      for (int e = 0; e < 8; ++e) accg[e] += d[e] * xv[e];
     }
 #pragma unroll
     for (int e = 0; e < 8; ++e) accb[e] += d[e];
@@ -1014,14 +1039,36 @@ static void gru_bwd_launch_t(const void* grad_h, const void* w_img,
 
 template <typename T>
 static void gru_reduce_launch_t(const void* dpre, const void* gamma, const void* xg,
-                                float* dxg, float* dgamma, float* dbeta, int64_t BT,
+                                void* dxg, float* dgamma, float* dbeta,
+                                void* gamma_pi, void* xg_pi, int64_t BT,
                                 int C, hipStream_t stream) {
-  int gx = (C + 15) / 16;
-  // ~4 workgroups per CU over the 256-CU chip; never more y-slices than bt rows
-  int gy = (int)std::min<int64_t>(BT, std::max(1, 1024 / gx));
-  hipLaunchKernelGGL((gru_reduce_fused_kernel<T>), dim3(gx, gy), dim3(256), 0,
-                     stream, (const T*)dpre, (const T*)xg, (const T*)gamma,
-                     dxg, dgamma, dbeta, BT, C);
+  // stage pi-layout images of gamma and xg (once per backward, ~47 MB at
+  // bench scale) so the reduction inner loops are pure 16-byte loads
+  {
+    int64_t n = (int64_t)C * 48;
+    int grid = (int)std::min<int64_t>((n + 255) / 256, 4096);
+    hipLaunchKernelGGL((pi_permute_rows_kernel<T>), dim3(grid), dim3(256), 0,
+                       stream, (const T*)gamma, (T*)gamma_pi, (int64_t)C);
+  }
+  {
+    int64_t n = BT * 48;
+    int grid = (int)std::min<int64_t>((n + 255) / 256, 4096);
+    hipLaunchKernelGGL((pi_permute_rows_kernel<T>), dim3(grid), dim3(256), 0,
+                       stream, (const T*)xg, (T*)xg_pi, BT);
+  }
+  {
+    int64_t n = BT * 48;
+    int grid = (int)std::min<int64_t>((n + 255) / 256, 4096);
+    hipLaunchKernelGGL((gru_dxg_kernel<T>), dim3(grid), dim3(256), 0, stream,
+                       (const T*)dpre, (const T*)gamma_pi, (T*)dxg, BT, C);
+  }
+  {
+    int n_threads = C * 64;
+    int gx = (n_threads + 255) / 256;
+    int gy = 128;  // BT slices (parallelism for small C; atomics stay cheap)
+    hipLaunchKernelGGL((gru_dgamma_kernel<T>), dim3(gx, gy), dim3(256), 0, stream,
+                       (const T*)dpre, (const T*)xg_pi, dgamma, dbeta, BT, C);
+  }
 }
 
 }  // namespace dr
@@ -1056,14 +1103,15 @@ void dr_gru_bwd(const void* grad_h, const void* w_img, const void* w_fwd,
 }
 
 void dr_gru_bwd_reduce(const void* dpre, const void* gamma, const void* xg,
-                       float* dxg, float* dgamma, float* dbeta, int64_t BT, int C,
-                       int is_bf16, hipStream_t stream) {
+                       void* dxg, float* dgamma, float* dbeta, void* gamma_pi,
+                       void* xg_pi, int64_t BT, int C, int is_bf16,
+                       hipStream_t stream) {
   if (is_bf16)
-    dr::gru_reduce_launch_t<uint16_t>(dpre, gamma, xg, dxg, dgamma, dbeta, BT, C,
-                                      stream);
+    dr::gru_reduce_launch_t<uint16_t>(dpre, gamma, xg, dxg, dgamma, dbeta,
+                                      gamma_pi, xg_pi, BT, C, stream);
   else
-    dr::gru_reduce_launch_t<float>(dpre, gamma, xg, dxg, dgamma, dbeta, BT, C,
-                                   stream);
+    dr::gru_reduce_launch_t<float>(dpre, gamma, xg, dxg, dgamma, dbeta,
+                                   gamma_pi, xg_pi, BT, C, stream);
 }
 
 }  // extern "C"
