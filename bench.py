@@ -136,7 +136,8 @@ def shared_train_tensors(args, dist_ctx):
 def accuracy_probe(epochs: int, device: torch.device, n_apis: int = 13,
                    n_components: int = 12, windows_per_day: int = 240,
                    n_days: int = 8, step_size: int = 60,
-                   baseline_epochs: int = 100) -> dict:
+                   baseline_epochs: int = 100,
+                   lr_schedule: str = "none") -> dict:
     """Train at the reference config on a reference-scale app and report the
     three-estimator mean median absolute error (the prediction-MAE half of
     BASELINE.json's metric; anchors: resource-estimation/README.md:88-98)."""
@@ -155,7 +156,8 @@ def accuracy_probe(epochs: int, device: torch.device, n_apis: int = 13,
         data=DataConfig(step_size=step_size, split=0.40),
         train=TrainConfig(epochs=epochs, batch_size=32, lr=1e-3,
                           eval_cycles=9, baseline_epochs=baseline_epochs,
-                          log_every=0, eval_every=5, graph_step=True),
+                          log_every=0, eval_every=5, graph_step=True,
+                          lr_schedule=lr_schedule),
         model=DeepRestNetConfig(dropout=0.1),
     )
     trainer = Trainer(data, cfg, device=device)

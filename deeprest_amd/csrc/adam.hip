@@ -16,11 +16,17 @@ template <bool STEP_DEV>
 __global__ void fused_adam_kernel(const int64_t* __restrict__ meta, int nt,
                                   int64_t total, float lr, float beta1, float beta2,
                                   float eps, float weight_decay, float bias_c1,
-                                  float bias_c2, const int* __restrict__ step_ptr) {
+                                  float bias_c2, const int* __restrict__ step_ptr,
+                                  const float* __restrict__ lr_ptr) {
   if (STEP_DEV) {
+    // step counter AND learning rate live in device memory so a captured
+    // replay keeps exact bias correction and follows LR schedules (the
+    // host updates the scalar between replays; a by-value lr would be
+    // frozen at its capture-time value)
     const float s = (float)*step_ptr;
     bias_c1 = 1.f - __powf(beta1, s);
     bias_c2 = 1.f - __powf(beta2, s);
+    lr = *lr_ptr;
   }
   const int64_t* p_ptrs = meta;
   const int64_t* g_ptrs = meta + nt;
@@ -71,18 +77,19 @@ void dr_fused_adam(const int64_t* meta, int nt, int64_t total, float lr, float b
   float bias_c2 = 1.f - powf(beta2, (float)step);
   hipLaunchKernelGGL((dr::fused_adam_kernel<false>), dim3(grid), dim3(block), 0,
                      stream, meta, nt, total, lr, beta1, beta2, eps, weight_decay,
-                     bias_c1, bias_c2, nullptr);
+                     bias_c1, bias_c2, nullptr, nullptr);
 }
 
-void dr_fused_adam_dev(const int64_t* meta, int nt, int64_t total, float lr,
-                       float beta1, float beta2, float eps, float weight_decay,
-                       const int* step_ptr, hipStream_t stream) {
+void dr_fused_adam_dev(const int64_t* meta, int nt, int64_t total,
+                       const float* lr_ptr, float beta1, float beta2, float eps,
+                       float weight_decay, const int* step_ptr,
+                       hipStream_t stream) {
   const int block = 256;
   int grid = (int)std::min<int64_t>((total + block - 1) / block, 4096);
   if (grid == 0) grid = 1;
   hipLaunchKernelGGL((dr::fused_adam_kernel<true>), dim3(grid), dim3(block), 0,
-                     stream, meta, nt, total, lr, beta1, beta2, eps, weight_decay,
-                     0.f, 0.f, step_ptr);
+                     stream, meta, nt, total, 0.f, beta1, beta2, eps, weight_decay,
+                     0.f, 0.f, step_ptr, lr_ptr);
 }
 
 }  // extern "C"

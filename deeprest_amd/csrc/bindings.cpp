@@ -27,9 +27,10 @@ void dr_pinball_bwd(const float* out, const float* labels, const float* quantile
 void dr_fused_adam(const int64_t* meta, int nt, int64_t total, float lr, float beta1,
                    float beta2, float eps, float weight_decay, int step,
                    hipStream_t stream);
-void dr_fused_adam_dev(const int64_t* meta, int nt, int64_t total, float lr,
-                       float beta1, float beta2, float eps, float weight_decay,
-                       const int* step_ptr, hipStream_t stream);
+void dr_fused_adam_dev(const int64_t* meta, int nt, int64_t total,
+                       const float* lr_ptr, float beta1, float beta2, float eps,
+                       float weight_decay, const int* step_ptr,
+                       hipStream_t stream);
 void dr_gru_fwd(const void* xg, const void* gamma, const void* beta,
                 const void* w_hh, const float* b_hh, const void* h0, void* h_all,
                 void* saves, int B, int TT, int C, int reverse, int save,
@@ -173,16 +174,19 @@ void fused_adam(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
 // device int32 scalar, so a replayed graph keeps exact bias correction with
 // zero host work per step.
 void fused_adam_capturable(at::Tensor meta_dev, int64_t nt, int64_t total,
-                           at::Tensor step_t, double lr, double beta1, double beta2,
-                           double eps, double weight_decay) {
+                           at::Tensor step_t, at::Tensor lr_t, double beta1,
+                           double beta2, double eps, double weight_decay) {
   const at::cuda::CUDAGuard guard(meta_dev.device());
   TORCH_CHECK(meta_dev.is_cuda() && meta_dev.scalar_type() == at::kLong &&
               meta_dev.is_contiguous());
   TORCH_CHECK(step_t.is_cuda() && step_t.scalar_type() == at::kInt &&
               step_t.numel() == 1);
-  dr_fused_adam_dev(meta_dev.data_ptr<int64_t>(), (int)nt, total, (float)lr,
-                    (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
-                    step_t.data_ptr<int>(), cur_stream());
+  TORCH_CHECK(lr_t.is_cuda() && lr_t.scalar_type() == at::kFloat &&
+              lr_t.numel() == 1);
+  dr_fused_adam_dev(meta_dev.data_ptr<int64_t>(), (int)nt, total,
+                    lr_t.data_ptr<float>(), (float)beta1, (float)beta2,
+                    (float)eps, (float)weight_decay, step_t.data_ptr<int>(),
+                    cur_stream());
 }
 
 // -------------------------------------------------------------------- gru

@@ -47,6 +47,8 @@ class TrainConfig:
     log_every: int = 1
     eval_every: int = 1                   # per-epoch eval cadence (rank 0);
                                           # the final epoch always evaluates
+    lr_schedule: str = "none"             # "none" (reference: fixed lr) or
+                                          # "cosine" (decay to 5% over the run)
 
 
 @dataclass

@@ -156,6 +156,16 @@ class Trainer:
         total_samples = 0
         t_start = time.perf_counter()
         for epoch in range(self.start_epoch, cfg.epochs):
+            if cfg.lr_schedule == "cosine":
+                # decay to 5% of the base lr; graph-safe — FusedAdam.set_lr
+                # refreshes the device scalar captured replays read
+                frac = epoch / max(cfg.epochs - 1, 1)
+                lr_now = cfg.lr * (0.05 + 0.95 * 0.5 * (1.0 + np.cos(np.pi * frac)))
+                if hasattr(self.optimizer, "set_lr"):
+                    self.optimizer.set_lr(lr_now)
+                else:
+                    for group in self.optimizer.param_groups:
+                        group["lr"] = lr_now
             self.model.train()
             perm = torch.randperm(n, generator=gen)
             # data-parallel shard: equal-length strided slices — truncate to a

@@ -71,6 +71,7 @@ class FusedAdam(torch.optim.Optimizer):
             raise ValueError("capturable FusedAdam supports a single param group")
         self.capturable = capturable
         self._step_t: torch.Tensor | None = None   # device int32 scalar
+        self._lr_t: torch.Tensor | None = None     # device f32 scalar (lr)
         self._meta: torch.Tensor | None = None     # device int64 pointer table
         self._meta_key = None                      # pointer tuple behind _meta
         self._meta_total = 0
@@ -97,11 +98,24 @@ class FusedAdam(torch.optim.Optimizer):
             # this call, so seed the device counter one behind it
             start = int(self.state[params[0]]["step"]) - 1
             self._step_t = torch.full((1,), start, dtype=torch.int32, device=dev)
+        if self._lr_t is None:
+            # device-resident lr: captured replays read it, so LR schedules
+            # work under hipGraph (set_lr updates it between replays)
+            self._lr_t = torch.full((1,), float(group["lr"]),
+                                    dtype=torch.float32, device=dev)
         self._step_t += 1  # device add: captured, so replays keep counting
         beta1, beta2 = group["betas"]
         ext.fused_adam_capturable(self._meta, self._meta_nt, self._meta_total,
-                                  self._step_t, group["lr"], beta1, beta2,
+                                  self._step_t, self._lr_t, beta1, beta2,
                                   group["eps"], group["weight_decay"])
+
+    def set_lr(self, lr: float) -> None:
+        """Schedule-safe lr update: refreshes the host groups AND the
+        device scalar a captured (replayed) step reads."""
+        for group in self.param_groups:
+            group["lr"] = lr
+        if self._lr_t is not None:
+            self._lr_t.fill_(float(lr))
 
     def state_dict(self):
         # graph replays advance only the device counter; sync the host step

@@ -138,3 +138,15 @@ def test_resume_reproduces_uninterrupted_run(tmp_path):
     wa = torch.cat([p.detach().reshape(-1) for p in ta.model.parameters()])
     wb = torch.cat([p.detach().reshape(-1) for p in tb2.model.parameters()])
     assert torch.equal(wa, wb)
+
+
+def test_cosine_lr_schedule_cpu():
+    data = tiny_data()
+    cfg = tiny_config(epochs=3)
+    cfg.train.lr_schedule = "cosine"
+    cfg.train.run_baselines = False
+    trainer = Trainer(data, cfg, device=torch.device("cpu"))
+    result = trainer.train()
+    assert np.isfinite(result.train_losses).all()
+    # final-epoch lr decayed to the 5% floor
+    assert abs(trainer.optimizer.param_groups[0]["lr"] - 0.05 * cfg.train.lr) < 1e-9
