@@ -23,6 +23,9 @@ class DataConfig:
     raw_path: Optional[str] = None        # raw_data.pkl path (featurize first)
     step_size: int = 60                   # window length (estimate.py:18)
     split: float = 0.40                   # train fraction (estimate.py:17)
+    target_transform: str = "none"        # "log1p": quantile-preserving
+                                          # compression for unseen-scale
+                                          # extrapolation (dataset.py)
     # synthetic app fallback (when no input/raw path given)
     synth_apis: int = 8
     synth_components: int = 12

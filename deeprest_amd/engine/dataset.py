@@ -29,9 +29,17 @@ class EstimationDataset:
         data: FeaturizedData,
         step_size: int = 60,
         split_fraction: float = 0.40,
+        target_transform: str = "none",
     ) -> None:
+        """``target_transform="log1p"``: fit scalers on log1p(y) and invert
+        with expm1 at denormalization.  Monotone, so quantile outputs stay
+        quantiles; compresses the unseen-scale extrapolation range (a 3x
+        traffic query moves the target by +log 3 instead of 3x past the
+        fitted min-max — the measured failure mode of the bounded net on
+        the reference's unseen-scale axis, profiles/r02_unseen_traffic.md)."""
         self.data = data
         self.step_size = step_size
+        self.target_transform = target_transform
         self.metric_names = data.metric_names
 
         # The traffic windows are the BIG tensor at 4096-endpoint scale
@@ -54,6 +62,10 @@ class EstimationDataset:
 
         # raw (denormalized) labels, needed by the baselines and error eval
         self.y_raw = y.copy()
+        if target_transform == "log1p":
+            y = np.log1p(np.maximum(y, 0.0))
+        elif target_transform != "none":
+            raise ValueError(f"unknown target_transform {target_transform!r}")
 
         self.x_scaler = MinMaxScaler().fit(X, self.split)
         if self.x_scaler.scale != 0.0:
@@ -98,7 +110,10 @@ class EstimationDataset:
         return out
 
     def denormalize_metric(self, values: np.ndarray, idx: int) -> np.ndarray:
-        return self.y_scalers[idx].inverse_transform(values)
+        out = self.y_scalers[idx].inverse_transform(values)
+        if self.target_transform == "log1p":
+            out = np.expm1(out)
+        return out
 
     # ------------------------------------------------------------- checkpoint
     def scaler_state(self) -> dict:
@@ -108,4 +123,5 @@ class EstimationDataset:
             "metric_names": self.metric_names,
             "step_size": self.step_size,
             "split": self.split,
+            "target_transform": self.target_transform,
         }

@@ -64,7 +64,9 @@ class Trainer:
         self.world_size = dist_ctx.world_size if dist_ctx is not None else 1
 
         self.dataset = EstimationDataset(
-            data, step_size=self.cfg.data.step_size, split_fraction=self.cfg.data.split
+            data, step_size=self.cfg.data.step_size,
+            split_fraction=self.cfg.data.split,
+            target_transform=self.cfg.data.target_transform,
         )
         if model is None:
             spec = build_model_spec(data)

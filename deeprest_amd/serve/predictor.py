@@ -44,7 +44,9 @@ class Predictor:
         graph_batch: Optional[int] = None,
         graph_batches: Optional[Sequence[int]] = None,
         use_graph: bool = True,
+        target_transform: str = "none",
     ) -> None:
+        self.target_transform = target_transform
         self.device = device or torch.device(
             "cuda" if torch.cuda.is_available() else "cpu"
         )
@@ -70,6 +72,7 @@ class Predictor:
         y_scalers = [MinMaxScaler.from_state_dict(s) for s in sc["y_scalers"]]
         fs = (FeatureSpace.from_state_dict(state["feature_space"])
               if state.get("feature_space") else None)
+        kw.setdefault("target_transform", sc.get("target_transform", "none"))
         return Predictor(model, x_scaler, y_scalers, sc["metric_names"],
                          feature_space=fs, device=device, **kw)
 
@@ -162,9 +165,10 @@ class Predictor:
         out = np.sort(out, axis=-1)
         preds = {}
         for m, name in enumerate(self.metric_names):
-            preds[name] = np.maximum(
-                self.y_scalers[m].inverse_transform(out[:, :, m, :]), 1e-6
-            )
+            v = self.y_scalers[m].inverse_transform(out[:, :, m, :])
+            if self.target_transform == "log1p":
+                v = np.expm1(v)
+            preds[name] = np.maximum(v, 1e-6)
         return preds
 
     def predict_what_if(self, synthesizer, traffic_plan, step_size: int,

@@ -36,6 +36,8 @@ def main():
     ap.add_argument("--out", default="results_scenarios.pkl")
     ap.add_argument("--resources", type=int, choices=[3, 5], default=5,
                     help="3 = round-1 default, 5 = the reference's full set")
+    ap.add_argument("--log-targets", action="store_true",
+                    help="log1p target transform (unseen-scale extrapolation)")
     args = ap.parse_args()
 
     from deeprest_amd.data.synthetic import ALL_RESOURCES, DEFAULT_RESOURCES
@@ -47,7 +49,8 @@ def main():
         resource_noise=0.03, seed=args.seed))
 
     cfg = EngineConfig(
-        data=DataConfig(step_size=60, split=0.40),
+        data=DataConfig(step_size=60, split=0.40,
+                        target_transform="log1p" if args.log_targets else "none"),
         train=TrainConfig(epochs=args.epochs, batch_size=32, lr=1e-3,
                           run_baselines=True, log_every=0),
         model=DeepRestNetConfig(dropout=0.1),
