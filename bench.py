@@ -98,6 +98,15 @@ def build_train_tensors(args):
     # clone: X_train is a view — saving/keeping it would pin the whole
     # (train+test) storage
     X, y = ds.X_train.clone(), ds.y_train.clone()
+    pad = (-spec.num_paths) % 64
+    if pad and os.environ.get("DEEPREST_PAD64", "1") == "1":
+        # 128-byte row alignment for the in_proj GEMMs: an odd lda
+        # (12615 call paths) makes every A-row read cross cache lines
+        # unaligned; zero columns are inert under the global min-max
+        # (traffic min is 0) and the padded K is what the shipped
+        # TunableOp tables are keyed on
+        X = torch.nn.functional.pad(X, (0, pad))
+        spec.num_paths += pad
     del ds, data, app  # free the multi-GB host transients
     return X, y, spec
 
