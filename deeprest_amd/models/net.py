@@ -132,10 +132,12 @@ class DeepRestNetConfig:
     bidirectional: bool = True
     prop_rounds: int = 2                  # call-graph propagation rounds
     fp8_inference: bool = False           # fp8 MFMA GRU decode at eval time
-    linear_bias: bool = True              # biases on encoder/x_proj Linears
-                                          # (False: LLaMA-style biasless —
-                                          # kills ~1.1 ms/step of bias-grad
-                                          # reductions; A/B'd for accuracy)
+    linear_bias: bool = False             # biases on encoder/x_proj Linears:
+                                          # biasless default (LayerNorms and
+                                          # metric_bias absorb shifts) saves
+                                          # ~1.1 ms/step of bias-grad
+                                          # reductions at accuracy parity
+                                          # (profiles/r02_perf_notes.md)
 
     def to_dict(self) -> dict:
         return {
@@ -454,7 +456,10 @@ class DeepRestNet(nn.Module):
     @staticmethod
     def from_full_state(state: dict) -> "DeepRestNet":
         spec = ModelSpec(**state["spec"])
-        cfg = DeepRestNetConfig(**state["config"])
+        config = dict(state["config"])
+        # checkpoints from before the biasless default carried biases
+        config.setdefault("linear_bias", True)
+        cfg = DeepRestNetConfig(**config)
         model = DeepRestNet(spec, cfg)
         model.load_state_dict(state["state_dict"])
         return model
