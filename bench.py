@@ -274,7 +274,7 @@ def main():
         try:
             for o in offs:
                 g = GraphedTrainStep(
-                    model, opt, lambda out, t: model.loss(out.float(), t),
+                    model, opt, lambda out, t: model.loss(out, t),
                     X[o : o + B], y[o : o + B],
                     autocast_dtype=autocast_dtype if use_autocast else None,
                     warmup=2, static_inputs=True, pool=pool)

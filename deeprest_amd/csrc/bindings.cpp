@@ -18,12 +18,12 @@ void dr_layernorm_bwd(const void* dy, const void* x, const float* w,
                       const float* mean, const float* rstd, void* dx,
                       float* dwdb_part, int n_blocks, int64_t n_rows, int D,
                       int is_bf16, hipStream_t stream);
-void dr_pinball_fwd(const float* out, const float* labels, const float* quantiles,
-                    int Q, int64_t N, float inv_count, float* loss,
+void dr_pinball_fwd(const void* out, const float* labels, const float* quantiles,
+                    int Q, int64_t N, float inv_count, float* loss, int is_bf16,
                     hipStream_t stream);
-void dr_pinball_bwd(const float* out, const float* labels, const float* quantiles,
+void dr_pinball_bwd(const void* out, const float* labels, const float* quantiles,
                     int Q, int64_t N, const float* grad_loss, float inv_n,
-                    float* dout, hipStream_t stream);
+                    void* dout, int is_bf16, hipStream_t stream);
 void dr_fused_adam(const int64_t* meta, int nt, int64_t total, float lr, float beta1,
                    float beta2, float eps, float weight_decay, int step,
                    hipStream_t stream);
@@ -108,18 +108,18 @@ std::vector<at::Tensor> layer_norm_backward(at::Tensor dy, at::Tensor x, at::Ten
 // ---------------------------------------------------------------- pinball
 at::Tensor pinball_forward(at::Tensor outputs, at::Tensor labels, at::Tensor q) {
   const at::cuda::CUDAGuard guard(outputs.device());
-  TORCH_CHECK(outputs.scalar_type() == at::kFloat && labels.scalar_type() == at::kFloat,
-              "pinball expects f32");
+  check_dtype(outputs, "pinball outputs");
+  TORCH_CHECK(labels.scalar_type() == at::kFloat, "pinball labels must be f32");
   TORCH_CHECK(outputs.is_contiguous() && labels.is_contiguous());
   int Q = (int)outputs.size(-1);
   TORCH_CHECK(Q <= 8, "at most 8 quantiles");
   int64_t N = outputs.numel() / Q;
   TORCH_CHECK(labels.numel() == N);
-  auto loss = at::zeros({}, outputs.options());
+  auto loss = at::zeros({}, outputs.options().dtype(at::kFloat));
   float inv_count = N > 0 ? 1.0f / (float)N : 0.f;
-  dr_pinball_fwd(outputs.data_ptr<float>(), labels.data_ptr<float>(),
+  dr_pinball_fwd(outputs.data_ptr(), labels.data_ptr<float>(),
                  q.data_ptr<float>(), Q, N, inv_count, loss.data_ptr<float>(),
-                 cur_stream());
+                 is_bf16(outputs), cur_stream());
   return loss;
 }
 
@@ -135,9 +135,9 @@ at::Tensor pinball_backward(at::Tensor grad, at::Tensor outputs, at::Tensor labe
               "pinball grad must be a scalar f32 CUDA tensor");
   float inv_n = N > 0 ? 1.f / (float)N : 0.f;
   auto gc = grad.contiguous();
-  dr_pinball_bwd(outputs.data_ptr<float>(), labels.data_ptr<float>(),
+  dr_pinball_bwd(outputs.data_ptr(), labels.data_ptr<float>(),
                  q.data_ptr<float>(), Q, N, gc.data_ptr<float>(), inv_n,
-                 dout.data_ptr<float>(), cur_stream());
+                 dout.data_ptr(), is_bf16(outputs), cur_stream());
   return dout;
 }
 

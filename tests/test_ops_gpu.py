@@ -507,3 +507,25 @@ def test_gru_offspec_shape_uses_composed_path(dev):
         torch.testing.assert_close(out.float(), ref.float(), rtol=1e-4, atol=1e-4)
         out.sum().backward()   # autograd must work on the degraded path
         assert args[0].grad is not None and torch.isfinite(args[0].grad).all()
+
+
+def test_pinball_bf16_outputs(dev):
+    """bf16 predictions feed the pinball kernel directly; loss and grads
+    match the fp32 oracle at bf16 tolerance."""
+    from deeprest_amd.ops import pinball_loss
+    from deeprest_amd.ops.pinball import reference_pinball_loss
+
+    torch.manual_seed(6)
+    B, T, M, Q = 4, 9, 5, 3
+    out16 = (torch.randn(B, T, M, Q, device=dev) * 0.5).to(torch.bfloat16)
+    labels = torch.randn(B, T, M, device=dev)
+    a = out16.detach().clone().requires_grad_(True)
+    b = out16.float().detach().clone().requires_grad_(True)
+    la = pinball_loss(a, labels)
+    lb = reference_pinball_loss(b, labels, (0.05, 0.50, 0.95))
+    assert la.dtype == torch.float32
+    torch.testing.assert_close(la, lb.float(), rtol=2e-2, atol=1e-3)
+    la.backward()
+    lb.backward()
+    assert a.grad.dtype == torch.bfloat16
+    torch.testing.assert_close(a.grad.float(), b.grad, rtol=2e-2, atol=2e-3)
