@@ -216,9 +216,8 @@ std::vector<at::Tensor> gru_seq_forward(at::Tensor xg, at::Tensor w_hh,
   TORCH_CHECK(xg.numel() < (1LL << 31),
               "x_gates too large for 32-bit staging offsets");
   auto h_all = at::empty({B, TT, C, H}, xg.options());
-  // saves carry u8-quantized r|z (kernel-side packing; see gru.hip)
-  auto saves = save ? at::empty({B, TT, C, 2 * H}, xg.options().dtype(at::kByte))
-                    : at::empty({0}, xg.options().dtype(at::kByte));
+  auto saves = save ? at::empty({B, TT, C, 2 * H}, xg.options())
+                    : at::empty({0}, xg.options());
   // the GEMM streams a bf16 weight image from L2 regardless of T
   auto w_gemm = w_hh.scalar_type() == at::kBFloat16
                     ? w_hh
@@ -240,10 +239,9 @@ std::vector<at::Tensor> gru_seq_backward_kernel(at::Tensor grad_h, at::Tensor w_
   int B = (int)grad_h.size(0), TT = (int)grad_h.size(1);
   int C = (int)grad_h.size(2), H = (int)grad_h.size(3);
   TORCH_CHECK(H == 128);
-  TORCH_CHECK(saves.numel() == (int64_t)B * TT * C * 2 * H &&
-                  saves.scalar_type() == at::kByte,
-              "gru backward: saves tensor missing or wrong size/dtype "
-              "(forward must run with save=true)");
+  TORCH_CHECK(saves.numel() == (int64_t)B * TT * C * 2 * H,
+              "gru backward: saves tensor missing or wrong size (forward must "
+              "run with save=true)");
   TORCH_CHECK(w_img.scalar_type() == at::kBFloat16 &&
                   w_img.sizes() == at::IntArrayRef({H, 3 * H}) &&
                   w_img.is_contiguous(),

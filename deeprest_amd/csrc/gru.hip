@@ -178,10 +178,7 @@ __global__ __launch_bounds__(THREADS * NSUB) void gru_fwd_kernel(
     const float* __restrict__ b_hh,  // (3H,)
     const T* __restrict__ h0,      // (B, C, H)
     T* __restrict__ h_all,         // (B, TT, C, H)
-    uint8_t* __restrict__ saves,   // (B, TT, C, 2H) pi, u8-quantized r|z
-                                   // (sigmoid outputs in (0,1): 1/255
-                                   // absolute error ~ bf16's relative
-                                   // error; halves the saves HBM traffic)
+    T* __restrict__ saves,         // (B, TT, C, 4H) pi layout (SAVE only)
     int B, int TT, int C, int reverse) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   constexpr int ELT = FP8 ? 1 : 2;             // GEMM-tile element bytes
@@ -447,18 +444,10 @@ __global__ __launch_bounds__(THREADS * NSUB) void gru_fwd_kernel(
       }
       if (SAVE && live[i]) {
         // only r and z are saved (2H); the backward recomputes n and hh_n
-        // (an extra H x H GEMM there is cheaper than 2H of HBM both ways).
-        // u8 linear quantization of the (0,1) sigmoid outputs: one packed
-        // 8-byte store per gate per row
-        uint8_t* sv = saves + (sv_bc[i] + bc_toff) * (2 * H) + c_col * 8;
-        uint64_t pr = 0, pz = 0;
-#pragma unroll
-        for (int nt = 0; nt < 8; ++nt) {
-          pr |= (uint64_t)(uint8_t)(fr[nt] * 255.f + 0.5f) << (8 * nt);
-          pz |= (uint64_t)(uint8_t)(fz[nt] * 255.f + 0.5f) << (8 * nt);
-        }
-        *reinterpret_cast<uint64_t*>(sv) = pr;
-        *reinterpret_cast<uint64_t*>(sv + H) = pz;
+        // (an extra H x H GEMM there is cheaper than 2H of HBM both ways)
+        T* sv = saves + (sv_bc[i] + bc_toff) * (2 * H) + c_col * 8;
+        st8(sv, fr);
+        st8(sv + H, fz);
       }
       // fence: stop the scheduler interleaving all 4 rows' live ranges
       __builtin_amdgcn_sched_barrier(0);
@@ -514,7 +503,7 @@ __global__ __launch_bounds__(NSUB * THREADS) void gru_bwd_kernel(
     const float* __restrict__ b_hh, // (3H,)
     const T* __restrict__ h0,       // (B, C, H)
     const T* __restrict__ h_all,    // (B, TT, C, H)
-    const uint8_t* __restrict__ saves,  // (B, TT, C, 2H) pi, u8 r|z
+    const T* __restrict__ saves,    // (B, TT, C, 2H) pi layout: r|z
     T* __restrict__ dpre,           // (B, TT, C, 4H) pi: dr|dz|dn|d_hhn
     float* __restrict__ dh0,        // (B, C, H)
     int B, int TT, int C, int reverse) {
@@ -640,15 +629,10 @@ __global__ __launch_bounds__(NSUB * THREADS) void gru_bwd_kernel(
     for (int i = 0; i < 4; ++i) {
       if (!live[i]) continue;
       const int64_t bc = bc0[i] + bc_toff;
-      const uint8_t* sv_p = saves + bc * (2 * H) + c_col * 8;
-      const uint64_t pr = *reinterpret_cast<const uint64_t*>(sv_p);
-      const uint64_t pz = *reinterpret_cast<const uint64_t*>(sv_p + H);
+      const T* sv_p = saves + bc * (2 * H) + c_col * 8;
       float rp[8], zp[8];
-#pragma unroll
-      for (int nt = 0; nt < 8; ++nt) {
-        rp[nt] = (float)((pr >> (8 * nt)) & 0xff) * (1.f / 255.f);
-        zp[nt] = (float)((pz >> (8 * nt)) & 0xff) * (1.f / 255.f);
-      }
+      ld8(sv_p, rp);
+      ld8(sv_p + H, zp);
       float fdr[8], fdz[8], fdn[8], fdh[8];
       // h_prev sits at bc + dirC rows (tprev = t+dir); h0 at the boundary.
       // pointer select -> ONE load (the OOB address is never dereferenced)
@@ -1031,13 +1015,13 @@ static void gru_fwd_launch_t(const void* xg, const void* gamma, const void* beta
       hipLaunchKernelGGL((gru_fwd_kernel<T, true, false, 2>), dim3(tiles128),
                          dim3(2 * THREADS), LDS_FWD_8W, stream, (const T*)xg,
                          (const T*)gamma, (const T*)beta, (const uint16_t*)w_gemm,
-                         b_hh, (const T*)h0, (T*)h_all, (uint8_t*)saves, B, TT, C,
+                         b_hh, (const T*)h0, (T*)h_all, (T*)saves, B, TT, C,
                          reverse);
     else
       hipLaunchKernelGGL((gru_fwd_kernel<T, false, false, 2>), dim3(tiles128),
                          dim3(2 * THREADS), LDS_FWD_8W, stream, (const T*)xg,
                          (const T*)gamma, (const T*)beta, (const uint16_t*)w_gemm,
-                         b_hh, (const T*)h0, (T*)h_all, (uint8_t*)saves, B, TT, C,
+                         b_hh, (const T*)h0, (T*)h_all, (T*)saves, B, TT, C,
                          reverse);
     return;
   }
@@ -1045,17 +1029,17 @@ static void gru_fwd_launch_t(const void* xg, const void* gamma, const void* beta
     hipLaunchKernelGGL((gru_fwd_kernel<T, false, true>), dim3(grid), dim3(THREADS),
                        LDS_FWD_FP8, stream, (const T*)xg, (const T*)gamma,
                        (const T*)beta, (const uint16_t*)w_gemm, b_hh, (const T*)h0,
-                       (T*)h_all, (uint8_t*)saves, B, TT, C, reverse);
+                       (T*)h_all, (T*)saves, B, TT, C, reverse);
   else if (save)
     hipLaunchKernelGGL((gru_fwd_kernel<T, true>), dim3(grid), dim3(THREADS),
                        LDS_FWD_TOTAL, stream, (const T*)xg, (const T*)gamma,
                        (const T*)beta, (const uint16_t*)w_gemm, b_hh, (const T*)h0,
-                       (T*)h_all, (uint8_t*)saves, B, TT, C, reverse);
+                       (T*)h_all, (T*)saves, B, TT, C, reverse);
   else
     hipLaunchKernelGGL((gru_fwd_kernel<T, false>), dim3(grid), dim3(THREADS),
                        LDS_FWD_TOTAL, stream, (const T*)xg, (const T*)gamma,
                        (const T*)beta, (const uint16_t*)w_gemm, b_hh, (const T*)h0,
-                       (T*)h_all, (uint8_t*)saves, B, TT, C, reverse);
+                       (T*)h_all, (T*)saves, B, TT, C, reverse);
 }
 
 template <typename T>
@@ -1081,7 +1065,7 @@ static void gru_bwd_launch_t(const void* grad_h, const void* w_img,
                        LDS_WB, stream, (const T*)grad_h, (const uint16_t*)w_img,
                        (const uint16_t*)w_fwd, (const T*)xg, (const T*)gamma,
                        (const T*)beta, b_hh, (const T*)h0, (const T*)h_all,
-                       (const uint8_t*)saves, (T*)dpre, dh0, B, TT, C, reverse);
+                       (const T*)saves, (T*)dpre, dh0, B, TT, C, reverse);
     return;
   }
   int grid = (int)((R + ROWS - 1) / ROWS);
@@ -1089,7 +1073,7 @@ static void gru_bwd_launch_t(const void* grad_h, const void* w_img,
                      (const T*)grad_h, (const uint16_t*)w_img,
                      (const uint16_t*)w_fwd, (const T*)xg, (const T*)gamma,
                      (const T*)beta, b_hh, (const T*)h0, (const T*)h_all,
-                     (const uint8_t*)saves, (T*)dpre, dh0, B, TT, C, reverse);
+                     (const T*)saves, (T*)dpre, dh0, B, TT, C, reverse);
 }
 
 template <typename T>
