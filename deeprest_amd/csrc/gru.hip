@@ -807,69 +807,6 @@ __global__ void gru_dxg_kernel(const T* __restrict__ dpre,     // (BT, C, 4H) pi
   }
 }
 
-// dgamma/dbeta4 with block-tiled reads: a block owns 16 components and a
-// bt range; per bt it streams a CONTIGUOUS 16 KB slab of dpre (the
-// per-c-walk version strode 64 KB between loads - page-per-load).  Thread
-// t owns (c_local = t/64 + 4k, gate=(t%64)/16, colgroup=t%16): wave w
-// streams row c_w fully coalesced; accumulators live in registers and
-// flush once per block with f32 atomics.
-template <typename T>
-__global__ __launch_bounds__(256) void gru_dgamma_tiled_kernel(
-    const T* __restrict__ dpre,   // (BT, C, 4H) pi
-    const T* __restrict__ xg_pi,  // (BT, 3H) pi
-    float* __restrict__ dgamma,   // (C, 3H) zeroed
-    float* __restrict__ dbeta4,   // (C, 4H) zeroed
-    int64_t BT, int C) {
-  const int tid = threadIdx.x;
-  const int rem = tid & 63;
-  const int g = rem >> 4;
-  const int cc = rem & 15;
-  const bool has_x = g < 3;
-  const int pi0 = g * H + cc * 8;
-  const int jn = g * H + cc;
-  const int64_t bt_lo = BT * blockIdx.y / gridDim.y;
-  const int64_t bt_hi = BT * (blockIdx.y + 1) / gridDim.y;
-
-  int cs[4];
-  int ncs = 0;
-  for (int k = 0; k < 4; ++k) {
-    int c = (int)blockIdx.x * 16 + (tid >> 6) + 4 * k;
-    if (c < C) cs[ncs++] = c;
-  }
-  float accg[4][8], accb[4][8];
-#pragma unroll
-  for (int k = 0; k < 4; ++k)
-#pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      accg[k][e] = 0.f;
-      accb[k][e] = 0.f;
-    }
-  for (int64_t bt = bt_lo; bt < bt_hi; ++bt) {
-    float xv[8];
-    if (has_x) ld8(xg_pi + bt * G3H + pi0, xv);
-    for (int k = 0; k < ncs; ++k) {
-      float d[8];
-      ld8(dpre + (bt * C + cs[k]) * G4H + pi0, d);
-#pragma unroll
-      for (int e = 0; e < 8; ++e) accb[k][e] += d[e];
-      if (has_x) {
-#pragma unroll
-        for (int e = 0; e < 8; ++e) accg[k][e] += d[e] * xv[e];
-      }
-    }
-  }
-  for (int k = 0; k < ncs; ++k) {
-    float* b_out = dbeta4 + (int64_t)cs[k] * G4H + jn;
-#pragma unroll
-    for (int e = 0; e < 8; ++e) atomicAdd(b_out + e * 16, accb[k][e]);
-    if (has_x) {
-      float* g_out = dgamma + (int64_t)cs[k] * G3H + jn;
-#pragma unroll
-      for (int e = 0; e < 8; ++e) atomicAdd(g_out + e * 16, accg[k][e]);
-    }
-  }
-}
-
 // dgamma[c, j<3H] = sum_bt dpre[bt, c, pi(j)] * xg[bt, j]
 // dbeta4[c, j<4H] = sum_bt dpre[bt, c, pi(j)]   (4th slice -> db_hh_n)
 // grid.y slices BT; f32 atomics finalize.  Natural-order outputs.
@@ -1165,12 +1102,11 @@ static void gru_reduce_launch_t(const void* dpre, const void* gamma, const void*
                        (const T*)dpre, (const T*)gamma_pi, (T*)dxg, BT, C);
   }
   {
-    // block-tiled variant: contiguous 16 KB slab reads per bt per block
-    int gx = (C + 15) / 16;
-    int gy = (int)std::min<int64_t>(BT, std::max(1, 1024 / gx));
-    hipLaunchKernelGGL((gru_dgamma_tiled_kernel<T>), dim3(gx, gy), dim3(256),
-                       0, stream, (const T*)dpre, (const T*)xg_pi, dgamma,
-                       dbeta, BT, C);
+    int n_threads = C * 64;
+    int gx = (n_threads + 255) / 256;
+    int gy = 128;  // BT slices (parallelism for small C; atomics stay cheap)
+    hipLaunchKernelGGL((gru_dgamma_kernel<T>), dim3(gx, gy), dim3(256), 0, stream,
+                       (const T*)dpre, (const T*)xg_pi, dgamma, dbeta, BT, C);
   }
 }
 
