@@ -118,7 +118,17 @@ def shared_train_tensors(args, dist_ctx):
     if dist_ctx is None:
         return build_train_tensors(args)
     key = f"{args.batch}-{args.endpoints}-{args.components}-{args.seq_len}"
-    base = "/dev/shm" if os.path.isdir("/dev/shm") else tempfile.gettempdir()
+    # estimated file size: X f32 (need, T, ~P) + slack; containers often
+    # mount a small /dev/shm — fall back to TMPDIR rather than ENOSPC the
+    # driver's 8-rank scale run
+    need_windows = int(args.batch / 0.8) + args.seq_len + 64
+    est_bytes = int(need_windows * args.seq_len
+                    * (args.endpoints * 55 + 64) * 4 * 1.3)
+    base = tempfile.gettempdir()
+    if os.path.isdir("/dev/shm"):
+        st = os.statvfs("/dev/shm")
+        if st.f_bavail * st.f_frsize > est_bytes:
+            base = "/dev/shm"
     path = os.path.join(
         base, f"deeprest_bench_{hashlib.md5(key.encode()).hexdigest()[:12]}.pt")
     if dist_ctx.rank == 0:

@@ -276,3 +276,61 @@ def test_results_entry_reanchors_monotone_metrics():
                               train_peak=train_peak, reanchor=True,
                               anchor_value=100.0)
     assert re2["scale_ours"][0] == (140.0 - 130.0 + 100.0) / train_peak
+
+
+def test_old_checkpoint_with_biases_loads(tmp_path):
+    """Checkpoints from before the biasless default carried biases and no
+    linear_bias key — the from_full_state shim must restore them."""
+    from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+    from deeprest_amd.models.net import (DeepRestNet, DeepRestNetConfig,
+                                         build_model_spec)
+
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=3, n_components=4, windows_per_day=40, n_days=1, seed=3))
+    data = app.generate_featurized()
+    spec = build_model_spec(data)
+    old = DeepRestNet(spec, DeepRestNetConfig(
+        d_model=32, n_heads=2, n_layers=1, d_ff=64, hidden=16, comp_dim=8,
+        linear_bias=True))
+    state = old.full_state()
+    del state["config"]["linear_bias"]          # emulate the old format
+    loaded = DeepRestNet.from_full_state(state)
+    assert loaded.cfg.linear_bias is True
+    assert loaded.in_proj.bias is not None
+    x = torch.rand(2, 10, spec.num_paths)
+    old.eval(), loaded.eval()          # dropout off for the equality check
+    torch.testing.assert_close(loaded(x), old(x))
+
+
+def test_predictor_restores_target_transform(trained, tmp_path):
+    """log1p-trained checkpoints serve with expm1 denormalization."""
+    import numpy as np
+
+    from deeprest_amd.engine.config import EngineConfig, DataConfig, TrainConfig
+    from deeprest_amd.engine.trainer import Trainer
+    from deeprest_amd.models.net import DeepRestNetConfig
+    from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=3, n_components=4, windows_per_day=60, n_days=2, seed=3))
+    data = app.generate_featurized()
+    cfg = EngineConfig(
+        data=DataConfig(step_size=20, split=0.4, target_transform="log1p"),
+        train=TrainConfig(epochs=1, batch_size=16, run_baselines=False,
+                          log_every=0),
+        model=DeepRestNetConfig(d_model=32, n_heads=2, n_layers=1, d_ff=64,
+                                hidden=16, comp_dim=8, dropout=0.0))
+    tr = Trainer(data, cfg, device=torch.device("cpu"))
+    tr.train()
+    ckpt = str(tmp_path / "log1p.pt")
+    tr.save(ckpt, 1)
+    pred = Predictor.from_checkpoint(ckpt, device=torch.device("cpu"))
+    assert pred.target_transform == "log1p"
+    from deeprest_amd.data.windows import sliding_window
+
+    w = sliding_window(np.asarray(data.traffic, dtype=np.float64), 20)[:4]
+    out = pred.predict(w)
+    # denormalized predictions land in raw-resource magnitude, not log space
+    first = out[data.metric_names[0]]
+    raw_med = float(np.median(data.resources[data.metric_names[0]]))
+    assert float(np.median(first)) > np.log1p(raw_med)  # clearly not log-space
