@@ -230,16 +230,10 @@ __global__ __launch_bounds__(A_THREADS) void mha_bwd_kernel(
     T* __restrict__ dv,
     int T_len, int D, float scale) {
   // LDS: q / do tiles (32 x D bf16, padded rows) + per-wave P/ds scratch +
-  // per-q-tile lse/Drow + the block's 64-key K tile (read three times per
-  // q-tile: kfrag A-operand, dQ B-operand — global scalar re-reads were the
-  // kernel's hot spot) + a dq accumulator tile (when one key-block covers
-  // all of T, cross-wave dq accumulation stays in LDS and dq is a plain
-  // store, not global f32 atomics)
+  // per-q-tile lse/Drow
   __shared__ __attribute__((aligned(16))) char q_lds[32 * K_STRIDE];
   __shared__ __attribute__((aligned(16))) char do_lds[32 * K_STRIDE];
-  __shared__ __attribute__((aligned(16))) char k_tile[64 * K_STRIDE];
   __shared__ __attribute__((aligned(16))) char p_sc[A_WAVES * 16 * (32 * 2 + 16)];
-  __shared__ float dq_s[32 * DMAX];
   __shared__ float lse_s[32];
   __shared__ float drow_s[32];
 
@@ -260,39 +254,20 @@ __global__ __launch_bounds__(A_THREADS) void mha_bwd_kernel(
   const int n_kt_d = (D + 31) / 32;   // K-tiles over the D axis
   const int n_d_tiles = (D + 15) / 16;
 
-  // ---- stage the block's 64-key K tile in LDS (zero-padded) ----
-  {
-    const int kbase = blockIdx.x * 64;
-    for (int id = tid; id < 64 * (DMAX / 8); id += A_THREADS) {
-      int row = id / (DMAX / 8);
-      int blk = id % (DMAX / 8);
-      uint16_t* dst = reinterpret_cast<uint16_t*>(k_tile + row * K_STRIDE + blk * 16);
-      int krow = kbase + row;
-#pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        int d = blk * 8 + e;
-        bool ok = krow < T_len && d < D;
-        dst[e] = f2bf(ok ? ldf(kb + (int64_t)krow * D + d) : 0.f);
-      }
-    }
-  }
-  __syncthreads();
-
   // ---- K and V fragments for this wave's 16 keys (A operands, row = key) ----
   bf16x8 kfrag[2], vfrag[2];
 #pragma unroll
   for (int kt = 0; kt < 2; ++kt) {
-    int krow_l = wv * 16 + (lane & 15);          // key row within the tile
-    kfrag[kt] = *reinterpret_cast<const bf16x8*>(
-        k_tile + krow_l * K_STRIDE + (kt * 32 + rgrp * 8) * 2);
-    uint16_t tv[8];
+    uint16_t tk[8], tv[8];
     int krow = key0 + (lane & 15);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int d = kt * 32 + rgrp * 8 + j;
       bool ok = krow < T_len && d < D;
+      tk[j] = f2bf(ok ? ldf(kb + (int64_t)krow * D + d) : 0.f);
       tv[j] = f2bf(ok ? ldf(vb + (int64_t)krow * D + d) : 0.f);
     }
+    kfrag[kt] = *reinterpret_cast<bf16x8*>(tk);
     vfrag[kt] = *reinterpret_cast<bf16x8*>(tv);
   }
 
@@ -325,32 +300,18 @@ __global__ __launch_bounds__(A_THREADS) void mha_bwd_kernel(
         dstd[e] = f2bf(ok ? ldf(dob + (int64_t)qrow * D + d) : 0.f);
       }
     }
-    // Drow[r] = sum_d do[r,d] * o[r,d]; 8 threads per row partial-sum into
-    // LDS (the 1-thread-per-row version serialized 2*D scalar loads while
-    // 224 lanes idled), lse per row
+    // Drow[r] = sum_d do[r,d] * o[r,d]; lse per row (one wave's worth of rows)
     if (tid < 32) {
-      drow_s[tid] = 0.f;
       int qrow = q0 + tid;
-      lse_s[tid] = qrow < T_len ? lse[bh * T_len + qrow] : 0.f;
-    }
-    // zero the dq accumulator tile for this q-tile (single-key-block path)
-    if (gridDim.x == 1) {
-      for (int id = tid; id < 32 * DMAX; id += A_THREADS) dq_s[id] = 0.f;
-    }
-    __syncthreads();
-    {
-      int row = tid >> 3;              // 32 rows x 8 partials
-      int part = tid & 7;
-      if (row < 32) {
-        int qrow = q0 + row;
-        float s = 0.f;
-        if (qrow < T_len) {
-          int d0 = part * (D / 8);
-          for (int d = d0; d < d0 + D / 8; ++d)
-            s += ldf(dob + (int64_t)qrow * D + d) * ldf(ob + (int64_t)qrow * D + d);
-        }
-        if (s != 0.f) atomicAdd(&drow_s[row], s);
+      float s = 0.f;
+      if (qrow < T_len) {
+        for (int d = 0; d < D; ++d)
+          s += ldf(dob + (int64_t)qrow * D + d) * ldf(ob + (int64_t)qrow * D + d);
+        lse_s[tid] = lse[bh * T_len + qrow];
+      } else {
+        lse_s[tid] = 0.f;
       }
+      drow_s[tid] = s;
     }
     __syncthreads();
 
@@ -491,42 +452,21 @@ __global__ __launch_bounds__(A_THREADS) void mha_bwd_kernel(
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           int keyr = rgrp * 8 + j;
-          bk[j] = (keyr < 16 && d < D)
-                      ? *reinterpret_cast<const uint16_t*>(
-                            k_tile + (wv * 16 + keyr) * K_STRIDE + d * 2)
-                      : (uint16_t)0;
+          int krow = key0 + keyr;
+          bool ok = keyr < 16 && krow < T_len && d < D;
+          bk[j] = f2bf(ok ? ldf(kb + (int64_t)krow * D + d) : 0.f);
         }
         bf16x8 bfrag = *reinterpret_cast<bf16x8*>(bk);
         f32x4 a = {0.f, 0.f, 0.f, 0.f};
         a = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, a, 0, 0, 0);
-        if (gridDim.x == 1) {
-          // sole key-block: accumulate across waves in LDS, store below
+        // scatter-add into dq (fp32)
 #pragma unroll
-          for (int i = 0; i < 4; ++i) {
-            int qrow_t = qh * 16 + rgrp * 4 + i;
-            int d2 = nt * 16 + c_col;
-            if (d2 < D) atomicAdd(&dq_s[qrow_t * DMAX + d2], a[i]);
-          }
-        } else {
-          // multiple key-blocks contend on dq: global f32 atomics
-#pragma unroll
-          for (int i = 0; i < 4; ++i) {
-            int qrow = q0 + qh * 16 + rgrp * 4 + i;
-            int d2 = nt * 16 + c_col;
-            if (qrow < T_len && d2 < D)
-              atomicAdd(dq + (bh * T_len + qrow) * D + d2, a[i]);
-          }
+        for (int i = 0; i < 4; ++i) {
+          int qrow = q0 + qh * 16 + rgrp * 4 + i;
+          int d2 = nt * 16 + c_col;
+          if (qrow < T_len && d2 < D)
+            atomicAdd(dq + (bh * T_len + qrow) * D + d2, a[i]);
         }
-      }
-    }
-    if (gridDim.x == 1) {
-      __syncthreads();   // all waves' dq_s adds done
-      for (int id = tid; id < 32 * DMAX; id += A_THREADS) {
-        int row = id / DMAX;
-        int d = id % DMAX;
-        int qrow = q0 + row;
-        if (qrow < T_len && d < D)
-          dq[(bh * T_len + qrow) * D + d] = dq_s[id];
       }
     }
   }
