@@ -52,6 +52,14 @@ class TrainConfig:
                                           # the final epoch always evaluates
     lr_schedule: str = "none"             # "none" (reference: fixed lr) or
                                           # "cosine" (decay to 5% over the run)
+    residual_base: str = "none"           # "trace-ridge": the net learns the
+                                          # RESIDUAL over a closed-form ridge
+                                          # on call-path features — the ridge
+                                          # extrapolates unseen traffic scale
+                                          # linearly, the net corrects
+                                          # in-range nonlinearity (quantiles
+                                          # are shift-equivariant, so the
+                                          # quantile semantics are unchanged)
 
 
 @dataclass

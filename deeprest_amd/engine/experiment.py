@@ -45,6 +45,9 @@ def run_experiment(
             enabled=(trainer.device.type == "cuda")):
         xb = ds.X_test[eval_idx].to(trainer.device)
         out = trainer.model(xb).float().cpu().numpy()        # (K, T, M, Q)
+    rb = trainer.ridge_apply(ds.X_test[eval_idx].numpy())
+    if rb is not None:
+        out = out + rb[..., None]
     median_q = len(trainer.model.cfg.quantiles) // 2
 
     # baselines (whole-test-range predictions, subset to eval windows)
@@ -262,6 +265,10 @@ def run_scenario_suite(
                 device_type=trainer.device.type, dtype=trainer.autocast_dtype,
                 enabled=(trainer.device.type == "cuda")):
             out = trainer.model(xb.to(trainer.device)).float().cpu().numpy()
+        rb = trainer.ridge_apply(xq_norm)
+        if rb is not None:
+            # residual head: ridge carries the unseen-scale extrapolation
+            out = out + rb[..., None]
 
         n_flat = len(qdata.traffic)
         split_flat = ds.split + step - 1   # base learning period, flat steps

@@ -90,6 +90,17 @@ class Trainer:
         )
         self.start_epoch = 0
         self._baseline_preds: Optional[Dict[str, np.ndarray]] = None
+        # ridge-residual base (train.residual_base="trace-ridge"): the net
+        # learns y_norm - ridge(x_norm); the ridge carries linear
+        # extrapolation to unseen traffic scale (the bounded net saturates
+        # outside its fitted min-max — profiles/r02_unseen_traffic.md)
+        self._ridge_w: Optional[np.ndarray] = None       # (P+1, M) f64
+        self._ridge_norm: Optional[np.ndarray] = None    # (N, T, M) f32
+        if self.cfg.train.residual_base == "trace-ridge":
+            self._fit_residual_base()
+        elif self.cfg.train.residual_base != "none":
+            raise ValueError(
+                f"unknown residual_base {self.cfg.train.residual_base!r}")
         self.autocast_dtype = (
             torch.bfloat16 if self.cfg.train.dtype == "bf16" else torch.float32
         )
@@ -99,6 +110,33 @@ class Trainer:
             autocast_dtype=self.autocast_dtype
             if self.device.type == "cuda" else None,
         )
+
+    # -------------------------------------------------------- residual base
+    def _fit_residual_base(self) -> None:
+        ds = self.dataset
+        X = ds.X.numpy()
+        N, T, P = X.shape
+        if P > 20000:
+            raise ValueError(
+                "trace-ridge residual base solves (P+1)^2 normal equations; "
+                f"P={P} is past the closed-form range (use <= ~20k paths)")
+        M = ds.y.shape[-1]
+        Xf = X[: ds.split].reshape(-1, P).astype(np.float64)
+        Xf = np.concatenate([Xf, np.ones((len(Xf), 1))], axis=1)
+        yf = ds.y[: ds.split].numpy().reshape(-1, M).astype(np.float64)
+        A = Xf.T @ Xf + 1e-3 * np.eye(P + 1)
+        self._ridge_w = np.linalg.solve(A, Xf.T @ yf)          # (P+1, M)
+        self._ridge_norm = self.ridge_apply(X).astype(np.float32)
+
+    def ridge_apply(self, x_norm: np.ndarray) -> Optional[np.ndarray]:
+        """Normalized windows (N, T, P) -> normalized-space ridge predictions
+        (N, T, M); None when no residual base is fitted."""
+        if self._ridge_w is None:
+            return None
+        N, T, P = x_norm.shape
+        flat = (x_norm.reshape(-1, P) @ self._ridge_w[:P]
+                + self._ridge_w[P])
+        return flat.reshape(N, T, -1)
 
     # ------------------------------------------------------------- baselines
     def run_baselines(self) -> Dict[str, np.ndarray]:
@@ -147,7 +185,12 @@ class Trainer:
                 pass
 
         X_train = ds.X_train.to(self.device)
-        y_train = ds.y_train.to(self.device)
+        y_train = ds.y_train
+        if self._ridge_norm is not None:
+            # train on the residual: quantiles are shift-equivariant, so
+            # residual quantiles + ridge = target quantiles
+            y_train = y_train - torch.from_numpy(self._ridge_norm[: ds.split])
+        y_train = y_train.to(self.device)
         n = X_train.shape[0]
         gen = torch.Generator().manual_seed(cfg.seed)
         # resume determinism: replay the skipped epochs' permutation draws so
@@ -236,6 +279,10 @@ class Trainer:
         ):
             out = self.model(xb)
         out = out.float()
+        if self._ridge_norm is not None:
+            ridge = torch.from_numpy(
+                self._ridge_norm[ds.split :][eval_idx]).to(self.device)
+            out = out + ridge.unsqueeze(-1)
         test_loss = float(self.model.loss(out, yb).item())
 
         outputs = np.maximum(out.cpu().numpy(), 1e-6)      # (K, T, M, Q)
@@ -263,7 +310,8 @@ class Trainer:
             scaler_state=self.dataset.scaler_state(),
             feature_space_state=self.feature_space_state,
             epoch=epoch,
-            extra={"config": self.cfg.to_dict()},
+            extra={"config": self.cfg.to_dict(),
+                   "residual_ridge": self._ridge_w},
         )
 
     def load(self, path: str) -> None:

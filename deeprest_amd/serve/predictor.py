@@ -45,8 +45,12 @@ class Predictor:
         graph_batches: Optional[Sequence[int]] = None,
         use_graph: bool = True,
         target_transform: str = "none",
+        residual_ridge: Optional[np.ndarray] = None,   # (P+1, M): the net's
+        # outputs are residuals over this ridge (train.residual_base)
     ) -> None:
         self.target_transform = target_transform
+        self.residual_ridge = (np.asarray(residual_ridge)
+                               if residual_ridge is not None else None)
         self.device = device or torch.device(
             "cuda" if torch.cuda.is_available() else "cpu"
         )
@@ -73,6 +77,8 @@ class Predictor:
         fs = (FeatureSpace.from_state_dict(state["feature_space"])
               if state.get("feature_space") else None)
         kw.setdefault("target_transform", sc.get("target_transform", "none"))
+        kw.setdefault("residual_ridge",
+                      (state.get("extra") or {}).get("residual_ridge"))
         return Predictor(model, x_scaler, y_scalers, sc["metric_names"],
                          feature_space=fs, device=device, **kw)
 
@@ -159,6 +165,12 @@ class Predictor:
         xn = self.x_scaler.transform(x)
         out = self.predict_normalized(torch.from_numpy(xn).float())
         out = out.float().cpu().numpy()            # (N, T, M, Q)
+        if self.residual_ridge is not None:
+            P = xn.shape[-1]
+            base = (xn.reshape(-1, P) @ self.residual_ridge[:P]
+                    + self.residual_ridge[P]).reshape(
+                        xn.shape[0], xn.shape[1], -1)
+            out = out + base[..., None]
         # quantile regression can emit crossed quantiles (q95 < q50) early in
         # training; serving consumers (anomaly bands, demo plots) assume a
         # monotone triple, so sort the Q axis — a no-op once calibrated

@@ -38,6 +38,8 @@ def main():
                     help="3 = round-1 default, 5 = the reference's full set")
     ap.add_argument("--log-targets", action="store_true",
                     help="log1p target transform (unseen-scale extrapolation)")
+    ap.add_argument("--residual", action="store_true",
+                    help="trace-ridge residual head (unseen-scale extrapolation)")
     args = ap.parse_args()
 
     from deeprest_amd.data.synthetic import ALL_RESOURCES, DEFAULT_RESOURCES
@@ -52,7 +54,9 @@ def main():
         data=DataConfig(step_size=60, split=0.40,
                         target_transform="log1p" if args.log_targets else "none"),
         train=TrainConfig(epochs=args.epochs, batch_size=32, lr=1e-3,
-                          run_baselines=True, log_every=0),
+                          run_baselines=True, log_every=0,
+                          residual_base="trace-ridge" if args.residual
+                          else "none"),
         model=DeepRestNetConfig(dropout=0.1),
     )
     torch.manual_seed(0)
