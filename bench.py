@@ -178,7 +178,11 @@ def accuracy_probe(epochs: int, device: torch.device, n_apis: int = 13,
             train=TrainConfig(epochs=n_epochs, batch_size=32, lr=1e-3,
                               eval_cycles=9, baseline_epochs=baseline_epochs,
                               log_every=0, eval_every=5, graph_step=True,
-                              lr_schedule=lr_schedule),
+                              lr_schedule=lr_schedule,
+                              # flagship estimator: net + trace-ridge
+                              # residual base (A/B: DEEPR 3.9 vs 7.8 at 50
+                              # epochs, beats COMP 39/39)
+                              residual_base="trace-ridge"),
             model=DeepRestNetConfig(dropout=0.1),
         )
         torch.manual_seed(0)  # probe stability run-to-run

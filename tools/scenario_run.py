@@ -38,8 +38,10 @@ def main():
                     help="3 = round-1 default, 5 = the reference's full set")
     ap.add_argument("--log-targets", action="store_true",
                     help="log1p target transform (unseen-scale extrapolation)")
-    ap.add_argument("--residual", action="store_true",
-                    help="trace-ridge residual head (unseen-scale extrapolation)")
+    ap.add_argument("--no-residual", dest="residual", action="store_false",
+                    help="disable the trace-ridge residual head (the "
+                         "default head; carries unseen-scale extrapolation)")
+    ap.set_defaults(residual=True)
     args = ap.parse_args()
 
     from deeprest_amd.data.synthetic import ALL_RESOURCES, DEFAULT_RESOURCES
