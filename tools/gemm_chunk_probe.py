@@ -1,13 +1,23 @@
-import sys
+"""Chunked-GEMM probe for the in_proj shapes (BT=122880, K=12672, N=256).
+
+The bench's in_proj GEMMs (F.linear TN layout) run at ~1.5 TB/s: the
+6.3 MB weight panel is re-read by every M-tile and thrashes per-XCD L2.
+K-chunked accumulation keeps each weight chunk L2-resident. Also checks
+whether the plain NN layout (pre-transposed weight) is simply faster.
+"""
 import time
+
 import torch
+import torch.nn.functional as F
 
 torch.manual_seed(0)
 dev = torch.device("cuda")
 BT, K, N = 122880, 12672, 256
 A = torch.randn(BT, K, device=dev, dtype=torch.bfloat16)
-B = torch.randn(K, N, device=dev, dtype=torch.bfloat16)
+W = torch.randn(N, K, device=dev, dtype=torch.bfloat16)     # nn.Linear layout
+Wt = W.t().contiguous()                                     # (K, N) NN layout
 dY = torch.randn(BT, N, device=dev, dtype=torch.bfloat16)
+
 
 def timeit(fn, iters=20):
     for _ in range(5):
@@ -17,32 +27,29 @@ def timeit(fn, iters=20):
     for _ in range(iters):
         fn()
     torch.cuda.synchronize()
-    return (time.perf_counter() - t0) / iters * 1000
+    return round((time.perf_counter() - t0) / iters * 1000, 3)
 
-# forward: C = A @ B
-C = torch.empty(BT, N, device=dev, dtype=torch.bfloat16)
-print("fwd full     :", round(timeit(lambda: torch.mm(A, B, out=C)), 3), "ms")
-Cf = torch.empty(BT, N, device=dev, dtype=torch.float32)
-for nch in (4, 8, 16):
+
+print("fwd linear(A, W)  [TN, the model's path]:", timeit(lambda: F.linear(A, W)), "ms")
+print("fwd mm(A, Wt)     [NN, pre-transposed]  :", timeit(lambda: torch.mm(A, Wt)), "ms")
+for nch in (4, 8):
     ks = K // nch
     def chunked():
-        Cf.zero_()
-        for i in range(nch):
-            Cf.addmm_(A[:, i*ks:(i+1)*ks], B[i*ks:(i+1)*ks], beta=1.0)
-    print(f"fwd chunk{nch:3d} :", round(timeit(chunked), 3), "ms")
+        acc = torch.mm(A[:, :ks], Wt[:ks])
+        for i in range(1, nch):
+            acc += torch.mm(A[:, i * ks:(i + 1) * ks], Wt[i * ks:(i + 1) * ks])
+        return acc
+    print(f"fwd NN chunked x{nch}                     :", timeit(chunked), "ms")
 
-# dW = A^T @ dY  (the big-K gradient)
-W = torch.empty(K, N, device=dev, dtype=torch.bfloat16)
-print("dW full      :", round(timeit(lambda: torch.mm(A.t(), dY, out=W)), 3), "ms")
-Wf = torch.empty(K, N, device=dev, dtype=torch.float32)
-for nch in (4, 8, 16):
+print("dW mm(dY.t, A)    [the model's dW]      :", timeit(lambda: torch.mm(dY.t(), A)), "ms")
+print("dW mm(A.t, dY)    [transposed variant]  :", timeit(lambda: torch.mm(A.t(), dY)), "ms")
+for nch in (4, 8):
     bs = BT // nch
     def chunked_dw():
-        Wf.zero_()
-        for i in range(nch):
-            Wf.addmm_(A[i*bs:(i+1)*bs].t(), dY[i*bs:(i+1)*bs], beta=1.0)
-    print(f"dW chunk{nch:3d}  :", round(timeit(chunked_dw), 3), "ms")
+        acc = torch.mm(dY[:bs].t(), A[:bs])
+        for i in range(1, nch):
+            acc += torch.mm(dY[i * bs:(i + 1) * bs].t(), A[i * bs:(i + 1) * bs])
+        return acc
+    print(f"dW chunked x{nch}                         :", timeit(chunked_dw), "ms")
 
-# dX = dY @ B^T (N=K-wide output, K=256 small) — reference
-X2 = torch.empty(BT, K, device=dev, dtype=torch.bfloat16)
-print("dX full      :", round(timeit(lambda: torch.mm(dY, B.t(), out=X2)), 3), "ms")
+print("dX mm(dY, W)      [the model's dX]      :", timeit(lambda: torch.mm(dY, W)), "ms")
