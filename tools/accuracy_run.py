@@ -29,6 +29,9 @@ def main():
     ap.add_argument("--days", type=int, default=8)       # reference demo: 8 days
     ap.add_argument("--windows-per-day", type=int, default=240)
     ap.add_argument("--seed", type=int, default=77)
+    ap.add_argument("--residual", action="store_true",
+                    help="trace-ridge residual head")
+    ap.add_argument("--eval-every", type=int, default=1)
     args = ap.parse_args()
 
     app = SyntheticApp(SyntheticAppConfig(
@@ -40,7 +43,10 @@ def main():
     cfg = EngineConfig(
         data=DataConfig(step_size=60, split=0.40),
         train=TrainConfig(epochs=args.epochs, batch_size=32, lr=1e-3,
-                          eval_cycles=9, baseline_epochs=100, log_every=0),
+                          eval_cycles=9, baseline_epochs=100, log_every=0,
+                          eval_every=args.eval_every, graph_step=True,
+                          residual_base="trace-ridge" if args.residual
+                          else "none"),
         model=DeepRestNetConfig(dropout=0.1),
     )
     device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
