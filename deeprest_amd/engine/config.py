@@ -58,8 +58,11 @@ class TrainConfig:
                                           # extrapolates unseen traffic scale
                                           # linearly, the net corrects
                                           # in-range nonlinearity (quantiles
-                                          # are shift-equivariant, so the
-                                          # quantile semantics are unchanged)
+                                          # are shift-equivariant). Use when
+                                          # train rows >> call paths
+                                          # (measured: DEEPR 3.9 vs 7.8 at
+                                          # P=454; HURTS at P~rows, see
+                                          # profiles/r02_unseen_traffic.md)
 
 
 @dataclass
