@@ -90,3 +90,24 @@ def test_accuracy_probe_shape():
     assert acc["metrics"] == 5 * 3          # (4 comps + frontend) x 3 resources
     assert 0 <= acc["deepr_beats_comp"] <= acc["metrics"]
     assert acc["probe_seconds"] > 0
+
+
+def test_bench_torchrun_dp8_contract():
+    """The driver's largest launch shape: 8 ranks (gloo on CPU here, RCCL
+    on the node). Exercises 8-reader dataset sharing, rank walks, and the
+    rank-0-only JSON line at world size 8."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", "29538", "bench.py", "--gpus", "8",
+         "--steps", "1", "--warmup", "0", "--batch", "4",
+         "--endpoints", "6", "--components", "5", "--seq-len", "12"],
+        cwd=REPO, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, f"exactly one JSON line from rank 0, got {lines}"
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 8
+    assert d["config"]["parallelism"] == "dp8"
+    assert d["config"]["global_batch"] == 32
