@@ -83,13 +83,15 @@ def cmd_experiment(args):
     if args.suite:
         # unseen-traffic scenario suite needs a traffic-plan-capable app,
         # i.e. the synthetic generator (external data has no counterfactuals)
-        from .data.synthetic import SyntheticApp, SyntheticAppConfig
+        from .data.synthetic import ALL_RESOURCES, SyntheticApp, SyntheticAppConfig
 
         d = cfg.data
         app = SyntheticApp(SyntheticAppConfig(
             n_apis=d.synth_apis, n_components=d.synth_components,
             windows_per_day=d.synth_windows_per_day, n_days=d.synth_days,
             seed=d.synth_seed,
+            # the reference's full resource set (utils.py:8-26)
+            resources=ALL_RESOURCES,
         ))
         store = run_scenario_suite(app, base_name=args.name, config=cfg)
         from .engine.experiment import scenario_error_tables
