@@ -875,107 +875,6 @@ __global__ void gru_dgamma_kernel(const T* __restrict__ dpre,   // (BT, C, 4H) p
   }
 }
 
-// FUSED single-pass reductions: dpre (~4 GB at bench scale) is read ONCE
-// and produces all three outputs — dxg via block-staged LDS partials
-// (plain store when one c-block covers all C, f32 global atomics
-// otherwise), dgamma/dbeta4 via per-thread register accumulators flushed
-// once per block.  The two-kernel version read dpre twice at ~3 TB/s
-// each (2.58 ms/step at the flagship config); one pass is HBM-bound once.
-// Thread t owns (c_local = t/64 + 4k, gate = (t%64)/16, column group
-// cc = t%16): wave w streams row c_w fully coalesced (16 B/lane).
-template <typename T>
-__global__ __launch_bounds__(256) void gru_reduce_fused_kernel(
-    const T* __restrict__ dpre,   // (BT, C, 4H) pi layout
-    const T* __restrict__ xg,     // (BT, 3H) natural
-    const T* __restrict__ gamma,  // (C, 3H) natural
-    float* __restrict__ dxg,      // (BT, 3H) f32 zero-init
-    float* __restrict__ dgamma,   // (C, 3H) f32 zero-init
-    float* __restrict__ dbeta4,   // (C, 4H) f32 zero-init
-    int64_t BT, int C) {
-  __shared__ float dxg_s[G3H];
-  const int tid = threadIdx.x;
-  const int rem = tid & 63;
-  const int g = rem >> 4;            // gate 0..3 (3 = the d_hh_n slice)
-  const int cc = rem & 15;           // pi column group
-  const bool has_x = g < 3;
-  const int pi0 = g * H + cc * 8;
-  const int jn = g * H + cc;         // natural base column (elems at +16e)
-  const int64_t bt_lo = BT * blockIdx.y / gridDim.y;
-  const int64_t bt_hi = BT * (blockIdx.y + 1) / gridDim.y;
-
-  int cs[4];
-  int ncs = 0;
-  for (int k = 0; k < 4; ++k) {
-    int c = (int)blockIdx.x * 16 + (tid >> 6) + 4 * k;
-    if (c < C) cs[ncs++] = c;
-  }
-  float gm[4][8];
-  if (has_x) {
-    for (int k = 0; k < ncs; ++k) {
-      const T* gp = gamma + (int64_t)cs[k] * G3H + jn;
-#pragma unroll
-      for (int e = 0; e < 8; ++e) gm[k][e] = ldf(gp + e * 16);
-    }
-  }
-  float accg[4][8], accb[4][8];
-#pragma unroll
-  for (int k = 0; k < 4; ++k)
-#pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      accg[k][e] = 0.f;
-      accb[k][e] = 0.f;
-    }
-
-  for (int64_t bt = bt_lo; bt < bt_hi; ++bt) {
-    for (int i = tid; i < G3H; i += 256) dxg_s[i] = 0.f;
-    __syncthreads();
-    float xv[8];
-    if (has_x) {
-      const T* x = xg + bt * G3H + jn;
-#pragma unroll
-      for (int e = 0; e < 8; ++e) xv[e] = ldf(x + e * 16);
-    }
-    float dxacc[8];
-#pragma unroll
-    for (int e = 0; e < 8; ++e) dxacc[e] = 0.f;
-    for (int k = 0; k < ncs; ++k) {
-      float d[8];
-      ld8(dpre + (bt * C + cs[k]) * G4H + pi0, d);
-#pragma unroll
-      for (int e = 0; e < 8; ++e) accb[k][e] += d[e];
-      if (has_x) {
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          accg[k][e] += d[e] * xv[e];
-          dxacc[e] += d[e] * gm[k][e];
-        }
-      }
-    }
-    if (has_x) {
-#pragma unroll
-      for (int e = 0; e < 8; ++e) atomicAdd(&dxg_s[jn + 16 * e], dxacc[e]);
-    }
-    __syncthreads();
-    float* out = dxg + bt * G3H;
-    if (gridDim.x == 1) {            // sole contributor: plain store
-      for (int i = tid; i < G3H; i += 256) out[i] = dxg_s[i];
-    } else {
-      for (int i = tid; i < G3H; i += 256) atomicAdd(out + i, dxg_s[i]);
-    }
-  }
-
-  for (int k = 0; k < ncs; ++k) {
-    float* b_out = dbeta4 + (int64_t)cs[k] * G4H + jn;
-#pragma unroll
-    for (int e = 0; e < 8; ++e) atomicAdd(b_out + e * 16, accb[k][e]);
-    if (has_x) {
-      float* g_out = dgamma + (int64_t)cs[k] * G3H + jn;
-#pragma unroll
-      for (int e = 0; e < 8; ++e) atomicAdd(g_out + e * 16, accg[k][e]);
-    }
-  }
-}
-
 // ------------------------------------------------------------- launchers
 // fp8 tiles halve the GEMM-side LDS: 48K W + 8K h + 24K xg
 constexpr int LDS_FWD_FP8 = G3H * H + ROWS * H + XG_SLOTS * G3H * 2;  // 81920 B
