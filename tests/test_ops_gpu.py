@@ -529,3 +529,43 @@ def test_pinball_bf16_outputs(dev):
     lb.backward()
     assert a.grad.dtype == torch.bfloat16
     torch.testing.assert_close(a.grad.float(), b.grad, rtol=2e-2, atol=2e-3)
+
+
+def test_resource_aware_batch_on_device(dev):
+    """Batched RESRC baseline on GPU == CPU sequential at full batch."""
+    from deeprest_amd.models.baselines import (ResourceAwareBaseline,
+                                               ResourceAwareBatchBaseline)
+
+    rng = np.random.default_rng(7)
+    M, N, W = 4, 80, 12
+    y = np.cumsum(rng.normal(1.0, 0.3, size=(M, N, W)), axis=1) + 5.0
+    gpu_out = ResourceAwareBatchBaseline(
+        split=40, window=W, epochs=3, batch_size=10**6, seed=3,
+        device=dev).fit_and_estimate(y)
+    cpu_out = ResourceAwareBatchBaseline(
+        split=40, window=W, epochs=3, batch_size=10**6, seed=3
+    ).fit_and_estimate(y)
+    np.testing.assert_allclose(gpu_out, cpu_out, rtol=2e-4, atol=2e-4)
+
+
+def test_bench_graph_cycle_env(dev):
+    """DEEPREST_GRAPH_STEP=1 runs the zero-copy multi-graph training cycle
+    end-to-end (off by default; measured slower at the flagship config but
+    a supported mode)."""
+    import json as _json
+    import os as _os
+    import subprocess
+    import sys as _sys
+
+    repo = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+    env = dict(_os.environ, DEEPREST_GRAPH_STEP="1")
+    out = subprocess.run(
+        [_sys.executable, "bench.py", "--steps", "3", "--warmup", "1",
+         "--batch", "64", "--endpoints", "8", "--components", "7",
+         "--seq-len", "16", "--accuracy", "off"],
+        cwd=repo, env=env, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][0]
+    d = _json.loads(line)
+    assert d["value"] > 0
+    assert "graph cycle capture failed" not in out.stderr
