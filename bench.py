@@ -345,7 +345,8 @@ def main():
     if rank == 0:
         n_gpus = world if on_gpu else args.gpus
         print(json.dumps({
-            "metric": "training samples/sec, 256-endpoint 3-resource estimation model",
+            "metric": f"training samples/sec, {args.endpoints}-endpoint "
+                      "3-resource estimation model",
             "value": round(samples_per_sec, 2),
             "unit": "windows/s",
             "n_gpus": n_gpus,
