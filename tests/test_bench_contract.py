@@ -10,6 +10,8 @@ import os
 import subprocess
 import sys
 
+import numpy as np
+
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
@@ -90,6 +92,9 @@ def test_accuracy_probe_shape():
     assert acc["metrics"] == 5 * 3          # (4 comps + frontend) x 3 resources
     assert 0 <= acc["deepr_beats_comp"] <= acc["metrics"]
     assert acc["probe_seconds"] > 0
+    # the extended-convergence row the driver line carries
+    assert acc["extended"]["epochs"] == 2
+    assert np.isfinite(acc["extended"]["deepr"])
 
 
 def test_bench_torchrun_dp8_contract():
