@@ -25,7 +25,8 @@ from ..data.featurize import FeaturizedData
 from ..models.baselines import ComponentAwareBaseline
 from ..models.net import DeepRestNet, build_model_spec
 from ..ops.adam import FusedAdam
-from ..utils.errors import error_percentiles, format_error_table
+from ..utils.errors import (error_percentiles, format_error_table,
+                            quantile_coverage)
 from .checkpoint import load_checkpoint, save_checkpoint
 from .config import EngineConfig
 from .dataset import EstimationDataset
@@ -37,6 +38,9 @@ class TrainResult:
     train_losses: List[float] = field(default_factory=list)
     test_losses: List[float] = field(default_factory=list)
     error_tables: Dict[str, Dict[str, Dict[str, float]]] = field(default_factory=dict)
+    # per-metric calibration of the outer quantile band on the eval windows
+    # (nominal 0.90 for (.05, .95)); see utils.errors.quantile_coverage
+    coverage: Dict[str, Dict[str, float]] = field(default_factory=dict)
     samples_per_sec: float = 0.0
 
     def summary(self) -> str:
@@ -246,6 +250,7 @@ class Trainer:
                 test_loss, tables = self.evaluate(baselines)
                 result.test_losses.append(test_loss)
                 result.error_tables = tables
+                result.coverage = getattr(self, "last_coverage", {})
                 if cfg.log_every and (epoch + 1) % cfg.log_every == 0:
                     print(
                         f"Epoch [{epoch + 1}/{cfg.epochs}], "
@@ -288,6 +293,8 @@ class Trainer:
         outputs = np.maximum(out.cpu().numpy(), 1e-6)      # (K, T, M, Q)
         labels = yb.cpu().numpy()
         tables: Dict[str, Dict[str, Dict[str, float]]] = {}
+        coverage: Dict[str, Dict[str, float]] = {}
+        Q = outputs.shape[-1]
         median_q = len(self.model.cfg.quantiles) // 2
         for m, name in enumerate(ds.metric_names):
             labels_d = ds.denormalize_metric(labels[:, :, m], m).ravel()
@@ -299,6 +306,13 @@ class Trainer:
                     per_est[est_name] = error_percentiles(np.abs(bl - labels_d))
             per_est["deepr"] = error_percentiles(np.abs(pred_d - labels_d))
             tables[name] = per_est
+            # outer-band calibration (quantiles sorted: serving semantics)
+            band = np.sort(outputs[:, :, m, :], axis=-1)
+            coverage[name] = quantile_coverage(
+                labels_d,
+                ds.denormalize_metric(band[:, :, 0], m).ravel(),
+                ds.denormalize_metric(band[:, :, Q - 1], m).ravel())
+        self.last_coverage = coverage
         return test_loss, tables
 
     # ------------------------------------------------------------ checkpoints

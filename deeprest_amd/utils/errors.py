@@ -35,3 +35,23 @@ def format_error_table(name: str, per_estimator: Dict[str, Dict[str, float]]) ->
             % (est.upper(), stats["median"], stats["p95"], stats["p99"], stats["max"])
         )
     return "\n".join(lines)
+
+
+def quantile_coverage(labels: np.ndarray, q_lo: np.ndarray,
+                      q_hi: np.ndarray) -> Dict[str, float]:
+    """Empirical calibration of a quantile band: the fraction of held-out
+    observations inside [q_lo, q_hi] (nominal coverage for the (.05, .95)
+    band is 0.90), plus the one-sided miss rates.  The reference never
+    evaluates its bands; for the sanity-check use case (anomaly = utilization
+    outside the band) calibration IS the operating characteristic."""
+    y = np.asarray(labels, dtype=np.float64).ravel()
+    lo = np.asarray(q_lo, dtype=np.float64).ravel()
+    hi = np.asarray(q_hi, dtype=np.float64).ravel()
+    if y.size == 0:
+        return {"coverage": float("nan"), "below": float("nan"),
+                "above": float("nan")}
+    return {
+        "coverage": float(np.mean((y >= lo) & (y <= hi))),
+        "below": float(np.mean(y < lo)),
+        "above": float(np.mean(y > hi)),
+    }

@@ -207,3 +207,22 @@ def test_residual_base_trainer_and_predictor(tmp_path):
     exp = ds.denormalize_metric(full[:, :, 1], 0)
     np.testing.assert_allclose(got, np.maximum(exp, 1e-6), rtol=1e-3,
                                atol=1e-3)
+
+
+def test_band_coverage_reported():
+    from deeprest_amd.utils.errors import quantile_coverage
+
+    # unit semantics
+    y = np.array([1.0, 2.0, 3.0, 10.0])
+    cov = quantile_coverage(y, np.zeros(4), np.full(4, 5.0))
+    assert cov == {"coverage": 0.75, "below": 0.0, "above": 0.25}
+
+    data = tiny_data()
+    cfg = tiny_config(epochs=2)
+    cfg.train.run_baselines = False
+    trainer = Trainer(data, cfg, device=torch.device("cpu"))
+    res = trainer.train()
+    assert set(res.coverage.keys()) == set(data.metric_names)
+    for c in res.coverage.values():
+        assert 0.0 <= c["coverage"] <= 1.0
+        assert abs(c["coverage"] + c["below"] + c["above"] - 1.0) < 1e-9

@@ -66,11 +66,15 @@ def main():
         if per_est["deepr"]["median"] <= per_est["comp"]["median"]:
             wins_comp += 1
     import numpy as np
+    cov = [c["coverage"] for c in result.coverage.values()]
     print(json.dumps({
         "metrics": total,
         "deepr_beats_resrc": wins_resrc,
         "deepr_beats_comp": wins_comp,
         "mean_median_abs_err": {k: round(float(np.mean(v)), 4) for k, v in med.items()},
+        # calibration of the (.05,.95) band on the eval windows: nominal 0.90
+        "band_coverage_mean": round(float(np.mean(cov)), 4) if cov else None,
+        "band_coverage_min": round(float(np.min(cov)), 4) if cov else None,
         "epochs": args.epochs,
         "train_windows": trainer.dataset.split,
         "samples_per_sec": round(result.samples_per_sec, 1),
