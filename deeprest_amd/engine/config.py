@@ -52,6 +52,10 @@ class TrainConfig:
                                           # the final epoch always evaluates
     lr_schedule: str = "none"             # "none" (reference: fixed lr) or
                                           # "cosine" (decay to 5% over the run)
+    conformal: float = 0.0                # >0: target coverage for split-
+                                          # conformal band widening (CQR) on
+                                          # held-out calibration windows,
+                                          # e.g. 0.9 for the (.05,.95) band
     residual_base: str = "none"           # "trace-ridge": the net learns the
                                           # RESIDUAL over a closed-form ridge
                                           # on call-path features — the ridge

@@ -100,6 +100,7 @@ class Trainer:
         # outside its fitted min-max — profiles/r02_unseen_traffic.md)
         self._ridge_w: Optional[np.ndarray] = None       # (P+1, M) f64
         self._ridge_norm: Optional[np.ndarray] = None    # (N, T, M) f32
+        self._conformal: Optional[np.ndarray] = None     # (M,) band widening
         if self.cfg.train.residual_base == "trace-ridge":
             self._fit_residual_base()
         elif self.cfg.train.residual_base != "none":
@@ -264,7 +265,60 @@ class Trainer:
 
         elapsed = time.perf_counter() - t_start
         result.samples_per_sec = total_samples / max(elapsed, 1e-9)
+
+        if self.rank == 0 and cfg.conformal > 0.0:
+            self.conformalize(cfg.conformal)
+            # re-evaluate so the reported coverage reflects the widened band
+            test_loss, tables = self.evaluate(baselines)
+            result.test_losses.append(test_loss)
+            result.error_tables = tables
+            result.coverage = getattr(self, "last_coverage", {})
+            if cfg.checkpoint_path:
+                # the widening travels with the checkpoint (Predictor reads it)
+                self.save(cfg.checkpoint_path, cfg.epochs)
         return result
+
+    # ----------------------------------------------------------- conformal
+    @torch.no_grad()
+    def conformalize(self, target: float = 0.9, max_windows: int = 512) -> None:
+        """Split-conformal band widening (conformalized quantile regression):
+        on CALIBRATION windows — test-split windows disjoint from the
+        non-overlapping eval set — compute per metric the conformity score
+        s = max(q_lo - y, y - q_hi) in normalized space and widen the outer
+        band by its ceil((n+1)(1-a))/n quantile, giving finite-sample
+        coverage >= target under exchangeability.  The median is untouched,
+        so the error tables are unchanged."""
+        ds = self.dataset
+        eval_set = set(ds.eval_window_indices(self.cfg.train.eval_cycles))
+        n_test = ds.num_windows - ds.split
+        cal_idx = [iv for iv in range(n_test) if iv not in eval_set]
+        if not cal_idx:
+            cal_idx = list(range(n_test))      # degenerate tiny datasets
+        cal_idx = cal_idx[:max_windows]
+        self.model.eval()
+        xb = ds.X_test[cal_idx].to(self.device)
+        yb = ds.y_test[cal_idx].cpu().numpy()            # normalized (K,T,M)
+        with torch.autocast(
+            device_type=self.device.type, dtype=self.autocast_dtype,
+            enabled=(self.device.type == "cuda"),
+        ):
+            out = self.model(xb)
+        out = out.float().cpu().numpy()
+        rb = self.ridge_apply(ds.X_test[cal_idx].numpy())
+        if rb is not None:
+            out = out + rb[..., None]
+        band = np.sort(out, axis=-1)                     # (K,T,M,Q)
+        Q = band.shape[-1]
+        alpha = 1.0 - target
+        M = yb.shape[-1]
+        offs = np.zeros(M, dtype=np.float32)
+        for m in range(M):
+            s = np.maximum(band[:, :, m, 0] - yb[:, :, m],
+                           yb[:, :, m] - band[:, :, m, Q - 1]).ravel()
+            n = len(s)
+            q = min((np.ceil((n + 1) * (1.0 - alpha)) / n), 1.0)
+            offs[m] = np.quantile(s, q)
+        self._conformal = offs
 
     # ------------------------------------------------------------------- eval
     @torch.no_grad()
@@ -306,8 +360,12 @@ class Trainer:
                     per_est[est_name] = error_percentiles(np.abs(bl - labels_d))
             per_est["deepr"] = error_percentiles(np.abs(pred_d - labels_d))
             tables[name] = per_est
-            # outer-band calibration (quantiles sorted: serving semantics)
+            # outer-band calibration (quantiles sorted: serving semantics;
+            # conformal widening applied when fitted)
             band = np.sort(outputs[:, :, m, :], axis=-1)
+            if self._conformal is not None:
+                band[:, :, 0] -= self._conformal[m]
+                band[:, :, -1] += self._conformal[m]
             coverage[name] = quantile_coverage(
                 labels_d,
                 ds.denormalize_metric(band[:, :, 0], m).ravel(),
@@ -325,7 +383,8 @@ class Trainer:
             feature_space_state=self.feature_space_state,
             epoch=epoch,
             extra={"config": self.cfg.to_dict(),
-                   "residual_ridge": self._ridge_w},
+                   "residual_ridge": self._ridge_w,
+                   "conformal": self._conformal},
         )
 
     def load(self, path: str) -> None:

@@ -47,10 +47,13 @@ class Predictor:
         target_transform: str = "none",
         residual_ridge: Optional[np.ndarray] = None,   # (P+1, M): the net's
         # outputs are residuals over this ridge (train.residual_base)
+        conformal: Optional[np.ndarray] = None,        # (M,) CQR band widening
     ) -> None:
         self.target_transform = target_transform
         self.residual_ridge = (np.asarray(residual_ridge)
                                if residual_ridge is not None else None)
+        self.conformal = (np.asarray(conformal)
+                          if conformal is not None else None)
         self.device = device or torch.device(
             "cuda" if torch.cuda.is_available() else "cpu"
         )
@@ -79,6 +82,7 @@ class Predictor:
         kw.setdefault("target_transform", sc.get("target_transform", "none"))
         kw.setdefault("residual_ridge",
                       (state.get("extra") or {}).get("residual_ridge"))
+        kw.setdefault("conformal", (state.get("extra") or {}).get("conformal"))
         return Predictor(model, x_scaler, y_scalers, sc["metric_names"],
                          feature_space=fs, device=device, **kw)
 
@@ -175,6 +179,10 @@ class Predictor:
         # training; serving consumers (anomaly bands, demo plots) assume a
         # monotone triple, so sort the Q axis — a no-op once calibrated
         out = np.sort(out, axis=-1)
+        if self.conformal is not None:
+            # split-conformal band widening fitted at train time (CQR)
+            out[..., 0] -= self.conformal
+            out[..., -1] += self.conformal
         preds = {}
         for m, name in enumerate(self.metric_names):
             v = self.y_scalers[m].inverse_transform(out[:, :, m, :])
