@@ -31,6 +31,8 @@ def main():
     ap.add_argument("--seed", type=int, default=77)
     ap.add_argument("--residual", action="store_true",
                     help="trace-ridge residual head")
+    ap.add_argument("--conformal", type=float, default=0.0,
+                    help="target coverage for CQR band widening (e.g. 0.9)")
     ap.add_argument("--eval-every", type=int, default=1)
     args = ap.parse_args()
 
@@ -46,7 +48,7 @@ def main():
                           eval_cycles=9, baseline_epochs=100, log_every=0,
                           eval_every=args.eval_every, graph_step=True,
                           residual_base="trace-ridge" if args.residual
-                          else "none"),
+                          else "none", conformal=args.conformal),
         model=DeepRestNetConfig(dropout=0.1),
     )
     device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
