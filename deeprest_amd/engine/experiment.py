@@ -238,6 +238,11 @@ def run_scenario_suite(
             window=step, split=ds.split).fit(ds.y_raw[:, :, m]))
 
     store = ResultsStore()
+    # per scenario, per non-monotone resource: empirical coverage of the
+    # (conformalized) outer band UNDER the scenario's distribution shift —
+    # the conformal guarantee assumes exchangeability, so this measures how
+    # it degrades off-distribution (attached to the store, not pickled)
+    store.scenario_coverage = {}
     scenarios = DEFAULT_SCENARIOS if scenarios is None else scenarios
     for scen_name, plan_kw in scenarios:
         if plan_kw.get("composition") == "unseen":
@@ -269,6 +274,10 @@ def run_scenario_suite(
         if rb is not None:
             # residual head: ridge carries the unseen-scale extrapolation
             out = out + rb[..., None]
+        band = np.sort(out, axis=-1)                 # (K, T, M, Q) normalized
+        if trainer._conformal is not None:
+            band[..., 0] -= trainer._conformal
+            band[..., -1] += trainer._conformal
 
         n_flat = len(qdata.traffic)
         split_flat = ds.split + step - 1   # base learning period, flat steps
@@ -290,6 +299,19 @@ def run_scenario_suite(
             }
             preds = {k: np.maximum(np.asarray(v, dtype=np.float64), 1e-6)
                      for k, v in preds.items()}
+            if resource not in REANCHOR_METRICS:
+                from ..utils.errors import quantile_coverage
+
+                lo = ds.denormalize_metric(band[:, :, m, 0], m)
+                hi = ds.denormalize_metric(band[:, :, m, -1], m)
+                kk = min(len(eval_idx), len(measurement) // step)
+                if kk > 0:
+                    meas_w = np.stack([measurement[k * step:(k + 1) * step]
+                                       for k in range(kk)])
+                    cov = quantile_coverage(meas_w, lo[:kk], hi[:kk])
+                    store.scenario_coverage.setdefault(
+                        f"{base_name}-{scen_name}", {}).setdefault(
+                        resource, []).append(cov["coverage"])
             entry = build_results_entry(
                 measurement=measurement,
                 predictions=preds,

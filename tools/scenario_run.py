@@ -15,6 +15,7 @@ import sys
 
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
+import numpy as np
 import torch
 
 from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
@@ -41,6 +42,8 @@ def main():
     ap.add_argument("--no-residual", dest="residual", action="store_false",
                     help="disable the trace-ridge residual head (the "
                          "default head; carries unseen-scale extrapolation)")
+    ap.add_argument("--no-conformal", action="store_true",
+                    help="skip CQR band calibration")
     ap.set_defaults(residual=True)
     args = ap.parse_args()
 
@@ -58,7 +61,8 @@ def main():
         train=TrainConfig(epochs=args.epochs, batch_size=32, lr=1e-3,
                           run_baselines=True, log_every=0,
                           residual_base="trace-ridge" if args.residual
-                          else "none"),
+                          else "none",
+                          conformal=0.0 if args.no_conformal else 0.9),
         model=DeepRestNetConfig(dropout=0.1),
     )
     torch.manual_seed(0)
@@ -84,6 +88,13 @@ def main():
         for res, res_est in by_res[exp].items():
             print(f"  --- {res} ---")
             show(res_est, "  ")
+    cov = getattr(store, "scenario_coverage", {})
+    if cov:
+        print("===== band coverage under scenario shift "
+              "(nominal 0.90; non-monotone resources) =====")
+        for exp, per_res in cov.items():
+            parts = [f"{r}: {np.mean(v):.3f}" for r, v in per_res.items()]
+            print(f"  {exp}: " + " | ".join(parts))
     print(f"saved {args.out}")
 
 
