@@ -154,9 +154,15 @@ def test_scenario_suite_five_resources_smoke():
                           log_every=0),
         model=DeepRestNetConfig(d_model=32, n_heads=4, n_layers=1, d_ff=64,
                                 hidden=16, comp_dim=8, dropout=0.0))
+    cfg.train.conformal = 0.9
     store = run_scenario_suite(app, config=cfg, device=torch.device("cpu"),
                                scenarios=[("waves_unseen-3x", {"scale": 3.0})])
     exp = store.experiments()[0]
+    # coverage-under-shift side table (non-monotone resources only)
+    cov = store.scenario_coverage[exp]
+    assert set(cov.keys()) == {"cpu", "write-iops", "write-tp"}
+    for vals in cov.values():
+        assert all(0.0 <= v <= 1.0 for v in vals)
     comps = store.results[exp]
     # every component has entries for all five resource types
     for comp, metrics in comps.items():
