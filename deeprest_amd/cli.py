@@ -70,6 +70,13 @@ def cmd_train(args):
     trainer = Trainer(data, cfg)
     result = trainer.train()
     print(result.summary())
+    if result.coverage:
+        import numpy as _np
+
+        covs = [c["coverage"] for c in result.coverage.values()]
+        print(f"quantile-band coverage: mean {float(_np.mean(covs)):.3f} "
+              f"min {float(_np.min(covs)):.3f} (nominal "
+              f"{1.0 - 2 * cfg.model.quantiles[0]:.2f})")
     print(f"samples/sec: {result.samples_per_sec:.1f}")
     if cfg.train.checkpoint_path:
         print(f"checkpoint: {cfg.train.checkpoint_path}")

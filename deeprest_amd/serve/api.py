@@ -137,10 +137,19 @@ def create_app(checkpoint_path: Optional[str] = None, predictor: Optional[Predic
     def health():
         from ..ops import native_available
 
+        pred = state["predictor"]
         return {
             "status": "ok",
             "native_extension": native_available(),
-            "model_loaded": state["predictor"] is not None,
+            "model_loaded": pred is not None,
+            # estimator-head observability: what transforms this server
+            # applies on top of the net's quantile outputs
+            "residual_base": bool(getattr(pred, "residual_ridge", None)
+                                  is not None) if pred else None,
+            "conformal_bands": bool(getattr(pred, "conformal", None)
+                                    is not None) if pred else None,
+            "target_transform": getattr(pred, "target_transform", None)
+            if pred else None,
         }
 
     @app.post("/ingest")

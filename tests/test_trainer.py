@@ -259,3 +259,33 @@ def test_conformal_widening_improves_coverage(tmp_path):
     m0 = ds.metric_names[0]
     assert (out[m0][..., 0] <= out[m0][..., 1] + 1e-6).all()
     assert (out[m0][..., 1] <= out[m0][..., 2] + 1e-6).all()
+
+
+def test_log1p_residual_conformal_compose(tmp_path):
+    """All three estimator-head options compose: log1p targets +
+    trace-ridge residual + conformal bands, end to end through the
+    checkpointed Predictor with positive monotone outputs."""
+    from deeprest_amd.serve.predictor import Predictor
+
+    data = tiny_data()
+    cfg = tiny_config(tmp_path, epochs=2)
+    cfg.data.target_transform = "log1p"
+    cfg.train.run_baselines = False
+    cfg.train.residual_base = "trace-ridge"
+    cfg.train.conformal = 0.9
+    trainer = Trainer(data, cfg, device=torch.device("cpu"))
+    res = trainer.train()
+    assert np.isfinite(res.train_losses).all()
+    assert res.coverage and all(
+        0 <= c["coverage"] <= 1 for c in res.coverage.values())
+    pred = Predictor.from_checkpoint(cfg.train.checkpoint_path,
+                                     device=torch.device("cpu"))
+    assert (pred.target_transform, pred.residual_ridge is not None,
+            pred.conformal is not None) == ("log1p", True, True)
+    ds = trainer.dataset
+    x_norm = ds.X_test[:3].numpy().astype(np.float64)
+    raw = ds.x_scaler.min_val + x_norm * (ds.x_scaler.scale or 1.0)
+    out = pred.predict(raw)
+    for v in out.values():
+        assert (v > 0).all() and np.isfinite(v).all()
+        assert (np.diff(v, axis=-1) >= -1e-6).all()
