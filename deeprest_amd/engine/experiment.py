@@ -276,8 +276,11 @@ def run_scenario_suite(
             out = out + rb[..., None]
         band = np.sort(out, axis=-1)                 # (K, T, M, Q) normalized
         if trainer._conformal is not None:
-            band[..., 0] -= trainer._conformal
-            band[..., -1] += trainer._conformal
+            # clamp shrinks at the median (serving semantics, predictor.py)
+            band[..., 0] = np.minimum(band[..., 0] - trainer._conformal,
+                                      band[..., 1])
+            band[..., -1] = np.maximum(band[..., -1] + trainer._conformal,
+                                       band[..., -2])
 
         n_flat = len(qdata.traffic)
         split_flat = ds.split + step - 1   # base learning period, flat steps

@@ -364,8 +364,12 @@ class Trainer:
             # conformal widening applied when fitted)
             band = np.sort(outputs[:, :, m, :], axis=-1)
             if self._conformal is not None:
-                band[:, :, 0] -= self._conformal[m]
-                band[:, :, -1] += self._conformal[m]
+                # negative scores shrink; clamp at the median (serving
+                # semantics, predictor.py)
+                band[:, :, 0] = np.minimum(
+                    band[:, :, 0] - self._conformal[m], band[:, :, 1])
+                band[:, :, -1] = np.maximum(
+                    band[:, :, -1] + self._conformal[m], band[:, :, -2])
             coverage[name] = quantile_coverage(
                 labels_d,
                 ds.denormalize_metric(band[:, :, 0], m).ravel(),

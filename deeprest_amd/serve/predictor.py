@@ -180,9 +180,14 @@ class Predictor:
         # monotone triple, so sort the Q axis — a no-op once calibrated
         out = np.sort(out, axis=-1)
         if self.conformal is not None:
-            # split-conformal band widening fitted at train time (CQR)
-            out[..., 0] -= self.conformal
-            out[..., -1] += self.conformal
+            # split-conformal band widening fitted at train time (CQR);
+            # negative scores shrink the band — clamp at the median so the
+            # served triple stays monotone (clamping a shrink only raises
+            # coverage, the guarantee direction)
+            out[..., 0] = np.minimum(out[..., 0] - self.conformal,
+                                     out[..., 1])
+            out[..., -1] = np.maximum(out[..., -1] + self.conformal,
+                                      out[..., -2])
         preds = {}
         for m, name in enumerate(self.metric_names):
             v = self.y_scalers[m].inverse_transform(out[:, :, m, :])
