@@ -171,7 +171,8 @@ def cmd_serve(args):
 
     from .serve.api import create_app
 
-    app = create_app(checkpoint_path=args.checkpoint, results_path=args.results)
+    app = create_app(checkpoint_path=args.checkpoint, results_path=args.results,
+                     micro_batch=args.micro_batch)
     uvicorn.run(app, host=args.host, port=args.port)
     return 0
 
@@ -224,6 +225,9 @@ def main(argv=None):
     # binding must be an explicit opt-in (--host 0.0.0.0)
     v.add_argument("--host", default="127.0.0.1")
     v.add_argument("--port", type=int, default=2021)
+    v.add_argument("--micro-batch", action="store_true",
+                   help="coalesce concurrent /predict requests into one "
+                        "hipGraph replay (serve/batcher.py)")
 
     args = p.parse_args(argv)
     return {

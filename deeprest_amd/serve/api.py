@@ -10,6 +10,10 @@ Endpoints (FastAPI; run with `uvicorn deeprest_amd.serve.api:create_app`):
                                   ingested windows
 - POST /estimate                  what-if estimation for a traffic plan
                                   [{api: count}, ...] using the loaded model
+- POST /predict                   raw (N, T, P) call-path count windows ->
+                                  quantile predictions; with --micro-batch,
+                                  concurrent requests coalesce into one
+                                  hipGraph replay (serve/batcher.py)
 - POST /anomaly                   sanity-check measured series against the
                                   model's quantile band
 - GET  /apis                      known API endpoints (for what-if queries)
@@ -69,14 +73,22 @@ class IngestStore:
 
 
 def create_app(checkpoint_path: Optional[str] = None, predictor: Optional[Predictor] = None,
-               results_path: Optional[str] = None):
+               results_path: Optional[str] = None, micro_batch: bool = False,
+               micro_batch_wait_ms: float = 2.0):
     from fastapi import FastAPI, HTTPException
 
     app = FastAPI(title="deeprest-amd", version="0.1.0")
     store = IngestStore()
-    state = {"predictor": predictor, "results": None}
+    state = {"predictor": predictor, "results": None, "batcher": None}
     if checkpoint_path and predictor is None:
         state["predictor"] = Predictor.from_checkpoint(checkpoint_path)
+    if micro_batch and state["predictor"] is not None:
+        from .batcher import MicroBatcher
+
+        state["batcher"] = MicroBatcher(
+            state["predictor"],
+            max_batch=max(state["predictor"].graph_batches),
+            max_wait_ms=micro_batch_wait_ms)
     if results_path:
         from .results import ResultsStore
 
@@ -224,6 +236,32 @@ def create_app(checkpoint_path: Optional[str] = None, predictor: Optional[Predic
             "quantiles": [0.05, 0.50, 0.95],
             "predictions": {k: v.tolist() for k, v in out.items()},
         }
+
+    @app.post("/predict")
+    def predict_windows(payload: Dict[str, Any]):
+        """payload: {'windows': (N, T, P) raw call-path count windows} ->
+        per-metric quantile predictions. With micro-batching enabled,
+        concurrent requests coalesce into one hipGraph replay."""
+        pred: Optional[Predictor] = state["predictor"]
+        if pred is None:
+            raise HTTPException(status_code=400, detail="no model loaded")
+        w = payload.get("windows")
+        if w is None:
+            raise HTTPException(status_code=422, detail="windows required")
+        w = np.asarray(w, dtype=np.float64)
+        if w.ndim != 3 or w.shape[-1] != pred.model.spec.num_paths:
+            raise HTTPException(
+                status_code=422,
+                detail=f"windows must be (N, T, {pred.model.spec.num_paths})")
+        target = state["batcher"] or pred
+        out = target.predict(w)
+        resp = {k: v.tolist() for k, v in out.items()}
+        stats = None
+        if state["batcher"] is not None:
+            stats = {"batches_run": state["batcher"].batches_run,
+                     "requests_served": state["batcher"].requests_served}
+        return {"predictions": resp, "quantiles": list(pred.model.cfg.quantiles),
+                "micro_batch": stats}
 
     @app.post("/anomaly")
     def anomaly(payload: Dict[str, Any]):

@@ -334,3 +334,27 @@ def test_predictor_restores_target_transform(trained, tmp_path):
     first = out[data.metric_names[0]]
     raw_med = float(np.median(data.resources[data.metric_names[0]]))
     assert float(np.median(first)) > np.log1p(raw_med)  # clearly not log-space
+
+
+def test_predict_endpoint_with_micro_batching(trained):
+    from starlette.testclient import TestClient
+
+    from deeprest_amd.serve.api import create_app
+
+    app_syn, raw, data, cfg, ckpt = trained
+    client = TestClient(create_app(checkpoint_path=ckpt, micro_batch=True))
+    h = client.get("/health").json()
+    assert h["model_loaded"] is True
+    P = len(data.feature_space)
+    w = np.asarray(data.traffic[:24], dtype=float).reshape(2, 12, P)
+    r = client.post("/predict", json={"windows": w.tolist()})
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert body["quantiles"] == [0.05, 0.5, 0.95]
+    m0 = data.metric_names[0]
+    arr = np.asarray(body["predictions"][m0])
+    assert arr.shape == (2, 12, 3)
+    assert body["micro_batch"]["requests_served"] >= 1
+    # bad shape rejected
+    r2 = client.post("/predict", json={"windows": [[1, 2], [3, 4]]})
+    assert r2.status_code == 422
