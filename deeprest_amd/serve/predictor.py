@@ -174,7 +174,12 @@ class Predictor:
         256-endpoint width and GIL-serialized concurrent serving."""
         x = np.ascontiguousarray(np.asarray(traffic_windows),
                                  dtype=np.float32)
-        xt = torch.from_numpy(x).to(self.device)
+        return self.predict_tensor(torch.from_numpy(x).to(self.device))
+
+    def predict_tensor(self, xt: torch.Tensor) -> Dict[str, np.ndarray]:
+        """Raw count windows already on device as f32 (N, T, P) — the
+        zero-extra-copy entry the micro-batcher uses (client threads do
+        their own H2D, the batch concatenates on device)."""
         if self.x_scaler.scale != 0.0:
             xt = (xt - self.x_scaler.min_val) * (1.0 / self.x_scaler.scale)
         out = self.predict_normalized(xt).float()
