@@ -28,6 +28,11 @@ class MicroBatcher:
         self.max_wait_ms = max_wait_ms
         self.timeout_s = timeout_s
         self._lock = threading.Lock()
+        # one flush at a time: the Predictor's graph buffers are shared
+        # state (two overlapping flushes raced into concurrent graph
+        # capture); batches still COALESCE freely, execution serializes —
+        # the correct discipline for one GPU
+        self._run_lock = threading.Lock()
         self._pending: List[tuple] = []        # (windows, event, slot)
         self._timer: Optional[threading.Timer] = None
         self.batches_run = 0                   # observability
@@ -48,6 +53,10 @@ class MicroBatcher:
             self._run(batch)
 
     def _run(self, batch: List[tuple]) -> None:
+        with self._run_lock:
+            self._run_locked(batch)
+
+    def _run_locked(self, batch: List[tuple]) -> None:
         try:
             if isinstance(batch[0][0], torch.Tensor):
                 # device tensors (each client thread already did its H2D):
