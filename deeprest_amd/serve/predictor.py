@@ -176,6 +176,7 @@ class Predictor:
                                  dtype=np.float32)
         return self.predict_tensor(torch.from_numpy(x).to(self.device))
 
+    @torch.no_grad()
     def predict_tensor(self, xt: torch.Tensor) -> Dict[str, np.ndarray]:
         """Raw count windows already on device as f32 (N, T, P) — the
         zero-extra-copy entry the micro-batcher uses (client threads do
