@@ -569,3 +569,34 @@ def test_bench_graph_cycle_env(dev):
     d = _json.loads(line)
     assert d["value"] > 0
     assert "graph cycle capture failed" not in out.stderr
+
+
+def test_forward_long_gpu_matches_forward(dev):
+    """Long-horizon streaming path on the NATIVE kernels: a single chunk
+    covering the horizon equals the plain forward; multi-chunk runs stay
+    finite and the fp8 decode variant tracks bf16."""
+    from deeprest_amd.data.synthetic import SyntheticApp, SyntheticAppConfig
+    from deeprest_amd.models.net import (DeepRestNet, DeepRestNetConfig,
+                                         build_model_spec)
+
+    app = SyntheticApp(SyntheticAppConfig(
+        n_apis=5, n_components=6, windows_per_day=120, n_days=1, seed=23))
+    data = app.generate_featurized()
+    spec = build_model_spec(data)
+    torch.manual_seed(4)
+    model = DeepRestNet(spec, DeepRestNetConfig(
+        d_model=64, n_heads=2, n_layers=1, d_ff=128, hidden=128, comp_dim=16,
+        dropout=0.0)).to(dev).eval()
+    x = torch.rand(2, 96, spec.num_paths, device=dev)
+    with torch.no_grad():
+        full = model(x)
+        one_chunk = model.forward_long(x, chunk_size=96)
+        multi = model.forward_long(x, chunk_size=32)
+    torch.testing.assert_close(one_chunk, full, rtol=2e-4, atol=2e-4)
+    assert multi.shape == full.shape and torch.isfinite(multi).all()
+
+    model.cfg.fp8_inference = True
+    with torch.no_grad():
+        fp8_out = model.forward_long(x, chunk_size=96)
+    err = (fp8_out - full).abs().max().item()
+    assert err < 0.5, f"fp8 decode drifted: {err}"
