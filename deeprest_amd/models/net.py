@@ -404,6 +404,10 @@ class DeepRestNet(nn.Module):
                 x = layer(x)
             return x
 
+        # fp8 MFMA decode (BASELINE config 5) applies to the long-horizon
+        # path exactly as to forward()
+        fp8 = bool(cfg.fp8_inference and traffic.is_cuda)
+
         # forward sweep
         h_f = torch.tanh(dec.h0_proj(comp)).unsqueeze(0).expand(B, C, H).contiguous().to(dt)
         gamma = dec.cond_gamma(comp)
@@ -415,7 +419,7 @@ class DeepRestNet(nn.Module):
             enc_cache.append(enc)
             xg = dec.x_proj(enc)
             out = fused_gru_sequence(xg, dec.w_hh, dec.b_hh, h_f, gamma, beta,
-                                     reverse=False)
+                                     reverse=False, fp8=fp8)
             h_f = out[:, -1].contiguous()
             fwd_outs.append(out)
 
@@ -427,7 +431,8 @@ class DeepRestNet(nn.Module):
             for ci in range(len(chunks) - 1, -1, -1):
                 xg_r = dec.x_proj_r(enc_cache[ci])
                 out_r = fused_gru_sequence(xg_r, dec.w_hh_r, dec.b_hh_r, h_r,
-                                           gamma_r, beta_r, reverse=True)
+                                           gamma_r, beta_r, reverse=True,
+                                           fp8=fp8)
                 h_r = out_r[:, 0].contiguous()
                 rev_outs[ci] = out_r
             h_chunks = list(zip(fwd_outs, rev_outs))
